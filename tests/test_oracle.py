@@ -1,0 +1,98 @@
+"""Oracle self-consistency: the dpark-semantics restatement vs the
+independent numpy/pandas second oracle vs the committed golden fixtures,
+plus hypothesis property tests."""
+
+import glob
+import os
+
+import numpy as np
+import pytest
+from hypothesis import given, settings, strategies as st
+
+from oracle import dpark_oracle as dpo
+from oracle import second_oracle as so
+from oracle.synth import gen_pairs, splitmix64
+
+GOLDEN = glob.glob(os.path.join(os.path.dirname(__file__), "golden", "*.npz"))
+
+
+@pytest.mark.parametrize("path", GOLDEN, ids=[os.path.basename(p) for p in GOLDEN])
+def test_golden_fixtures(path):
+    g = np.load(path)
+    keys, vals = g["keys"], g["vals"]
+    uk, s = dpo.reduce_by_key_sum(keys, vals)
+    assert np.array_equal(uk, g["rbk_keys"])
+    np.testing.assert_allclose(s, g["rbk_sums"], rtol=1e-12)
+    assert np.array_equal(dpo.sort_by_key(keys), g["sorted_keys"])
+    uk2, s2 = so.reduce_by_key_sum(keys, vals)
+    assert np.array_equal(uk2, g["rbk_keys"])
+    np.testing.assert_allclose(s2, g["rbk_sums"], rtol=1e-12)
+
+
+pairs_strategy = st.lists(
+    st.tuples(st.integers(min_value=-(2**63), max_value=2**63 - 1),
+              st.floats(min_value=-1e6, max_value=1e6, allow_nan=False)),
+    max_size=300)
+
+
+@settings(max_examples=60, deadline=None)
+@given(pairs_strategy, st.integers(1, 7), st.integers(1, 5))
+def test_reduce_partition_independence(pairs, n_parts, n_maps):
+    """Contract §8(a1): result independent of partitioning / map split."""
+    keys = np.array([k for k, _ in pairs], dtype=np.int64)
+    vals = np.array([v for _, v in pairs], dtype=np.float64)
+    d = dpo.combine_by_key(pairs, lambda v: v, lambda a, b: a + b,
+                           lambda a, b: a + b, n_parts, n_maps)
+    uk, s = so.reduce_by_key_sum(keys, vals)
+    assert sorted(d.keys()) == uk.tolist()
+    got = np.array([d[int(k)] for k in uk])
+    np.testing.assert_allclose(got, s, rtol=1e-9, atol=1e-9)
+
+
+@settings(max_examples=40, deadline=None)
+@given(st.lists(st.integers(min_value=-(2**63), max_value=2**63 - 1),
+                max_size=300))
+def test_sort_total_order(keys):
+    keys = np.array(keys, dtype=np.int64)
+    out = dpo.sort_by_key(keys)
+    assert np.array_equal(out, so.sort_by_key(keys))
+    assert np.all(out[1:] >= out[:-1]) if len(out) > 1 else True
+    # same multiset
+    assert np.array_equal(np.sort(keys), out)
+
+
+@settings(max_examples=30, deadline=None)
+@given(pairs_strategy, pairs_strategy)
+def test_join_product_contract(pa, pb):
+    """join == per-key cross product; multiset comparison."""
+    jj = dpo.join(pa, pb)
+    ka = np.array([k for k, _ in pa], dtype=np.int64)
+    va = np.array([v for _, v in pa], dtype=np.float64)
+    kb = np.array([k for k, _ in pb], dtype=np.int64)
+    vb = np.array([v for _, v in pb], dtype=np.float64)
+    jk, jva, jvb = so.join_multiset(ka, va, kb, vb)
+    assert len(jj) == len(jk)
+    got = sorted((k, v, w) for k, (v, w) in jj)
+    want = sorted(zip(jk.tolist(), jva.tolist(), jvb.tolist()))
+    for (gk, gv, gw), (wk, wv, ww) in zip(got, want):
+        assert gk == wk
+        np.testing.assert_allclose([gv, gw], [wv, ww], rtol=1e-12)
+
+
+def test_synth_spec_stability():
+    """The synth spec is frozen: first values pinned so the device PRNG
+    has a fixed target (tests/test_gpu.py compares byte-identically)."""
+    k, v = gen_pairs(seed=42, n=4, key_universe=1 << 24)
+    assert k.tolist() == [int(x) & 0xFFFFFF for x in
+                          splitmix64((np.uint64(42) << np.uint64(32))
+                                     + np.arange(4, dtype=np.uint64))]
+    assert v.dtype == np.float64 and np.all((v >= 0) & (v < 1))
+    k2, v2 = gen_pairs(seed=42, n=4, key_universe=1 << 24)
+    assert np.array_equal(k, k2) and np.array_equal(v, v2)
+
+
+def test_group_by_key_multiset():
+    pairs = [(1, 10.0), (2, 20.0), (1, 30.0), (1, 10.0)]
+    g = dpo.group_by_key(pairs)
+    assert sorted(g[1]) == [10.0, 10.0, 30.0]
+    assert g[2] == [20.0]
