@@ -1,0 +1,89 @@
+"""DparkContext — mirrors dpark/context.py :: DparkContext [U]
+(reference mount empty, §0 — citations symbol-level).
+
+Master selection collapses to local modes on this single-node target:
+  'local'        — threaded LocalScheduler (dpark -m local, default)
+  'local[N]'     — N worker threads (config 1 uses local[4])
+Mesos / process modes are out of scope; the
+multi-GPU path is one process per GPU via dpark_amd.engine +
+dpark_amd.distshuffle, not a cluster scheduler.
+"""
+
+from __future__ import annotations
+
+import re
+
+from .rdd import (LocalScheduler, ParallelCollection, ShuffleStore,
+                  TextFileRDD, UnionRDD)
+
+
+class Broadcast:
+    """dpark/broadcast.py :: TreeBroadcast [U] degenerates to a value
+    holder on a single node (no zmq tree needed)."""
+
+    def __init__(self, value):
+        self.value = value
+
+
+class Accumulator:
+    """dpark/accumulator.py :: Accumulator [U] — single-process variant."""
+
+    def __init__(self, value=0, op=None):
+        self.value = value
+        self._op = op or (lambda a, b: a + b)
+
+    def add(self, v):
+        self.value = self._op(self.value, v)
+
+    __iadd__ = None  # use .add(), matching dpark's accum += in tasks [U?]
+
+
+class DparkContext:
+    def __init__(self, master="local", parallel=None):
+        m = re.match(r"local\[(\d+)\]", master or "local")
+        workers = int(m.group(1)) if m else 4
+        self.master = master
+        self.defaultParallelism = parallel or workers
+        self.shuffle_store = ShuffleStore()
+        self.scheduler = LocalScheduler(self, workers=workers)
+        self._stopped = False
+
+    # -- RDD constructors -------------------------------------------------
+    def parallelize(self, data, numSlices=None):
+        return ParallelCollection(self, data,
+                                  numSlices or self.defaultParallelism)
+
+    makeRDD = parallelize
+
+    def textFile(self, path, numSplits=None):
+        return TextFileRDD(self, path, numSplits)
+
+    def union(self, rdds):
+        return UnionRDD(self, list(rdds))
+
+    def broadcast(self, value):
+        return Broadcast(value)
+
+    def accumulator(self, value=0, op=None):
+        return Accumulator(value, op)
+
+    # -- job running ------------------------------------------------------
+    def runJob(self, rdd, func):
+        """dpark/context.py :: DparkContext.runJob -> DAGScheduler.runJob
+        [U]; returns one result per partition."""
+        return self.scheduler.run_job(rdd, func)
+
+    def runJob_collect(self, rdd):
+        out = []
+        for part in self.runJob(rdd, list):
+            out.extend(part)
+        return out
+
+    def stop(self):
+        self._stopped = True
+
+    def __enter__(self):
+        return self
+
+    def __exit__(self, *exc):
+        self.stop()
