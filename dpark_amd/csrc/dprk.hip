@@ -1,0 +1,1509 @@
+/* dprk — MI355X-native (gfx950/CDNA4) shuffle/sort/group-by engine.
+ *
+ * Brand-new kernels designed for CDNA4 (wave64, LDS-histogram radix
+ * machinery, HBM-streaming layout) — NOT a translation: the reference
+ * (douban/dpark) has zero GPU code; its hot path is Python dicts + pickle
+ * + file shuffle (citations [U]).  What each piece
+ * replaces is documented in include/dprk.h.
+ *
+ * Pipeline (DESIGN.md has the full data-layout and roofline derivation):
+ *   k_gen            device PRNG (splitmix64; spec mirrored in oracle/synth.py)
+ *   k_minmax_or      OR/AND reduce of biased keys -> digit-skip detection
+ *   k_tile_hist      per-tile 256-bin digit histogram      (8 B/pair read)
+ *   k_seg_sums/k_scan_mid/k_tile_off   3-level scan of tile histograms
+ *   k_scatter        rank + LDS exchange + coalesced write (32 B/pair r+w)
+ *   k_reduce_tiles / k_stitch / k_compact_runs   segmented float64 reduce
+ *   k_head_* / k_join_*                group/join machinery
+ * Sorting int64 ascending == unsigned radix on u = key XOR 0x8000...0.
+ */
+
+#include <hip/hip_runtime.h>
+#include <rccl/rccl.h>
+
+#include <cstdarg>
+#include <cstdio>
+#include <cstdint>
+#include <cstring>
+
+#include "../../include/dprk.h"
+
+/* ------------------------------------------------------------------ */
+/* tunables                                                            */
+/* ------------------------------------------------------------------ */
+#define RADIX 256
+#define RADIX_BITS 8
+#define TILE 4096          /* elements per scatter/reduce workgroup      */
+#define NTHREADS 256       /* 4 waves of 64                              */
+#define SEG_TILES 4096     /* tiles per mid-level scan segment           */
+#define ELEMS_PER_THREAD (TILE / NTHREADS) /* 16 */
+#define SCAN_CHUNK 16384   /* generic u64 scan: elements per workgroup   */
+#define MAX_EVTS 16384
+
+#define SIGNFLIP 0x8000000000000000ull
+
+/* ------------------------------------------------------------------ */
+/* error handling                                                      */
+/* ------------------------------------------------------------------ */
+static char g_err[2048] = "";
+
+static void set_err(const char *fmt, ...) {
+  va_list ap;
+  va_start(ap, fmt);
+  vsnprintf(g_err, sizeof(g_err), fmt, ap);
+  va_end(ap);
+}
+
+extern "C" const char *dprk_last_error(void) { return g_err; }
+
+#define DPRK_STR_(x) #x
+#define DPRK_STR(x) DPRK_STR_(x)
+extern "C" const char *dprk_build_info(void) {
+  return "dprk/0.1 gfx950 TILE=" DPRK_STR(TILE);
+}
+
+#define HIP_CHECK(x)                                                       \
+  do {                                                                     \
+    hipError_t e_ = (x);                                                   \
+    if (e_ != hipSuccess) {                                                \
+      set_err("HIP error '%s' at %s:%d", hipGetErrorString(e_), __FILE__,  \
+              __LINE__);                                                   \
+      return 1;                                                            \
+    }                                                                      \
+  } while (0)
+
+#define NCCL_CHECK(x)                                                      \
+  do {                                                                     \
+    ncclResult_t r_ = (x);                                                 \
+    if (r_ != ncclSuccess) {                                               \
+      set_err("RCCL error '%s' at %s:%d", ncclGetErrorString(r_),          \
+              __FILE__, __LINE__);                                         \
+      return 1;                                                            \
+    }                                                                      \
+  } while (0)
+
+/* ------------------------------------------------------------------ */
+/* context                                                             */
+/* ------------------------------------------------------------------ */
+enum KernKind {
+  K_GEN = 0, K_HIST, K_TILEHIST, K_SCAN, K_SCATTER, K_REDUCE, K_RUNEXT,
+  K_JOIN, K_EXCHANGE, K_OTHER, K_NKINDS
+};
+
+struct EvtRec { hipEvent_t a, b; int kind; };
+
+struct dprk_ctx {
+  int device = -1;
+  hipStream_t stream = nullptr;
+
+  /* current dataset: ping-pong buffers, engine-owned */
+  int64_t *keys[2] = {nullptr, nullptr};
+  double *vals[2] = {nullptr, nullptr};
+  int cur = 0;
+  uint64_t n = 0;
+  uint64_t cap = 0;
+  int with_vals = 0;
+  int sorted = 0;
+
+  /* radix scratch */
+  uint32_t *tile_hist = nullptr;  /* [RADIX][ntiles] raw                 */
+  uint32_t *tile_off = nullptr;   /* [RADIX][ntiles] scanned (+ base)    */
+  uint32_t *seg_sums = nullptr;   /* [RADIX][nseg]                       */
+  uint64_t *row_tot = nullptr;    /* [RADIX] per-bucket totals (partition)*/
+  uint64_t *d_orand = nullptr;    /* {OR, AND} of biased keys            */
+  uint64_t max_tiles = 0;
+
+  /* segmented-reduce / run-extract scratch */
+  double *tile_T = nullptr;       /* per-tile trailing open sum          */
+  uint32_t *tile_anyend = nullptr;
+  uint32_t *tile_nruns = nullptr;
+  uint64_t *tile_outoff = nullptr;
+  double *tile_C = nullptr;       /* stitched incoming carry             */
+  uint64_t *d_scalar = nullptr;   /* small readback slot (totals)        */
+
+  /* group layout (valid after dprk_group_by_key) */
+  int64_t *grp_keys = nullptr;
+  uint64_t *grp_starts = nullptr;
+  uint64_t *grp_counts = nullptr;
+  uint64_t n_groups = 0;
+
+  /* join output (valid after dprk_join) */
+  int64_t *join_k = nullptr;
+  double *join_a = nullptr, *join_b = nullptr;
+  uint64_t join_n = 0;
+  /* generic scan scratch */
+  uint64_t *scan_part = nullptr;
+  uint64_t scan_part_cap = 0;
+  uint64_t *run_prod = nullptr, *run_boff = nullptr;
+  uint64_t run_prod_cap = 0;
+
+  /* comm */
+  ncclComm_t comm = nullptr;
+  int nranks = 1, rank = 0;
+
+  /* timing */
+  hipEvent_t evts[2 * MAX_EVTS];
+  int n_evts_created = 0;
+  EvtRec recs[MAX_EVTS];
+  int n_recs = 0;
+  dprk_stats_t stats;
+};
+
+/* event-pair timing around a launch; drained at op end (after stream sync)
+ * so bench.py's roofline.achieved comes from HIP events on the launch
+ * stream, as required. */
+static int timed_begin(dprk_ctx *c, int kind) {
+  if (c->n_recs >= MAX_EVTS) return 0; /* drop timing, never the launch */
+  EvtRec &r = c->recs[c->n_recs];
+  if (2 * c->n_recs + 1 >= c->n_evts_created) {
+    if (hipEventCreate(&c->evts[c->n_evts_created]) != hipSuccess) return 0;
+    if (hipEventCreate(&c->evts[c->n_evts_created + 1]) != hipSuccess) return 0;
+    c->n_evts_created += 2;
+  }
+  r.a = c->evts[2 * c->n_recs];
+  r.b = c->evts[2 * c->n_recs + 1];
+  r.kind = kind;
+  hipEventRecord(r.a, c->stream);
+  return 1;
+}
+static void timed_end(dprk_ctx *c, int began) {
+  if (!began) return;
+  hipEventRecord(c->recs[c->n_recs].b, c->stream);
+  c->n_recs++;
+}
+
+#define TIMED(c, kind, launch)                                             \
+  do {                                                                     \
+    int began_ = timed_begin(c, kind);                                     \
+    launch;                                                                \
+    HIP_CHECK(hipGetLastError());                                          \
+    timed_end(c, began_);                                                  \
+  } while (0)
+
+static int drain_timing(dprk_ctx *c) {
+  HIP_CHECK(hipStreamSynchronize(c->stream));
+  double *ms[K_NKINDS] = {
+      &c->stats.ms_gen, &c->stats.ms_hist, &c->stats.ms_tilehist,
+      &c->stats.ms_scan, &c->stats.ms_scatter, &c->stats.ms_reduce,
+      &c->stats.ms_runext, &c->stats.ms_join, &c->stats.ms_exchange,
+      &c->stats.ms_other};
+  uint64_t *nn[K_NKINDS] = {
+      &c->stats.n_gen, &c->stats.n_hist, &c->stats.n_tilehist,
+      &c->stats.n_scan, &c->stats.n_scatter, &c->stats.n_reduce,
+      &c->stats.n_runext, &c->stats.n_join, &c->stats.n_exchange,
+      &c->stats.n_other};
+  for (int i = 0; i < c->n_recs; i++) {
+    float t = 0.f;
+    HIP_CHECK(hipEventElapsedTime(&t, c->recs[i].a, c->recs[i].b));
+    *ms[c->recs[i].kind] += t;
+    (*nn[c->recs[i].kind])++;
+  }
+  c->n_recs = 0;
+  return 0;
+}
+
+/* ------------------------------------------------------------------ */
+/* device utilities                                                    */
+/* ------------------------------------------------------------------ */
+__device__ __forceinline__ uint64_t d_splitmix64(uint64_t x) {
+  uint64_t z = x + 0x9E3779B97F4A7C15ull;
+  z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ull;
+  z = (z ^ (z >> 27)) * 0x94D049BB133111EBull;
+  return z ^ (z >> 31);
+}
+
+/* digit extraction.  mode 0: radix digit of the sign-biased key (sort);
+ * mode 1: TOP bits of splitmix64(key) (the GPU HashPartitioner,
+ * dpark/dependency.py :: HashPartitioner equivalent [U] — layout freedom
+ * per SURVEY.md §8(a2)). */
+__device__ __forceinline__ uint32_t d_digit(int64_t key, int mode, int shift,
+                                            uint32_t mask) {
+  uint64_t u = mode ? d_splitmix64((uint64_t)key) : ((uint64_t)key ^ SIGNFLIP);
+  return (uint32_t)(u >> shift) & mask;
+}
+
+/* ------------------------------------------------------------------ */
+/* kernels: generation                                                 */
+/* ------------------------------------------------------------------ */
+__global__ __launch_bounds__(NTHREADS) void k_gen(
+    int64_t *__restrict__ keys, double *__restrict__ vals, uint64_t n,
+    uint64_t seed_hi, uint64_t keymask) {
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    uint64_t x = seed_hi + i;
+    uint64_t k = d_splitmix64(x) & keymask;
+    keys[i] = (int64_t)k;
+    if (vals) {
+      uint64_t vb = d_splitmix64(x ^ 0xDEADBEEFCAFEBABEull);
+      vals[i] = (double)(vb >> 11) * (1.0 / 9007199254740992.0); /* 2^-53 */
+    }
+  }
+}
+
+/* OR/AND reduction of biased keys -> constant-digit (skip-pass) detection */
+__global__ __launch_bounds__(NTHREADS) void k_minmax_or(
+    const int64_t *__restrict__ keys, uint64_t n,
+    uint64_t *__restrict__ orand /* [2] pre-set to {0, ~0} */) {
+  __shared__ uint64_t s_or[NTHREADS / 64], s_and[NTHREADS / 64];
+  uint64_t o = 0, a = ~0ull;
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    uint64_t u = (uint64_t)keys[i] ^ SIGNFLIP;
+    o |= u;
+    a &= u;
+  }
+  /* wave reduce */
+  for (int off = 32; off; off >>= 1) {
+    o |= __shfl_down(o, off, 64);
+    a &= __shfl_down(a, off, 64);
+  }
+  int wave = threadIdx.x / 64, lane = threadIdx.x % 64;
+  if (lane == 0) { s_or[wave] = o; s_and[wave] = a; }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    for (int w = 1; w < NTHREADS / 64; w++) { o |= s_or[w]; a &= s_and[w]; }
+    atomicOr((unsigned long long *)&orand[0], (unsigned long long)o);
+    atomicAnd((unsigned long long *)&orand[1], (unsigned long long)a);
+  }
+}
+
+/* ------------------------------------------------------------------ */
+/* kernels: radix pass (tile histogram -> 3-level scan -> scatter)     */
+/* ------------------------------------------------------------------ */
+
+/* per-tile 256-bin histogram; one workgroup per tile; per-wave LDS
+ * replicas to cut same-digit ds_add serialization (SURVEY.md §7 hard
+ * part 2) */
+__global__ __launch_bounds__(NTHREADS) void k_tile_hist(
+    const int64_t *__restrict__ keys, uint64_t n,
+    uint32_t *__restrict__ tile_hist, uint64_t ntiles, int mode, int shift,
+    uint32_t mask) {
+  __shared__ uint32_t h[NTHREADS / 64][RADIX];
+  for (uint64_t t = blockIdx.x; t < ntiles; t += gridDim.x) {
+    int wave = threadIdx.x / 64;
+    for (int i = threadIdx.x; i < (NTHREADS / 64) * RADIX; i += NTHREADS)
+      ((uint32_t *)h)[i] = 0;
+    __syncthreads();
+    uint64_t start = t * (uint64_t)TILE;
+    uint64_t end = min(start + TILE, n);
+    for (uint64_t i = start + threadIdx.x; i < end; i += NTHREADS)
+      atomicAdd(&h[wave][d_digit(keys[i], mode, shift, mask)], 1u);
+    __syncthreads();
+    for (int d = threadIdx.x; d < RADIX; d += NTHREADS) {
+      uint32_t s = 0;
+      for (int w = 0; w < NTHREADS / 64; w++) s += h[w][d];
+      tile_hist[(uint64_t)d * ntiles + t] = s; /* digit-major for the scan */
+    }
+    __syncthreads();
+  }
+}
+
+/* scan level 1: per-(digit, segment) sums. grid.x = nseg, grid.y = RADIX */
+__global__ __launch_bounds__(NTHREADS) void k_seg_sums(
+    const uint32_t *__restrict__ tile_hist, uint64_t ntiles, uint64_t nseg,
+    uint32_t *__restrict__ seg_sums) {
+  __shared__ uint32_t sh[NTHREADS];
+  uint64_t d = blockIdx.y, s = blockIdx.x;
+  uint64_t t0 = s * (uint64_t)SEG_TILES;
+  uint64_t t1 = min(t0 + SEG_TILES, ntiles);
+  uint32_t acc = 0;
+  for (uint64_t t = t0 + threadIdx.x; t < t1; t += NTHREADS)
+    acc += tile_hist[d * ntiles + t];
+  sh[threadIdx.x] = acc;
+  __syncthreads();
+  for (int off = NTHREADS / 2; off; off >>= 1) {
+    if (threadIdx.x < off) sh[threadIdx.x] += sh[threadIdx.x + off];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) seg_sums[d * nseg + s] = sh[0];
+}
+
+/* scan level 2 (single workgroup): exclusive-scan each digit row of
+ * seg_sums in place; compute digit base offsets (exclusive over digit
+ * rows' totals) and fold them in; export per-digit totals (u64) for the
+ * host/partition path.  nseg <= 256 and RADIX == NTHREADS == 256: thread
+ * d owns digit row d. */
+__global__ __launch_bounds__(NTHREADS) void k_scan_mid(
+    uint32_t *__restrict__ seg_sums, uint64_t nseg,
+    uint64_t *__restrict__ row_tot) {
+  __shared__ uint64_t tot[RADIX];
+  int d = threadIdx.x;
+  uint64_t run = 0;
+  for (uint64_t s = 0; s < nseg; s++) {
+    uint32_t v = seg_sums[(uint64_t)d * nseg + s];
+    seg_sums[(uint64_t)d * nseg + s] = (uint32_t)run;
+    run += v;
+  }
+  tot[d] = run;
+  if (row_tot) row_tot[d] = run;
+  __syncthreads();
+  /* exclusive scan of tot[256] -> base, Hillis-Steele in LDS */
+  __shared__ uint64_t sc[RADIX];
+  sc[d] = tot[d];
+  __syncthreads();
+  for (int off = 1; off < RADIX; off <<= 1) {
+    uint64_t add = (d >= off) ? sc[d - off] : 0;
+    __syncthreads();
+    sc[d] += add;
+    __syncthreads();
+  }
+  uint64_t base = sc[d] - tot[d]; /* inclusive -> exclusive */
+  for (uint64_t s = 0; s < nseg; s++)
+    seg_sums[(uint64_t)d * nseg + s] += (uint32_t)base;
+  /* NOTE: offsets fit u32 because n < 2^32 (enforced in dprk_reserve) */
+}
+
+/* scan level 3: within each (digit, segment) produce exclusive tile
+ * offsets (+ segment base).  grid.x = nseg, grid.y = RADIX. */
+__global__ __launch_bounds__(NTHREADS) void k_tile_off(
+    const uint32_t *__restrict__ tile_hist, uint64_t ntiles, uint64_t nseg,
+    const uint32_t *__restrict__ seg_sums, uint32_t *__restrict__ tile_off) {
+  __shared__ uint32_t sh[NTHREADS];
+  uint64_t d = blockIdx.y, s = blockIdx.x;
+  uint64_t t0 = s * (uint64_t)SEG_TILES;
+  uint64_t t1 = min(t0 + SEG_TILES, ntiles);
+  uint32_t seg_base = seg_sums[d * nseg + s];
+  /* chunks of NTHREADS tiles: thread-per-tile load, LDS scan, running base */
+  uint32_t run = seg_base;
+  for (uint64_t c = t0; c < t1; c += NTHREADS) {
+    uint64_t t = c + threadIdx.x;
+    uint32_t v = (t < t1) ? tile_hist[d * ntiles + t] : 0;
+    sh[threadIdx.x] = v;
+    __syncthreads();
+    /* inclusive scan */
+    for (int off = 1; off < NTHREADS; off <<= 1) {
+      uint32_t add = (threadIdx.x >= (unsigned)off) ? sh[threadIdx.x - off] : 0;
+      __syncthreads();
+      sh[threadIdx.x] += add;
+      __syncthreads();
+    }
+    if (t < t1) tile_off[d * ntiles + t] = run + sh[threadIdx.x] - v;
+    uint32_t chunk_tot = sh[NTHREADS - 1];
+    __syncthreads();
+    run += chunk_tot;
+  }
+}
+
+/* the scatter: stable rank within tile + LDS exchange + coalesced write.
+ * One workgroup per tile (grid-stride), NTHREADS threads, TILE elements.
+ * LDS: keys 32 KiB + vals 32 KiB + control ~6 KiB => 2 blocks/CU. */
+__global__ __launch_bounds__(NTHREADS, 2) void k_scatter(
+    const int64_t *__restrict__ keys_in, const double *__restrict__ vals_in,
+    int64_t *__restrict__ keys_out, double *__restrict__ vals_out,
+    uint64_t n, const uint32_t *__restrict__ tile_hist,
+    const uint32_t *__restrict__ tile_off, uint64_t ntiles, int mode,
+    int shift, uint32_t mask) {
+  __shared__ int64_t lk[TILE];
+  __shared__ double lv[TILE];
+  __shared__ uint32_t excl[RADIX];      /* tile-local digit exclusive scan */
+  __shared__ uint32_t goff[RADIX];      /* global write base per digit     */
+  __shared__ uint32_t wcnt[NTHREADS / 64][RADIX];
+  __shared__ uint32_t drun[RADIX];      /* running per-digit count         */
+
+  const int wave = threadIdx.x / 64, lane = threadIdx.x % 64;
+
+  for (uint64_t t = blockIdx.x; t < ntiles; t += gridDim.x) {
+    const uint64_t start = t * (uint64_t)TILE;
+    const uint64_t end = min(start + TILE, n);
+    const uint32_t tile_n = (uint32_t)(end - start);
+
+    /* tile-local exclusive digit scan from the raw tile histogram */
+    {
+      uint32_t v = tile_hist[(uint64_t)threadIdx.x * ntiles + t];
+      excl[threadIdx.x] = v;
+      goff[threadIdx.x] = tile_off[(uint64_t)threadIdx.x * ntiles + t];
+      drun[threadIdx.x] = 0;
+      __syncthreads();
+      /* inclusive scan then shift */
+      for (int off = 1; off < RADIX; off <<= 1) {
+        uint32_t add = (threadIdx.x >= (unsigned)off) ? excl[threadIdx.x - off] : 0;
+        __syncthreads();
+        excl[threadIdx.x] += add;
+        __syncthreads();
+      }
+      uint32_t inc = excl[threadIdx.x];
+      __syncthreads();
+      excl[threadIdx.x] = inc - v;
+      __syncthreads();
+    }
+
+    /* rank + LDS scatter, NTHREADS elements per sub-iteration (order =
+     * element order => stable) */
+    for (uint32_t s = 0; s < tile_n; s += NTHREADS) {
+      uint32_t idx = s + threadIdx.x;
+      int64_t key = 0;
+      double val = 0.0;
+      uint32_t dig = 0;
+      bool active = idx < tile_n;
+      /* zero wave counters (4 per thread) */
+      for (int i = threadIdx.x; i < (NTHREADS / 64) * RADIX; i += NTHREADS)
+        ((uint32_t *)wcnt)[i] = 0;
+      __syncthreads();
+      uint32_t pos = 0;
+      if (active) {
+        key = keys_in[start + idx];
+        if (vals_in) val = vals_in[start + idx];
+        dig = d_digit(key, mode, shift, mask);
+        /* wave-level same-digit mask via 8 ballots */
+        uint64_t same = __ballot(1); /* active lanes */
+        for (int b = 0; b < RADIX_BITS; b++) {
+          uint64_t bb = __ballot((dig >> b) & 1);
+          same &= ((dig >> b) & 1) ? bb : ~bb;
+        }
+        uint64_t lowmask = ((uint64_t)1 << lane) - 1;
+        uint32_t rank_in_wave = (uint32_t)__popcll(same & lowmask);
+        if (rank_in_wave == 0)
+          wcnt[wave][dig] = (uint32_t)__popcll(same);
+        pos = rank_in_wave;
+      }
+      __syncthreads();
+      if (active) {
+        uint32_t before = drun[dig];
+        for (int w = 0; w < wave; w++) before += wcnt[w][dig];
+        uint32_t p = excl[dig] + before + pos;
+        lk[p] = key;
+        if (vals_in) lv[p] = val;
+      }
+      __syncthreads();
+      /* fold this sub-iteration's counts into the running totals */
+      {
+        uint32_t d = threadIdx.x;
+        uint32_t add = 0;
+        for (int w = 0; w < NTHREADS / 64; w++) add += wcnt[w][d];
+        drun[d] += add;
+      }
+      __syncthreads();
+    }
+
+    /* write out in digit-sorted tile order: runs of equal digit land in
+     * consecutive global addresses => coalesced segments */
+    for (uint32_t idx = threadIdx.x; idx < tile_n; idx += NTHREADS) {
+      int64_t key = lk[idx];
+      uint32_t dig = d_digit(key, mode, shift, mask);
+      uint32_t dst = goff[dig] + (idx - excl[dig]);
+      keys_out[dst] = key;
+      if (vals_in) vals_out[dst] = lv[idx];
+    }
+    __syncthreads();
+  }
+}
+
+/* ------------------------------------------------------------------ */
+/* kernels: segmented float64 reduce (reduceByKey(+))                  */
+/*   dpark reduce-side Merger dict-merge equivalent [U], deterministic */
+/*   fixed-tree per SURVEY.md §7 hard part 3.                          */
+/* ------------------------------------------------------------------ */
+
+/* per-tile pass: emit (key, within-tile run sum) for every run END in the
+ * tile, compacted into scratch at tile*TILE; record per-tile trailing open
+ * sum T, any-end flag and run count for the stitch. Input must be sorted.
+ * Run ends are globally exact (next-key lookahead crosses tiles). */
+__global__ __launch_bounds__(NTHREADS, 2) void k_reduce_tiles(
+    const int64_t *__restrict__ keys, const double *__restrict__ vals,
+    uint64_t n, int64_t *__restrict__ keys_scr, double *__restrict__ sums_scr,
+    double *__restrict__ tile_T, uint32_t *__restrict__ tile_anyend,
+    uint32_t *__restrict__ tile_nruns, uint64_t ntiles) {
+  __shared__ int64_t lk[TILE + 1];   /* +1: next tile's first key        */
+  __shared__ double lv[TILE];
+  __shared__ double t_carry[NTHREADS];   /* per-thread trailing open sum */
+  __shared__ uint32_t t_flag[NTHREADS];  /* thread chunk has a head      */
+  __shared__ uint32_t t_ends[NTHREADS];  /* run ends in thread chunk     */
+
+  for (uint64_t t = blockIdx.x; t < ntiles; t += gridDim.x) {
+    const uint64_t start = t * (uint64_t)TILE;
+    const uint64_t end = min(start + TILE, n);
+    const uint32_t tile_n = (uint32_t)(end - start);
+    for (uint32_t i = threadIdx.x; i < tile_n; i += NTHREADS) {
+      lk[i] = keys[start + i];
+      lv[i] = vals[start + i];
+    }
+    if (threadIdx.x == 0)
+      lk[tile_n] = (end < n) ? keys[end] : 0; /* lookahead (unused if end==n) */
+    __syncthreads();
+
+    const uint32_t c0 = threadIdx.x * ELEMS_PER_THREAD;         /* chunk */
+    const uint32_t c1 = min(c0 + (uint32_t)ELEMS_PER_THREAD, tile_n);
+
+    /* pass 1: trailing open sum + has-head + end count per thread chunk */
+    double open = 0.0;
+    uint32_t has_head = 0, n_ends = 0;
+    int64_t prev_key = 0;
+    bool have_prev = false;
+    if (c0 < tile_n) {
+      if (c0 == 0) {
+        if (start == 0) { have_prev = false; }
+        else { prev_key = keys[start - 1]; have_prev = true; }
+      } else { prev_key = lk[c0 - 1]; have_prev = true; }
+      for (uint32_t i = c0; i < c1; i++) {
+        bool head = !have_prev || lk[i] != prev_key;
+        if (head) { open = lv[i]; has_head = 1; }
+        else open += lv[i];
+        bool is_end = (start + i == n - 1) ||
+                      ((i + 1 < tile_n) ? lk[i + 1] != lk[i]
+                                        : lk[tile_n] != lk[i]);
+        if (is_end) n_ends++;
+        prev_key = lk[i];
+        have_prev = true;
+      }
+    }
+    t_carry[threadIdx.x] = open;
+    t_flag[threadIdx.x] = has_head;
+    t_ends[threadIdx.x] = n_ends;
+    __syncthreads();
+
+    /* cross-thread segmented inclusive scan of (carry, flag); and plain
+     * exclusive scan of end counts. Hillis-Steele, fixed tree. */
+    for (int off = 1; off < NTHREADS; off <<= 1) {
+      double s_prev = 0.0;
+      uint32_t f_prev = 0, e_prev = 0;
+      if (threadIdx.x >= (unsigned)off) {
+        s_prev = t_carry[threadIdx.x - off];
+        f_prev = t_flag[threadIdx.x - off];
+        e_prev = t_ends[threadIdx.x - off];
+      }
+      __syncthreads();
+      if (threadIdx.x >= (unsigned)off) {
+        if (!t_flag[threadIdx.x]) {
+          t_carry[threadIdx.x] += s_prev;
+          t_flag[threadIdx.x] = f_prev;
+        }
+        t_ends[threadIdx.x] += e_prev;
+      }
+      __syncthreads();
+    }
+    /* incoming within-tile carry for thread = inclusive seg-scan of the
+     * PREVIOUS thread; end-write base = exclusive scan of ends */
+    double incoming = 0.0;
+    if (threadIdx.x > 0 && c0 < tile_n) incoming = t_carry[threadIdx.x - 1];
+    /* t_flag after scan is unused below; t_ends is inclusive -> make
+     * exclusive via subtraction of own count */
+    uint32_t ends_excl = t_ends[threadIdx.x] - n_ends;
+    __syncthreads();
+
+    /* pass 2: walk chunk again with the incoming carry; emit run ends */
+    if (c0 < tile_n) {
+      double running = incoming;
+      int64_t prev_key2 = 0;
+      bool have_prev2 = false;
+      if (c0 == 0) {
+        if (start != 0) { prev_key2 = keys[start - 1]; have_prev2 = true; }
+      } else { prev_key2 = lk[c0 - 1]; have_prev2 = true; }
+      uint32_t w = ends_excl;
+      for (uint32_t i = c0; i < c1; i++) {
+        bool head = !have_prev2 || lk[i] != prev_key2;
+        if (head) running = lv[i];
+        else running += lv[i];
+        bool is_end = (start + i == n - 1) ||
+                      ((i + 1 < tile_n) ? lk[i + 1] != lk[i]
+                                        : lk[tile_n] != lk[i]);
+        if (is_end) {
+          keys_scr[start + w] = lk[i];
+          sums_scr[start + w] = running;
+          w++;
+        }
+        prev_key2 = lk[i];
+        have_prev2 = true;
+      }
+      /* last thread with data records the tile trailing state */
+      if (c1 == tile_n) {
+        bool last_is_end = (start + tile_n - 1 == n - 1) ||
+                           (lk[tile_n] != lk[tile_n - 1]);
+        tile_T[t] = last_is_end ? 0.0 : running;
+        /* any_end for the tile = total ends > 0 */
+        tile_anyend[t] = (t_ends[NTHREADS - 1] > 0) ? 1u : 0u;
+        tile_nruns[t] = t_ends[NTHREADS - 1];
+      }
+    }
+    __syncthreads();
+  }
+}
+
+/* stitch (single workgroup, chunked over tiles): segmented exclusive scan
+ * over tiles of (T, any_end) -> incoming cross-tile carry C_t; plain
+ * exclusive scan of nruns -> output offsets; writes total run count to
+ * d_total. */
+__global__ __launch_bounds__(NTHREADS) void k_stitch(
+    const double *__restrict__ tile_T, const uint32_t *__restrict__ tile_anyend,
+    const uint32_t *__restrict__ tile_nruns, uint64_t ntiles,
+    double *__restrict__ tile_C, uint64_t *__restrict__ tile_outoff,
+    uint64_t *__restrict__ d_total) {
+  __shared__ double sT[NTHREADS];
+  __shared__ uint32_t sF[NTHREADS];
+  __shared__ uint64_t sN[NTHREADS];
+  double carry = 0.0;      /* running open sum entering current chunk */
+  uint64_t off_run = 0;    /* running output offset */
+  for (uint64_t c = 0; c < ntiles; c += NTHREADS) {
+    uint64_t t = c + threadIdx.x;
+    bool in = t < ntiles;
+    sT[threadIdx.x] = in ? tile_T[t] : 0.0;
+    sF[threadIdx.x] = in ? tile_anyend[t] : 1u; /* pad breaks chains safely */
+    sN[threadIdx.x] = in ? (uint64_t)tile_nruns[t] : 0;
+    __syncthreads();
+    /* write exclusive results BEFORE turning the arrays inclusive:
+     * C_t = carry chained through preceding tiles in this chunk */
+    /* compute via sequential-in-log segmented scan; first save inputs */
+    double myT = sT[threadIdx.x];
+    uint32_t myF = sF[threadIdx.x];
+    uint64_t myN = sN[threadIdx.x];
+    for (int off = 1; off < NTHREADS; off <<= 1) {
+      double s_prev = 0.0;
+      uint32_t f_prev = 0;
+      uint64_t n_prev = 0;
+      if (threadIdx.x >= (unsigned)off) {
+        s_prev = sT[threadIdx.x - off];
+        f_prev = sF[threadIdx.x - off];
+        n_prev = sN[threadIdx.x - off];
+      }
+      __syncthreads();
+      if (threadIdx.x >= (unsigned)off) {
+        if (!sF[threadIdx.x]) {
+          sT[threadIdx.x] += s_prev;
+          sF[threadIdx.x] = f_prev;
+        }
+        sN[threadIdx.x] += n_prev;
+      }
+      __syncthreads();
+    }
+    if (in) {
+      /* exclusive versions for this tile */
+      double excl_T;
+      uint32_t excl_F;
+      if (threadIdx.x == 0) { excl_T = 0.0; excl_F = 0; }
+      else { excl_T = sT[threadIdx.x - 1]; excl_F = sF[threadIdx.x - 1]; }
+      /* C_t = within-chunk chained carry, + chunk-incoming carry if the
+       * chain reaches the chunk start (no any_end tile before t) */
+      tile_C[t] = excl_T + (excl_F ? 0.0 : carry);
+      tile_outoff[t] = off_run + (sN[threadIdx.x] - myN);
+    }
+    __syncthreads();
+    /* chunk-level carries from the last thread's inclusive values */
+    if (threadIdx.x == NTHREADS - 1) {
+      sT[0] = sF[threadIdx.x] ? sT[threadIdx.x] : (sT[threadIdx.x] + carry);
+      sF[0] = sF[threadIdx.x];
+      sN[0] = sN[threadIdx.x];
+    }
+    __syncthreads();
+    double new_carry = sT[0];
+    /* if no end in the whole chunk, carry keeps accumulating (handled:
+     * sF inclusive==0 path added old carry above) */
+    carry = new_carry;
+    off_run += sN[0];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) *d_total = off_run;
+}
+
+/* compact: copy each tile's runs from scratch to the final arrays; the
+ * FIRST run of a tile absorbs the stitched carry C_t (C_t == 0.0 whenever
+ * the tile's leading segment is empty — see DESIGN.md). */
+__global__ __launch_bounds__(NTHREADS) void k_compact_runs(
+    const int64_t *__restrict__ keys_scr, const double *__restrict__ sums_scr,
+    const uint32_t *__restrict__ tile_nruns,
+    const uint64_t *__restrict__ tile_outoff,
+    const double *__restrict__ tile_C, uint64_t ntiles,
+    int64_t *__restrict__ keys_out, double *__restrict__ sums_out) {
+  for (uint64_t t = blockIdx.x; t < ntiles; t += gridDim.x) {
+    uint32_t nr = tile_nruns[t];
+    uint64_t base = tile_outoff[t];
+    uint64_t src = t * (uint64_t)TILE;
+    for (uint32_t j = threadIdx.x; j < nr; j += NTHREADS) {
+      double s = sums_scr[src + j];
+      if (j == 0) s += tile_C[t];
+      keys_out[base + j] = keys_scr[src + j];
+      sums_out[base + j] = s;
+    }
+  }
+}
+
+/* ------------------------------------------------------------------ */
+/* kernels: run extraction (groupByKey / join structure)               */
+/* ------------------------------------------------------------------ */
+
+/* per tile: compact (key, head_global_index) for every run HEAD */
+__global__ __launch_bounds__(NTHREADS) void k_head_tiles(
+    const int64_t *__restrict__ keys, uint64_t n,
+    int64_t *__restrict__ keys_scr, uint64_t *__restrict__ idx_scr,
+    uint32_t *__restrict__ tile_nheads, uint64_t ntiles) {
+  __shared__ uint32_t cnt[NTHREADS];
+  for (uint64_t t = blockIdx.x; t < ntiles; t += gridDim.x) {
+    const uint64_t start = t * (uint64_t)TILE;
+    const uint64_t end = min(start + TILE, n);
+    /* count heads per thread (strided chunks of ELEMS_PER_THREAD) */
+    const uint64_t c0 = start + threadIdx.x * ELEMS_PER_THREAD;
+    const uint64_t c1 = min(c0 + ELEMS_PER_THREAD, end);
+    uint32_t nh = 0;
+    for (uint64_t i = c0; i < c1 && i < end; i++) {
+      bool head = (i == 0) || keys[i] != keys[i - 1];
+      if (head) nh++;
+    }
+    cnt[threadIdx.x] = nh;
+    __syncthreads();
+    for (int off = 1; off < NTHREADS; off <<= 1) {
+      uint32_t prev = (threadIdx.x >= (unsigned)off) ? cnt[threadIdx.x - off] : 0;
+      __syncthreads();
+      cnt[threadIdx.x] += prev;
+      __syncthreads();
+    }
+    uint32_t w = cnt[threadIdx.x] - nh;
+    for (uint64_t i = c0; i < c1 && i < end; i++) {
+      bool head = (i == 0) || keys[i] != keys[i - 1];
+      if (head) {
+        keys_scr[start + w] = keys[i];
+        idx_scr[start + w] = i;
+        w++;
+      }
+    }
+    if (threadIdx.x == NTHREADS - 1) tile_nheads[t] = cnt[threadIdx.x];
+    __syncthreads();
+  }
+}
+
+/* compact heads -> (unique_keys, starts) using scanned tile offsets */
+__global__ __launch_bounds__(NTHREADS) void k_compact_heads(
+    const int64_t *__restrict__ keys_scr, const uint64_t *__restrict__ idx_scr,
+    const uint32_t *__restrict__ tile_nheads,
+    const uint64_t *__restrict__ tile_outoff, uint64_t ntiles,
+    int64_t *__restrict__ uk, uint64_t *__restrict__ starts) {
+  for (uint64_t t = blockIdx.x; t < ntiles; t += gridDim.x) {
+    uint32_t nh = tile_nheads[t];
+    uint64_t base = tile_outoff[t];
+    uint64_t src = t * (uint64_t)TILE;
+    for (uint32_t j = threadIdx.x; j < nh; j += NTHREADS) {
+      uk[base + j] = keys_scr[src + j];
+      starts[base + j] = idx_scr[src + j];
+    }
+  }
+}
+
+__global__ __launch_bounds__(NTHREADS) void k_run_counts(
+    const uint64_t *__restrict__ starts, uint64_t g, uint64_t n,
+    uint64_t *__restrict__ counts) {
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < g;
+       i += stride)
+    counts[i] = ((i + 1 < g) ? starts[i + 1] : n) - starts[i];
+}
+
+/* plain exclusive-scan of tile_nheads -> tile_outoff (+ total) — single
+ * workgroup, chunked (ntiles <= 1M => <=4096 chunks) */
+__global__ __launch_bounds__(NTHREADS) void k_scan_u32_single(
+    const uint32_t *__restrict__ in, uint64_t m,
+    uint64_t *__restrict__ out, uint64_t *__restrict__ d_total) {
+  __shared__ uint64_t sh[NTHREADS];
+  uint64_t run = 0;
+  for (uint64_t c = 0; c < m; c += NTHREADS) {
+    uint64_t i = c + threadIdx.x;
+    uint64_t v = (i < m) ? (uint64_t)in[i] : 0;
+    sh[threadIdx.x] = v;
+    __syncthreads();
+    for (int off = 1; off < NTHREADS; off <<= 1) {
+      uint64_t prev = (threadIdx.x >= (unsigned)off) ? sh[threadIdx.x - off] : 0;
+      __syncthreads();
+      sh[threadIdx.x] += prev;
+      __syncthreads();
+    }
+    if (i < m) out[i] = run + sh[threadIdx.x] - v;
+    uint64_t tot = sh[NTHREADS - 1];
+    __syncthreads();
+    run += tot;
+  }
+  if (threadIdx.x == 0 && d_total) *d_total = run;
+}
+
+/* ------------------------------------------------------------------ */
+/* kernels: generic u64 exclusive scan (3-kernel, for large arrays)    */
+/* ------------------------------------------------------------------ */
+__global__ __launch_bounds__(NTHREADS) void k_scanA(
+    uint64_t *__restrict__ data, uint64_t m, uint64_t *__restrict__ part) {
+  __shared__ uint64_t sh[NTHREADS];
+  uint64_t c0 = (uint64_t)blockIdx.x * SCAN_CHUNK;
+  uint64_t c1 = min(c0 + SCAN_CHUNK, m);
+  uint64_t run = 0;
+  for (uint64_t c = c0; c < c1; c += NTHREADS) {
+    uint64_t i = c + threadIdx.x;
+    uint64_t v = (i < c1) ? data[i] : 0;
+    sh[threadIdx.x] = v;
+    __syncthreads();
+    for (int off = 1; off < NTHREADS; off <<= 1) {
+      uint64_t prev = (threadIdx.x >= (unsigned)off) ? sh[threadIdx.x - off] : 0;
+      __syncthreads();
+      sh[threadIdx.x] += prev;
+      __syncthreads();
+    }
+    if (i < c1) data[i] = run + sh[threadIdx.x] - v; /* exclusive in place */
+    uint64_t tot = sh[NTHREADS - 1];
+    __syncthreads();
+    run += tot;
+  }
+  if (threadIdx.x == 0) part[blockIdx.x] = run;
+}
+
+__global__ __launch_bounds__(NTHREADS) void k_scanB(
+    uint64_t *__restrict__ part, uint64_t nparts, uint64_t *__restrict__ d_total) {
+  /* same single-WG chunked exclusive scan, in place */
+  __shared__ uint64_t sh[NTHREADS];
+  uint64_t run = 0;
+  for (uint64_t c = 0; c < nparts; c += NTHREADS) {
+    uint64_t i = c + threadIdx.x;
+    uint64_t v = (i < nparts) ? part[i] : 0;
+    sh[threadIdx.x] = v;
+    __syncthreads();
+    for (int off = 1; off < NTHREADS; off <<= 1) {
+      uint64_t prev = (threadIdx.x >= (unsigned)off) ? sh[threadIdx.x - off] : 0;
+      __syncthreads();
+      sh[threadIdx.x] += prev;
+      __syncthreads();
+    }
+    if (i < nparts) part[i] = run + sh[threadIdx.x] - v;
+    uint64_t tot = sh[NTHREADS - 1];
+    __syncthreads();
+    run += tot;
+  }
+  if (threadIdx.x == 0 && d_total) *d_total = run;
+}
+
+__global__ __launch_bounds__(NTHREADS) void k_scanC(
+    uint64_t *__restrict__ data, uint64_t m, const uint64_t *__restrict__ part) {
+  uint64_t c0 = (uint64_t)blockIdx.x * SCAN_CHUNK;
+  uint64_t c1 = min(c0 + SCAN_CHUNK, m);
+  uint64_t base = part[blockIdx.x];
+  for (uint64_t i = c0 + threadIdx.x; i < c1; i += NTHREADS) data[i] += base;
+}
+
+/* ------------------------------------------------------------------ */
+/* kernels: join                                                       */
+/* ------------------------------------------------------------------ */
+
+/* per A-run: binary-search the key among B's unique keys; product size */
+__global__ __launch_bounds__(NTHREADS) void k_join_count(
+    const int64_t *__restrict__ uka, const uint64_t *__restrict__ cnta,
+    uint64_t ga, const int64_t *__restrict__ ukb,
+    const uint64_t *__restrict__ cntb, uint64_t gb,
+    uint64_t *__restrict__ prod, uint64_t *__restrict__ bidx) {
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < ga;
+       i += stride) {
+    int64_t k = uka[i];
+    uint64_t lo = 0, hi = gb;
+    while (lo < hi) {
+      uint64_t mid = (lo + hi) >> 1;
+      if (ukb[mid] < k) lo = mid + 1;
+      else hi = mid;
+    }
+    if (lo < gb && ukb[lo] == k) {
+      prod[i] = cnta[i] * cntb[lo];
+      bidx[i] = lo;
+    } else {
+      prod[i] = 0;
+      bidx[i] = (uint64_t)-1;
+    }
+  }
+}
+
+/* emit: one wave per A-run (grid-stride); writes (k, va, vb) tuples.
+ * dpark join semantics: per-key cross product. */
+__global__ __launch_bounds__(NTHREADS) void k_join_emit(
+    const int64_t *__restrict__ uka, const uint64_t *__restrict__ starta,
+    const uint64_t *__restrict__ cnta, uint64_t ga,
+    const uint64_t *__restrict__ startb, const uint64_t *__restrict__ cntb,
+    const uint64_t *__restrict__ prod_off, const uint64_t *__restrict__ prod,
+    const uint64_t *__restrict__ bidx, const double *__restrict__ va,
+    const double *__restrict__ vb, int64_t *__restrict__ out_k,
+    double *__restrict__ out_a, double *__restrict__ out_b) {
+  const int wave = threadIdx.x / 64, lane = threadIdx.x % 64;
+  const uint64_t nwaves = (uint64_t)gridDim.x * (NTHREADS / 64);
+  for (uint64_t r = (uint64_t)blockIdx.x * (NTHREADS / 64) + wave; r < ga;
+       r += nwaves) {
+    uint64_t p = prod[r];
+    if (p == 0) continue;
+    int64_t k = uka[r];
+    uint64_t sa = starta[r];
+    uint64_t b = bidx[r];
+    uint64_t sb = startb[b];
+    uint64_t cb = cntb[b];
+    uint64_t base = prod_off[r];
+    for (uint64_t j = lane; j < p; j += 64) {
+      out_k[base + j] = k;
+      out_a[base + j] = va[sa + j / cb];
+      out_b[base + j] = vb[sb + j % cb];
+    }
+  }
+}
+
+/* ------------------------------------------------------------------ */
+/* host-side helpers                                                   */
+/* ------------------------------------------------------------------ */
+static uint64_t ceil_div(uint64_t a, uint64_t b) { return (a + b - 1) / b; }
+
+/* memory-bound grid size: enough workgroups to fill 256 CUs at the
+ * kernel's blocks/CU, grid-stride beyond (Guideline 11) */
+static int stream_grid(uint64_t work_items) {
+  uint64_t g = ceil_div(work_items, NTHREADS);
+  if (g > 2048) g = 2048;
+  if (g == 0) g = 1;
+  return (int)g;
+}
+
+static int free_all(dprk_ctx *c);
+
+extern "C" int dprk_init(dprk_ctx **out, int device_id) {
+  *out = nullptr;
+  HIP_CHECK(hipSetDevice(device_id));
+  dprk_ctx *c = new dprk_ctx();
+  c->device = device_id;
+  memset(&c->stats, 0, sizeof(c->stats));
+  if (hipStreamCreateWithFlags(&c->stream, hipStreamNonBlocking) !=
+      hipSuccess) {
+    delete c;
+    set_err("hipStreamCreate failed");
+    return 1;
+  }
+  *out = c;
+  return 0;
+}
+
+extern "C" void dprk_free(dprk_ctx *c) {
+  if (!c) return;
+  hipSetDevice(c->device);
+  free_all(c);
+  for (int i = 0; i < c->n_evts_created; i++) hipEventDestroy(c->evts[i]);
+  if (c->comm) ncclCommDestroy(c->comm);
+  if (c->stream) hipStreamDestroy(c->stream);
+  delete c;
+}
+
+static int free_group_join(dprk_ctx *c) {
+  if (c->grp_keys) HIP_CHECK(hipFree(c->grp_keys));
+  if (c->grp_starts) HIP_CHECK(hipFree(c->grp_starts));
+  if (c->grp_counts) HIP_CHECK(hipFree(c->grp_counts));
+  c->grp_keys = nullptr; c->grp_starts = nullptr; c->grp_counts = nullptr;
+  c->n_groups = 0;
+  if (c->join_k) HIP_CHECK(hipFree(c->join_k));
+  if (c->join_a) HIP_CHECK(hipFree(c->join_a));
+  if (c->join_b) HIP_CHECK(hipFree(c->join_b));
+  c->join_k = nullptr; c->join_a = nullptr; c->join_b = nullptr;
+  c->join_n = 0;
+  if (c->run_prod) HIP_CHECK(hipFree(c->run_prod));
+  if (c->run_boff) HIP_CHECK(hipFree(c->run_boff));
+  c->run_prod = nullptr; c->run_boff = nullptr; c->run_prod_cap = 0;
+  return 0;
+}
+
+static int free_all(dprk_ctx *c) {
+  for (int i = 0; i < 2; i++) {
+    if (c->keys[i]) HIP_CHECK(hipFree(c->keys[i]));
+    if (c->vals[i]) HIP_CHECK(hipFree(c->vals[i]));
+    c->keys[i] = nullptr; c->vals[i] = nullptr;
+  }
+  if (c->tile_hist) HIP_CHECK(hipFree(c->tile_hist));
+  if (c->tile_off) HIP_CHECK(hipFree(c->tile_off));
+  if (c->seg_sums) HIP_CHECK(hipFree(c->seg_sums));
+  if (c->row_tot) HIP_CHECK(hipFree(c->row_tot));
+  if (c->d_orand) HIP_CHECK(hipFree(c->d_orand));
+  if (c->tile_T) HIP_CHECK(hipFree(c->tile_T));
+  if (c->tile_anyend) HIP_CHECK(hipFree(c->tile_anyend));
+  if (c->tile_nruns) HIP_CHECK(hipFree(c->tile_nruns));
+  if (c->tile_outoff) HIP_CHECK(hipFree(c->tile_outoff));
+  if (c->tile_C) HIP_CHECK(hipFree(c->tile_C));
+  if (c->d_scalar) HIP_CHECK(hipFree(c->d_scalar));
+  if (c->scan_part) HIP_CHECK(hipFree(c->scan_part));
+  c->tile_hist = nullptr; c->tile_off = nullptr; c->seg_sums = nullptr;
+  c->row_tot = nullptr; c->d_orand = nullptr; c->tile_T = nullptr;
+  c->tile_anyend = nullptr; c->tile_nruns = nullptr; c->tile_outoff = nullptr;
+  c->tile_C = nullptr; c->d_scalar = nullptr; c->scan_part = nullptr;
+  c->scan_part_cap = 0;
+  c->cap = 0; c->max_tiles = 0; c->n = 0;
+  return free_group_join(c);
+}
+
+extern "C" int dprk_reserve(dprk_ctx *c, uint64_t max_n, int with_vals) {
+  if (!c) { set_err("null ctx"); return 1; }
+  if (max_n >= (1ull << 32)) {
+    set_err("dprk_reserve: max_n %llu >= 2^32 unsupported (u32 radix "
+            "offsets; out-of-core chunking is a next-row item, SURVEY.md "
+            "§8(f))", (unsigned long long)max_n);
+    return 1;
+  }
+  HIP_CHECK(hipSetDevice(c->device));
+  if (max_n <= c->cap && (!with_vals || c->with_vals)) {
+    c->with_vals = with_vals;
+    return 0;
+  }
+  free_all(c);
+  if (max_n == 0) max_n = 1;
+  uint64_t ntiles = ceil_div(max_n, TILE);
+  uint64_t nseg = ceil_div(ntiles, SEG_TILES);
+  if (nseg > RADIX) {
+    set_err("ntiles too large");
+    return 1;
+  }
+  HIP_CHECK(hipMalloc(&c->keys[0], max_n * 8));
+  HIP_CHECK(hipMalloc(&c->keys[1], max_n * 8));
+  if (with_vals) {
+    HIP_CHECK(hipMalloc(&c->vals[0], max_n * 8));
+    HIP_CHECK(hipMalloc(&c->vals[1], max_n * 8));
+  }
+  HIP_CHECK(hipMalloc(&c->tile_hist, (uint64_t)RADIX * ntiles * 4));
+  HIP_CHECK(hipMalloc(&c->tile_off, (uint64_t)RADIX * ntiles * 4));
+  HIP_CHECK(hipMalloc(&c->seg_sums, (uint64_t)RADIX * RADIX * 4));
+  HIP_CHECK(hipMalloc(&c->row_tot, RADIX * 8));
+  HIP_CHECK(hipMalloc(&c->d_orand, 2 * 8));
+  HIP_CHECK(hipMalloc(&c->tile_T, ntiles * 8));
+  HIP_CHECK(hipMalloc(&c->tile_anyend, ntiles * 4));
+  HIP_CHECK(hipMalloc(&c->tile_nruns, ntiles * 4));
+  HIP_CHECK(hipMalloc(&c->tile_outoff, ntiles * 8));
+  HIP_CHECK(hipMalloc(&c->tile_C, ntiles * 8));
+  HIP_CHECK(hipMalloc(&c->d_scalar, 4 * 8));
+  c->cap = max_n;
+  c->max_tiles = ntiles;
+  c->with_vals = with_vals;
+  c->n = 0;
+  c->cur = 0;
+  c->sorted = 0;
+  return 0;
+}
+
+extern "C" int dprk_count(dprk_ctx *c, uint64_t *n_out) {
+  if (!c) { set_err("null ctx"); return 1; }
+  *n_out = c->n;
+  return 0;
+}
+
+extern "C" int dprk_gen_pairs(dprk_ctx *c, uint64_t seed, uint64_t n,
+                              int dist, uint64_t key_universe) {
+  if (!c) { set_err("null ctx"); return 1; }
+  if (dist != 0) { set_err("dist %d not implemented (zipf is a later "
+                           "milestone)", dist); return 1; }
+  if (n > c->cap) { set_err("gen: n > reserved cap"); return 1; }
+  HIP_CHECK(hipSetDevice(c->device));
+  uint64_t keymask =
+      (key_universe == 0) ? ~0ull : (key_universe - 1); /* power of two */
+  TIMED(c, K_GEN,
+        hipLaunchKernelGGL(k_gen, dim3(stream_grid(n)), dim3(NTHREADS), 0,
+                           c->stream, c->keys[c->cur],
+                           c->with_vals ? c->vals[c->cur] : nullptr, n,
+                           seed << 32, keymask));
+  c->n = n;
+  c->sorted = 0;
+  return drain_timing(c);
+}
+
+extern "C" int dprk_put_pairs(dprk_ctx *c, const int64_t *keys,
+                              const double *vals, uint64_t n) {
+  if (!c) { set_err("null ctx"); return 1; }
+  if (n > c->cap) { set_err("put: n > reserved cap"); return 1; }
+  if (vals && !c->with_vals) { set_err("put: ctx reserved keys-only"); return 1; }
+  HIP_CHECK(hipSetDevice(c->device));
+  HIP_CHECK(hipMemcpyAsync(c->keys[c->cur], keys, n * 8,
+                           hipMemcpyHostToDevice, c->stream));
+  if (vals && c->with_vals)
+    HIP_CHECK(hipMemcpyAsync(c->vals[c->cur], vals, n * 8,
+                             hipMemcpyHostToDevice, c->stream));
+  HIP_CHECK(hipStreamSynchronize(c->stream));
+  c->n = n;
+  c->sorted = 0;
+  return 0;
+}
+
+extern "C" int dprk_get_pairs(dprk_ctx *c, int64_t *keys, double *vals,
+                              uint64_t cap, uint64_t *n_out) {
+  if (!c) { set_err("null ctx"); return 1; }
+  *n_out = c->n;
+  if (c->n > cap) { set_err("get: cap %llu < n %llu",
+                            (unsigned long long)cap,
+                            (unsigned long long)c->n); return 1; }
+  HIP_CHECK(hipSetDevice(c->device));
+  if (keys)
+    HIP_CHECK(hipMemcpyAsync(keys, c->keys[c->cur], c->n * 8,
+                             hipMemcpyDeviceToHost, c->stream));
+  if (vals && c->with_vals)
+    HIP_CHECK(hipMemcpyAsync(vals, c->vals[c->cur], c->n * 8,
+                             hipMemcpyDeviceToHost, c->stream));
+  HIP_CHECK(hipStreamSynchronize(c->stream));
+  return 0;
+}
+
+/* ---- one radix pass: tile_hist -> 3-level scan -> scatter ---------- */
+static int radix_pass(dprk_ctx *c, int mode, int shift, uint32_t mask,
+                      int want_row_tot) {
+  uint64_t n = c->n;
+  uint64_t ntiles = ceil_div(n, TILE);
+  uint64_t nseg = ceil_div(ntiles, SEG_TILES);
+  int src = c->cur, dst = c->cur ^ 1;
+  int thgrid = (int)min(ntiles, (uint64_t)2048);
+  TIMED(c, K_TILEHIST,
+        hipLaunchKernelGGL(k_tile_hist, dim3(thgrid), dim3(NTHREADS), 0,
+                           c->stream, c->keys[src], n, c->tile_hist, ntiles,
+                           mode, shift, mask));
+  TIMED(c, K_SCAN,
+        hipLaunchKernelGGL(k_seg_sums, dim3((uint32_t)nseg, RADIX),
+                           dim3(NTHREADS), 0, c->stream, c->tile_hist,
+                           ntiles, nseg, c->seg_sums));
+  TIMED(c, K_SCAN,
+        hipLaunchKernelGGL(k_scan_mid, dim3(1), dim3(NTHREADS), 0, c->stream,
+                           c->seg_sums, nseg,
+                           want_row_tot ? c->row_tot : nullptr));
+  TIMED(c, K_SCAN,
+        hipLaunchKernelGGL(k_tile_off, dim3((uint32_t)nseg, RADIX),
+                           dim3(NTHREADS), 0, c->stream, c->tile_hist,
+                           ntiles, nseg, c->seg_sums, c->tile_off));
+  TIMED(c, K_SCATTER,
+        hipLaunchKernelGGL(k_scatter, dim3(thgrid), dim3(NTHREADS), 0,
+                           c->stream, c->keys[src],
+                           c->with_vals ? c->vals[src] : nullptr,
+                           c->keys[dst], c->with_vals ? c->vals[dst] : nullptr,
+                           n, c->tile_hist, c->tile_off, ntiles, mode, shift,
+                           mask));
+  c->stats.bytes_scatter += n * (c->with_vals ? 32 : 16);
+  c->stats.bytes_total += n * (c->with_vals ? 40 : 24); /* +hist read */
+  c->cur = dst;
+  return 0;
+}
+
+extern "C" int dprk_sort_pairs(dprk_ctx *c) {
+  if (!c) { set_err("null ctx"); return 1; }
+  HIP_CHECK(hipSetDevice(c->device));
+  if (c->n == 0) { c->sorted = 1; return 0; }
+  if (c->sorted) return 0;
+  /* digit-skip detection: OR/AND of biased keys */
+  uint64_t init[2] = {0, ~0ull};
+  HIP_CHECK(hipMemcpyAsync(c->d_orand, init, 16, hipMemcpyHostToDevice,
+                           c->stream));
+  TIMED(c, K_HIST,
+        hipLaunchKernelGGL(k_minmax_or, dim3(stream_grid(c->n)),
+                           dim3(NTHREADS), 0, c->stream, c->keys[c->cur],
+                           c->n, c->d_orand));
+  uint64_t orand[2];
+  HIP_CHECK(hipMemcpyAsync(orand, c->d_orand, 16, hipMemcpyDeviceToHost,
+                           c->stream));
+  HIP_CHECK(hipStreamSynchronize(c->stream));
+  c->stats.bytes_total += c->n * 8;
+  c->stats.sort_passes = 0;
+  for (int p = 0; p < 8; p++) {
+    uint32_t ob = (uint32_t)(orand[0] >> (8 * p)) & 0xFF;
+    uint32_t ab = (uint32_t)(orand[1] >> (8 * p)) & 0xFF;
+    if (ob == ab) continue; /* constant digit across all keys: skip pass */
+    if (radix_pass(c, /*mode=*/0, /*shift=*/8 * p, /*mask=*/0xFF, 0))
+      return 1;
+    c->stats.sort_passes++;
+  }
+  c->sorted = 1;
+  return drain_timing(c);
+}
+
+extern "C" int dprk_partition(dprk_ctx *c, int nbuckets,
+                              uint64_t *bucket_counts) {
+  if (!c) { set_err("null ctx"); return 1; }
+  if (nbuckets < 1 || nbuckets > 256 || (nbuckets & (nbuckets - 1))) {
+    set_err("partition: nbuckets must be a power of two in [1,256]");
+    return 1;
+  }
+  HIP_CHECK(hipSetDevice(c->device));
+  if (nbuckets == 1) {
+    if (bucket_counts) bucket_counts[0] = c->n;
+    return 0;
+  }
+  int bits = 0;
+  while ((1 << bits) < nbuckets) bits++;
+  if (c->n > 0) {
+    if (radix_pass(c, /*mode=*/1, /*shift=*/64 - bits,
+                   /*mask=*/(uint32_t)(nbuckets - 1), 1))
+      return 1;
+    c->sorted = 0;
+  } else {
+    HIP_CHECK(hipMemsetAsync(c->row_tot, 0, RADIX * 8, c->stream));
+  }
+  if (bucket_counts) {
+    uint64_t rt[RADIX];
+    HIP_CHECK(hipMemcpyAsync(rt, c->row_tot, RADIX * 8,
+                             hipMemcpyDeviceToHost, c->stream));
+    HIP_CHECK(hipStreamSynchronize(c->stream));
+    for (int b = 0; b < nbuckets; b++) bucket_counts[b] = rt[b];
+    if (c->n == 0)
+      for (int b = 0; b < nbuckets; b++) bucket_counts[b] = 0;
+  }
+  return drain_timing(c);
+}
+
+extern "C" int dprk_reduce_by_key_sum(dprk_ctx *c, uint64_t *n_out) {
+  if (!c) { set_err("null ctx"); return 1; }
+  if (!c->with_vals) { set_err("reduce: ctx reserved keys-only"); return 1; }
+  HIP_CHECK(hipSetDevice(c->device));
+  if (c->n == 0) { if (n_out) *n_out = 0; return 0; }
+  if (!c->sorted && dprk_sort_pairs(c)) return 1;
+  uint64_t n = c->n;
+  uint64_t ntiles = ceil_div(n, TILE);
+  int src = c->cur, scr = c->cur ^ 1;
+  int grid = (int)min(ntiles, (uint64_t)2048);
+  TIMED(c, K_REDUCE,
+        hipLaunchKernelGGL(k_reduce_tiles, dim3(grid), dim3(NTHREADS), 0,
+                           c->stream, c->keys[src], c->vals[src], n,
+                           c->keys[scr], c->vals[scr], c->tile_T,
+                           c->tile_anyend, c->tile_nruns, ntiles));
+  TIMED(c, K_REDUCE,
+        hipLaunchKernelGGL(k_stitch, dim3(1), dim3(NTHREADS), 0, c->stream,
+                           c->tile_T, c->tile_anyend, c->tile_nruns, ntiles,
+                           c->tile_C, c->tile_outoff, c->d_scalar));
+  uint64_t total = 0;
+  HIP_CHECK(hipMemcpyAsync(&total, c->d_scalar, 8, hipMemcpyDeviceToHost,
+                           c->stream));
+  HIP_CHECK(hipStreamSynchronize(c->stream));
+  /* compact runs back into the src buffers (sorted data now dead) */
+  TIMED(c, K_REDUCE,
+        hipLaunchKernelGGL(k_compact_runs, dim3(grid), dim3(NTHREADS), 0,
+                           c->stream, c->keys[scr], c->vals[scr],
+                           c->tile_nruns, c->tile_outoff, c->tile_C, ntiles,
+                           c->keys[src], c->vals[src]));
+  c->stats.bytes_total += n * 16 + total * 32;
+  c->n = total;
+  c->cur = src;
+  c->sorted = 1;
+  if (n_out) *n_out = total;
+  return drain_timing(c);
+}
+
+/* run extraction into ctx group arrays (assumes sorted) */
+static int extract_runs(dprk_ctx *c) {
+  uint64_t n = c->n;
+  uint64_t ntiles = ceil_div(n, TILE);
+  int src = c->cur, scr = c->cur ^ 1;
+  int grid = (int)min(ntiles, (uint64_t)2048);
+  /* idx scratch reuses the alt VALS buffer as u64 (same 8B stride);
+   * keys scratch uses the alt keys buffer */
+  if (!c->with_vals) { set_err("groups need vals buffers reserved"); return 1; }
+  TIMED(c, K_RUNEXT,
+        hipLaunchKernelGGL(k_head_tiles, dim3(grid), dim3(NTHREADS), 0,
+                           c->stream, c->keys[src], n, c->keys[scr],
+                           (uint64_t *)c->vals[scr], c->tile_nruns, ntiles));
+  TIMED(c, K_RUNEXT,
+        hipLaunchKernelGGL(k_scan_u32_single, dim3(1), dim3(NTHREADS), 0,
+                           c->stream, c->tile_nruns, ntiles, c->tile_outoff,
+                           c->d_scalar));
+  uint64_t g = 0;
+  HIP_CHECK(hipMemcpyAsync(&g, c->d_scalar, 8, hipMemcpyDeviceToHost,
+                           c->stream));
+  HIP_CHECK(hipStreamSynchronize(c->stream));
+  free_group_join(c);
+  HIP_CHECK(hipMalloc(&c->grp_keys, (g ? g : 1) * 8));
+  HIP_CHECK(hipMalloc(&c->grp_starts, (g ? g : 1) * 8));
+  HIP_CHECK(hipMalloc(&c->grp_counts, (g ? g : 1) * 8));
+  TIMED(c, K_RUNEXT,
+        hipLaunchKernelGGL(k_compact_heads, dim3(grid), dim3(NTHREADS), 0,
+                           c->stream, c->keys[scr], (uint64_t *)c->vals[scr],
+                           c->tile_nruns, c->tile_outoff, ntiles,
+                           c->grp_keys, c->grp_starts));
+  TIMED(c, K_RUNEXT,
+        hipLaunchKernelGGL(k_run_counts, dim3(stream_grid(g)), dim3(NTHREADS),
+                           0, c->stream, c->grp_starts, g, n, c->grp_counts));
+  c->stats.bytes_total += n * 8 + g * 24;
+  c->n_groups = g;
+  return drain_timing(c);
+}
+
+extern "C" int dprk_group_by_key(dprk_ctx *c, uint64_t *n_groups_out) {
+  if (!c) { set_err("null ctx"); return 1; }
+  HIP_CHECK(hipSetDevice(c->device));
+  if (c->n == 0) { free_group_join(c); c->n_groups = 0;
+                   if (n_groups_out) *n_groups_out = 0; return 0; }
+  if (!c->sorted && dprk_sort_pairs(c)) return 1;
+  if (extract_runs(c)) return 1;
+  if (n_groups_out) *n_groups_out = c->n_groups;
+  return 0;
+}
+
+extern "C" int dprk_get_groups(dprk_ctx *c, int64_t *unique_keys,
+                               uint64_t *starts, uint64_t *counts,
+                               uint64_t cap, uint64_t *n_out) {
+  if (!c) { set_err("null ctx"); return 1; }
+  *n_out = c->n_groups;
+  if (c->n_groups > cap) { set_err("get_groups: cap too small"); return 1; }
+  HIP_CHECK(hipSetDevice(c->device));
+  if (c->n_groups == 0) return 0;
+  if (unique_keys)
+    HIP_CHECK(hipMemcpyAsync(unique_keys, c->grp_keys, c->n_groups * 8,
+                             hipMemcpyDeviceToHost, c->stream));
+  if (starts)
+    HIP_CHECK(hipMemcpyAsync(starts, c->grp_starts, c->n_groups * 8,
+                             hipMemcpyDeviceToHost, c->stream));
+  if (counts)
+    HIP_CHECK(hipMemcpyAsync(counts, c->grp_counts, c->n_groups * 8,
+                             hipMemcpyDeviceToHost, c->stream));
+  HIP_CHECK(hipStreamSynchronize(c->stream));
+  return 0;
+}
+
+extern "C" int dprk_join(dprk_ctx *a, dprk_ctx *b, uint64_t *n_out) {
+  if (!a || !b) { set_err("null ctx"); return 1; }
+  if (a->device != b->device) { set_err("join: ctxs on different devices");
+                                return 1; }
+  HIP_CHECK(hipSetDevice(a->device));
+  if (a->n == 0 || b->n == 0) {
+    free_group_join(a);
+    a->join_n = 0;
+    if (n_out) *n_out = 0;
+    return 0;
+  }
+  /* sort + run-extract both sides (group layout lives per ctx) */
+  if (!a->sorted && dprk_sort_pairs(a)) return 1;
+  if (!b->sorted && dprk_sort_pairs(b)) return 1;
+  if (extract_runs(a)) return 1;
+  if (extract_runs(b)) return 1;
+  uint64_t ga = a->n_groups, gb = b->n_groups;
+  /* per-A-run product sizes + B match indices */
+  if (a->run_prod) { HIP_CHECK(hipFree(a->run_prod)); a->run_prod = nullptr; }
+  if (a->run_boff) { HIP_CHECK(hipFree(a->run_boff)); a->run_boff = nullptr; }
+  HIP_CHECK(hipMalloc(&a->run_prod, ga * 8));
+  HIP_CHECK(hipMalloc(&a->run_boff, ga * 8));
+  uint64_t *d_bidx = nullptr;
+  HIP_CHECK(hipMalloc(&d_bidx, ga * 8));
+  TIMED(a, K_JOIN,
+        hipLaunchKernelGGL(k_join_count, dim3(stream_grid(ga)), dim3(NTHREADS),
+                           0, a->stream, a->grp_keys, a->grp_counts, ga,
+                           b->grp_keys, b->grp_counts, gb, a->run_prod,
+                           d_bidx));
+  /* exclusive scan of products -> output offsets (run_boff := scan copy) */
+  HIP_CHECK(hipMemcpyAsync(a->run_boff, a->run_prod, ga * 8,
+                           hipMemcpyDeviceToDevice, a->stream));
+  uint64_t nchunks = ceil_div(ga, SCAN_CHUNK);
+  if (a->scan_part_cap < nchunks) {
+    if (a->scan_part) HIP_CHECK(hipFree(a->scan_part));
+    HIP_CHECK(hipMalloc(&a->scan_part, nchunks * 8));
+    a->scan_part_cap = nchunks;
+  }
+  TIMED(a, K_SCAN,
+        hipLaunchKernelGGL(k_scanA, dim3((uint32_t)nchunks), dim3(NTHREADS),
+                           0, a->stream, a->run_boff, ga, a->scan_part));
+  TIMED(a, K_SCAN,
+        hipLaunchKernelGGL(k_scanB, dim3(1), dim3(NTHREADS), 0, a->stream,
+                           a->scan_part, nchunks, a->d_scalar));
+  TIMED(a, K_SCAN,
+        hipLaunchKernelGGL(k_scanC, dim3((uint32_t)nchunks), dim3(NTHREADS),
+                           0, a->stream, a->run_boff, ga, a->scan_part));
+  uint64_t total = 0;
+  HIP_CHECK(hipMemcpyAsync(&total, a->d_scalar, 8, hipMemcpyDeviceToHost,
+                           a->stream));
+  HIP_CHECK(hipStreamSynchronize(a->stream));
+  /* allocate output and emit */
+  if (a->join_k) HIP_CHECK(hipFree(a->join_k));
+  if (a->join_a) HIP_CHECK(hipFree(a->join_a));
+  if (a->join_b) HIP_CHECK(hipFree(a->join_b));
+  a->join_k = nullptr; a->join_a = nullptr; a->join_b = nullptr;
+  HIP_CHECK(hipMalloc(&a->join_k, (total ? total : 1) * 8));
+  HIP_CHECK(hipMalloc(&a->join_a, (total ? total : 1) * 8));
+  HIP_CHECK(hipMalloc(&a->join_b, (total ? total : 1) * 8));
+  TIMED(a, K_JOIN,
+        hipLaunchKernelGGL(k_join_emit, dim3(2048), dim3(NTHREADS), 0,
+                           a->stream, a->grp_keys, a->grp_starts,
+                           a->grp_counts, ga, b->grp_starts, b->grp_counts,
+                           a->run_boff, a->run_prod, d_bidx,
+                           a->vals[a->cur], b->vals[b->cur], a->join_k,
+                           a->join_a, a->join_b));
+  a->stats.bytes_total += total * 24 + ga * 32 + gb * 16;
+  int rc = drain_timing(a);
+  HIP_CHECK(hipFree(d_bidx));
+  a->join_n = total;
+  if (n_out) *n_out = total;
+  return rc;
+}
+
+extern "C" int dprk_get_join(dprk_ctx *c, int64_t *keys, double *va,
+                             double *vb, uint64_t cap, uint64_t *n_out) {
+  if (!c) { set_err("null ctx"); return 1; }
+  *n_out = c->join_n;
+  if (c->join_n > cap) { set_err("get_join: cap too small"); return 1; }
+  HIP_CHECK(hipSetDevice(c->device));
+  if (c->join_n == 0) return 0;
+  if (keys)
+    HIP_CHECK(hipMemcpyAsync(keys, c->join_k, c->join_n * 8,
+                             hipMemcpyDeviceToHost, c->stream));
+  if (va)
+    HIP_CHECK(hipMemcpyAsync(va, c->join_a, c->join_n * 8,
+                             hipMemcpyDeviceToHost, c->stream));
+  if (vb)
+    HIP_CHECK(hipMemcpyAsync(vb, c->join_b, c->join_n * 8,
+                             hipMemcpyDeviceToHost, c->stream));
+  HIP_CHECK(hipStreamSynchronize(c->stream));
+  return 0;
+}
+
+/* ---- multi-GPU exchange ------------------------------------------- */
+extern "C" int dprk_comm_unique_id(char uid[128]) {
+  static_assert(sizeof(ncclUniqueId) == 128, "ncclUniqueId size");
+  NCCL_CHECK(ncclGetUniqueId((ncclUniqueId *)uid));
+  return 0;
+}
+
+extern "C" int dprk_comm_init(dprk_ctx *c, int nranks, int rank,
+                              const char uid[128]) {
+  if (!c) { set_err("null ctx"); return 1; }
+  HIP_CHECK(hipSetDevice(c->device));
+  ncclUniqueId id;
+  memcpy(&id, uid, sizeof(id));
+  NCCL_CHECK(ncclCommInitRank(&c->comm, nranks, id, rank));
+  c->nranks = nranks;
+  c->rank = rank;
+  return 0;
+}
+
+/* One grouped send/recv all-to-allv of the rank-partitioned dataset —
+ * the ShuffleFetcher replacement (pairwise-direct over the
+ * 7 xGMI p2p links; never a ring for this exchange). */
+extern "C" int dprk_exchange(dprk_ctx *c, const uint64_t *send_counts,
+                             const uint64_t *recv_counts, int nranks) {
+  if (!c) { set_err("null ctx"); return 1; }
+  if (nranks == 1) return 0;
+  if (!c->comm) { set_err("exchange: comm not initialized"); return 1; }
+  if (nranks != c->nranks) { set_err("exchange: nranks mismatch"); return 1; }
+  HIP_CHECK(hipSetDevice(c->device));
+  uint64_t send_off[256], recv_off[256], total_recv = 0, total_send = 0;
+  for (int r = 0; r < nranks; r++) {
+    send_off[r] = total_send;
+    recv_off[r] = total_recv;
+    total_send += send_counts[r];
+    total_recv += recv_counts[r];
+  }
+  if (total_send != c->n) { set_err("exchange: send counts != n"); return 1; }
+  if (total_recv > c->cap) { set_err("exchange: recv %llu > cap",
+                                     (unsigned long long)total_recv);
+                             return 1; }
+  int src = c->cur, dst = c->cur ^ 1;
+  int began = timed_begin(c, K_EXCHANGE);
+  NCCL_CHECK(ncclGroupStart());
+  for (int r = 0; r < nranks; r++) {
+    if (send_counts[r])
+      NCCL_CHECK(ncclSend(c->keys[src] + send_off[r], send_counts[r],
+                          ncclInt64, r, c->comm, c->stream));
+    if (recv_counts[r])
+      NCCL_CHECK(ncclRecv(c->keys[dst] + recv_off[r], recv_counts[r],
+                          ncclInt64, r, c->comm, c->stream));
+    if (c->with_vals) {
+      if (send_counts[r])
+        NCCL_CHECK(ncclSend(c->vals[src] + send_off[r], send_counts[r],
+                            ncclDouble, r, c->comm, c->stream));
+      if (recv_counts[r])
+        NCCL_CHECK(ncclRecv(c->vals[dst] + recv_off[r], recv_counts[r],
+                            ncclDouble, r, c->comm, c->stream));
+    }
+  }
+  NCCL_CHECK(ncclGroupEnd());
+  timed_end(c, began);
+  c->stats.bytes_total += (total_send + total_recv) * (c->with_vals ? 16 : 8);
+  c->cur = dst;
+  c->n = total_recv;
+  c->sorted = 0;
+  return drain_timing(c);
+}
+
+/* ---- stats --------------------------------------------------------- */
+extern "C" int dprk_stats(dprk_ctx *c, dprk_stats_t *out) {
+  if (!c || !out) { set_err("null arg"); return 1; }
+  *out = c->stats;
+  return 0;
+}
+
+extern "C" int dprk_stats_reset(dprk_ctx *c) {
+  if (!c) { set_err("null ctx"); return 1; }
+  memset(&c->stats, 0, sizeof(c->stats));
+  return 0;
+}

@@ -1,0 +1,282 @@
+"""ctypes binding to libdprk.so — the drop-in boundary (include/dprk.h).
+
+Loads the engine's C-ABI and exposes a numpy-facing Engine class.  On a
+machine WITH a GPU the HIP path is the only path: a missing/unbuildable
+extension raises EngineUnavailable loudly (no silent CPU fallback —
+tier rule ③).  CPU-only containers (this build box) can import this
+module and introspect symbols; calling compute ops without a GPU fails
+in HIP with a clear error.
+"""
+
+from __future__ import annotations
+
+import ctypes
+import os
+
+import numpy as np
+
+_HERE = os.path.dirname(os.path.abspath(__file__))
+_SO = os.path.join(_HERE, "libdprk.so")
+
+_lib = None
+_load_error = None
+
+
+class EngineUnavailable(RuntimeError):
+    pass
+
+
+class EngineError(RuntimeError):
+    pass
+
+
+class DprkStats(ctypes.Structure):
+    _fields_ = [
+        ("ms_gen", ctypes.c_double), ("ms_hist", ctypes.c_double),
+        ("ms_tilehist", ctypes.c_double), ("ms_scan", ctypes.c_double),
+        ("ms_scatter", ctypes.c_double), ("ms_reduce", ctypes.c_double),
+        ("ms_runext", ctypes.c_double), ("ms_join", ctypes.c_double),
+        ("ms_exchange", ctypes.c_double), ("ms_other", ctypes.c_double),
+        ("n_gen", ctypes.c_uint64), ("n_hist", ctypes.c_uint64),
+        ("n_tilehist", ctypes.c_uint64), ("n_scan", ctypes.c_uint64),
+        ("n_scatter", ctypes.c_uint64), ("n_reduce", ctypes.c_uint64),
+        ("n_runext", ctypes.c_uint64), ("n_join", ctypes.c_uint64),
+        ("n_exchange", ctypes.c_uint64), ("n_other", ctypes.c_uint64),
+        ("bytes_scatter", ctypes.c_uint64), ("bytes_total", ctypes.c_uint64),
+        ("sort_passes", ctypes.c_uint64),
+    ]
+
+    def as_dict(self):
+        return {name: getattr(self, name) for name, _ in self._fields_}
+
+
+# every symbol include/dprk.h declares — tests/test_cabi.py asserts these
+SYMBOLS = [
+    "dprk_init", "dprk_free", "dprk_last_error", "dprk_build_info",
+    "dprk_reserve", "dprk_gen_pairs", "dprk_put_pairs", "dprk_get_pairs",
+    "dprk_count", "dprk_partition", "dprk_sort_pairs",
+    "dprk_reduce_by_key_sum", "dprk_group_by_key", "dprk_get_groups",
+    "dprk_join", "dprk_get_join", "dprk_comm_unique_id", "dprk_comm_init",
+    "dprk_exchange", "dprk_stats", "dprk_stats_reset",
+]
+
+
+def load_library():
+    """Load (once) and signature-check libdprk.so. Raises EngineUnavailable
+    with the underlying cause if the native engine cannot be loaded."""
+    global _lib, _load_error
+    if _lib is not None:
+        return _lib
+    if _load_error is not None:
+        raise EngineUnavailable(_load_error)
+    if not os.path.exists(_SO):
+        _load_error = (f"libdprk.so not found at {_SO}; build it with "
+                       f"`python __graft_entry__.py build` (hipcc, gfx950)")
+        raise EngineUnavailable(_load_error)
+    try:
+        lib = ctypes.CDLL(_SO)
+    except OSError as e:
+        _load_error = f"failed to load {_SO}: {e}"
+        raise EngineUnavailable(_load_error)
+    for s in SYMBOLS:
+        if not hasattr(lib, s):
+            _load_error = f"{_SO} missing symbol {s}"
+            raise EngineUnavailable(_load_error)
+    u64 = ctypes.c_uint64
+    p = ctypes.c_void_p
+    lib.dprk_last_error.restype = ctypes.c_char_p
+    lib.dprk_build_info.restype = ctypes.c_char_p
+    lib.dprk_init.argtypes = [ctypes.POINTER(p), ctypes.c_int]
+    lib.dprk_free.argtypes = [p]
+    lib.dprk_free.restype = None
+    lib.dprk_reserve.argtypes = [p, u64, ctypes.c_int]
+    lib.dprk_gen_pairs.argtypes = [p, u64, u64, ctypes.c_int, u64]
+    lib.dprk_put_pairs.argtypes = [p, ctypes.c_void_p, ctypes.c_void_p, u64]
+    lib.dprk_get_pairs.argtypes = [p, ctypes.c_void_p, ctypes.c_void_p, u64,
+                                   ctypes.POINTER(u64)]
+    lib.dprk_count.argtypes = [p, ctypes.POINTER(u64)]
+    lib.dprk_partition.argtypes = [p, ctypes.c_int, ctypes.POINTER(u64)]
+    lib.dprk_sort_pairs.argtypes = [p]
+    lib.dprk_reduce_by_key_sum.argtypes = [p, ctypes.POINTER(u64)]
+    lib.dprk_group_by_key.argtypes = [p, ctypes.POINTER(u64)]
+    lib.dprk_get_groups.argtypes = [p, ctypes.c_void_p, ctypes.c_void_p,
+                                    ctypes.c_void_p, u64, ctypes.POINTER(u64)]
+    lib.dprk_join.argtypes = [p, p, ctypes.POINTER(u64)]
+    lib.dprk_get_join.argtypes = [p, ctypes.c_void_p, ctypes.c_void_p,
+                                  ctypes.c_void_p, u64, ctypes.POINTER(u64)]
+    lib.dprk_comm_unique_id.argtypes = [ctypes.c_char_p]
+    lib.dprk_comm_init.argtypes = [p, ctypes.c_int, ctypes.c_int,
+                                   ctypes.c_char_p]
+    lib.dprk_exchange.argtypes = [p, ctypes.POINTER(u64), ctypes.POINTER(u64),
+                                  ctypes.c_int]
+    lib.dprk_stats.argtypes = [p, ctypes.POINTER(DprkStats)]
+    lib.dprk_stats_reset.argtypes = [p]
+    _lib = lib
+    return lib
+
+
+def _check(lib, rc, what):
+    if rc != 0:
+        raise EngineError(f"{what}: {lib.dprk_last_error().decode()}")
+
+
+class Engine:
+    """One GPU's engine context (numpy in/out, device memory engine-owned)."""
+
+    def __init__(self, device=0):
+        self._lib = load_library()
+        self._ctx = ctypes.c_void_p()
+        _check(self._lib, self._lib.dprk_init(ctypes.byref(self._ctx), device),
+               "dprk_init")
+        self.device = device
+
+    def close(self):
+        if self._ctx:
+            self._lib.dprk_free(self._ctx)
+            self._ctx = ctypes.c_void_p()
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass
+
+    # -- data ------------------------------------------------------------
+    def reserve(self, max_n, with_vals=True):
+        _check(self._lib, self._lib.dprk_reserve(self._ctx, max_n,
+                                                 1 if with_vals else 0),
+               "dprk_reserve")
+
+    def gen_pairs(self, seed, n, key_universe=0, dist=0):
+        _check(self._lib, self._lib.dprk_gen_pairs(self._ctx, seed, n, dist,
+                                                   key_universe),
+               "dprk_gen_pairs")
+
+    def put(self, keys, vals=None):
+        keys = np.ascontiguousarray(keys, dtype=np.int64)
+        vp = None
+        if vals is not None:
+            vals = np.ascontiguousarray(vals, dtype=np.float64)
+            assert len(vals) == len(keys)
+            vp = vals.ctypes.data_as(ctypes.c_void_p)
+        _check(self._lib,
+               self._lib.dprk_put_pairs(self._ctx,
+                                        keys.ctypes.data_as(ctypes.c_void_p),
+                                        vp, len(keys)),
+               "dprk_put_pairs")
+
+    def get(self, with_vals=True):
+        n = self.count()
+        keys = np.empty(n, dtype=np.int64)
+        vals = np.empty(n, dtype=np.float64) if with_vals else None
+        nout = ctypes.c_uint64()
+        _check(self._lib,
+               self._lib.dprk_get_pairs(
+                   self._ctx, keys.ctypes.data_as(ctypes.c_void_p),
+                   vals.ctypes.data_as(ctypes.c_void_p) if with_vals else None,
+                   n, ctypes.byref(nout)),
+               "dprk_get_pairs")
+        return (keys, vals) if with_vals else keys
+
+    def count(self):
+        n = ctypes.c_uint64()
+        _check(self._lib, self._lib.dprk_count(self._ctx, ctypes.byref(n)),
+               "dprk_count")
+        return n.value
+
+    # -- ops -------------------------------------------------------------
+    def partition(self, nbuckets):
+        counts = (ctypes.c_uint64 * nbuckets)()
+        _check(self._lib, self._lib.dprk_partition(self._ctx, nbuckets, counts),
+               "dprk_partition")
+        return np.array(counts, dtype=np.uint64)
+
+    def sort(self):
+        _check(self._lib, self._lib.dprk_sort_pairs(self._ctx),
+               "dprk_sort_pairs")
+
+    def reduce_by_key_sum(self):
+        n = ctypes.c_uint64()
+        _check(self._lib,
+               self._lib.dprk_reduce_by_key_sum(self._ctx, ctypes.byref(n)),
+               "dprk_reduce_by_key_sum")
+        return n.value
+
+    def group_by_key(self):
+        n = ctypes.c_uint64()
+        _check(self._lib,
+               self._lib.dprk_group_by_key(self._ctx, ctypes.byref(n)),
+               "dprk_group_by_key")
+        g = n.value
+        uk = np.empty(g, dtype=np.int64)
+        st = np.empty(g, dtype=np.uint64)
+        cn = np.empty(g, dtype=np.uint64)
+        nout = ctypes.c_uint64()
+        _check(self._lib,
+               self._lib.dprk_get_groups(
+                   self._ctx, uk.ctypes.data_as(ctypes.c_void_p),
+                   st.ctypes.data_as(ctypes.c_void_p),
+                   cn.ctypes.data_as(ctypes.c_void_p), g, ctypes.byref(nout)),
+               "dprk_get_groups")
+        return uk, st, cn
+
+    def join(self, other):
+        n = ctypes.c_uint64()
+        _check(self._lib,
+               self._lib.dprk_join(self._ctx, other._ctx, ctypes.byref(n)),
+               "dprk_join")
+        j = n.value
+        k = np.empty(j, dtype=np.int64)
+        va = np.empty(j, dtype=np.float64)
+        vb = np.empty(j, dtype=np.float64)
+        nout = ctypes.c_uint64()
+        _check(self._lib,
+               self._lib.dprk_get_join(
+                   self._ctx, k.ctypes.data_as(ctypes.c_void_p),
+                   va.ctypes.data_as(ctypes.c_void_p),
+                   vb.ctypes.data_as(ctypes.c_void_p), j, ctypes.byref(nout)),
+               "dprk_get_join")
+        return k, va, vb
+
+    def join_count(self, other):
+        """join without fetching the (possibly huge) result."""
+        n = ctypes.c_uint64()
+        _check(self._lib,
+               self._lib.dprk_join(self._ctx, other._ctx, ctypes.byref(n)),
+               "dprk_join")
+        return n.value
+
+    # -- comm ------------------------------------------------------------
+    @staticmethod
+    def comm_unique_id():
+        lib = load_library()
+        buf = ctypes.create_string_buffer(128)
+        _check(lib, lib.dprk_comm_unique_id(buf), "dprk_comm_unique_id")
+        return buf.raw
+
+    def comm_init(self, nranks, rank, uid: bytes):
+        assert len(uid) == 128
+        _check(self._lib,
+               self._lib.dprk_comm_init(self._ctx, nranks, rank, uid),
+               "dprk_comm_init")
+
+    def exchange(self, send_counts, recv_counts):
+        nr = len(send_counts)
+        sc = (ctypes.c_uint64 * nr)(*[int(x) for x in send_counts])
+        rc = (ctypes.c_uint64 * nr)(*[int(x) for x in recv_counts])
+        _check(self._lib, self._lib.dprk_exchange(self._ctx, sc, rc, nr),
+               "dprk_exchange")
+
+    # -- stats -----------------------------------------------------------
+    def stats(self):
+        s = DprkStats()
+        _check(self._lib, self._lib.dprk_stats(self._ctx, ctypes.byref(s)),
+               "dprk_stats")
+        return s.as_dict()
+
+    def stats_reset(self):
+        _check(self._lib, self._lib.dprk_stats_reset(self._ctx),
+               "dprk_stats_reset")
+
+    def build_info(self):
+        return self._lib.dprk_build_info().decode()

@@ -17,6 +17,7 @@ import bisect
 import itertools
 import operator
 import os
+import threading
 from concurrent.futures import ThreadPoolExecutor
 
 from .portable_hash import portable_hash
@@ -118,9 +119,13 @@ class ShuffleDependency(Dependency):
 class ShuffleStore:
     def __init__(self):
         self._outputs = {}  # shuffleId -> {mapId: {bucket: items}}
+        self._complete = set()
 
     def put(self, shuffle_id, map_id, buckets):
         self._outputs.setdefault(shuffle_id, {})[map_id] = buckets
+
+    def mark_complete(self, shuffle_id):
+        self._complete.add(shuffle_id)
 
     def fetch(self, shuffle_id, reduce_id):
         """dpark/shuffle.py :: SimpleShuffleFetcher.fetch [U]: yields each
@@ -130,7 +135,7 @@ class ShuffleStore:
             yield maps[map_id].get(reduce_id, [])
 
     def has(self, shuffle_id):
-        return shuffle_id in self._outputs
+        return shuffle_id in self._complete
 
 
 # ---------------------------------------------------------------------------
@@ -513,11 +518,27 @@ class LocalScheduler:
     def __init__(self, ctx, workers=4):
         self.ctx = ctx
         self.workers = max(1, workers)
+        self._shuffle_lock = threading.RLock()
 
     def ensure_shuffle(self, dep: ShuffleDependency):
+        """Materialize a shuffle's map outputs exactly once; concurrent
+        reduce tasks (threads of run_job) serialize here — the map-output
+        'tracker' of dpark/tracker.py [U] degenerates to this lock +
+        in-process store on a single node."""
         store = self.ctx.shuffle_store
         if store.has(dep.shuffleId):
             return
+        with self._shuffle_lock:
+            self._ensure_shuffle_locked(dep)
+
+    def _ensure_shuffle_locked(self, dep: ShuffleDependency):
+        store = self.ctx.shuffle_store
+        if store.has(dep.shuffleId):
+            return
+        # pre-materialize ancestor shuffles from THIS thread (RLock
+        # reentrant) so pool worker threads never block on the lock —
+        # the DAGScheduler's parent-stage-first order of SURVEY.md §3.1
+        self._materialize_ancestors(dep.rdd)
         parent, agg, part = dep.rdd, dep.aggregator, dep.partitioner
 
         def shuffle_map_task(map_id, split):
@@ -539,9 +560,24 @@ class LocalScheduler:
                     lambda args: shuffle_map_task(*args),
                     list(enumerate(splits))):
                 store.put(dep.shuffleId, map_id, out)
+        store.mark_complete(dep.shuffleId)
+
+    def _materialize_ancestors(self, rdd, seen=None):
+        seen = seen if seen is not None else set()
+        if rdd.id in seen:
+            return
+        seen.add(rdd.id)
+        for d in rdd.dependencies:
+            if isinstance(d, ShuffleDependency):
+                self._materialize_ancestors(d.rdd, seen)
+                self._ensure_shuffle_locked(d)
+            else:
+                self._materialize_ancestors(d.rdd, seen)
 
     def run_job(self, rdd, func):
         splits = rdd.splits()
+        with self._shuffle_lock:
+            self._materialize_ancestors(rdd)
         # materialize shuffle deps bottom-up first (single-thread walk
         # avoids re-entrant pools); compute() calls ensure_shuffle lazily
         # as well, so this is an optimization not a correctness need.

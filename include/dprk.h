@@ -1,0 +1,133 @@
+/* dprk — MI355X-native shuffle/sort/group-by engine C-ABI.
+ *
+ * This boundary replaces douban/dpark's shuffle seam (all citations [U]
+ * symbol-level; the reference mount is empty — SURVEY.md §0, §8(b)):
+ *   - dpark/task.py :: ShuffleMapTask.run bucket loop  -> dprk_partition
+ *   - dpark/shuffle.py :: LocalFileShuffle write + ShuffleFetcher.fetch
+ *     (HTTP/file shuffle exchange)                     -> dprk_exchange
+ *       (one RCCL all-to-allv over xGMI)
+ *   - dpark/shuffle.py :: Merger / sort-based external merger
+ *     (reduce-side combine)                            -> dprk_sort_pairs +
+ *       dprk_reduce_by_key_sum (LSD radix sort + segmented reduce)
+ *   - dpark/rdd.py :: RDD.sortByKey per-partition sort -> dprk_sort_pairs
+ *   - dpark/rdd.py :: RDD.join / CoGroupedRDD + CoGroupMerger
+ *                                                      -> dprk_join
+ *   - dpark/dependency.py :: HashPartitioner/portable_hash
+ *     -> splitmix64 digits device-side (partition layout intentionally
+ *        differs; collected results are layout-independent, SURVEY.md
+ *        §8(a2))
+ *
+ * Conventions: every function returns int status (0 = OK); on failure
+ * dprk_last_error() describes the most recent error (per-process).  Caller
+ * owns host buffers (numpy); the engine owns all device memory.  One
+ * dprk_ctx per GPU; a ctx is not thread-safe; ops are synchronous on the
+ * ctx's private HIP stream.  Loaded via ctypes from dpark_amd/engine.py;
+ * the reference-side binding a dpark maintainer would add is shown in
+ * INTEGRATION.md.  No torch types anywhere.
+ */
+
+#ifndef DPRK_H
+#define DPRK_H
+
+#include <stddef.h>
+#include <stdint.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+typedef struct dprk_ctx dprk_ctx;
+
+/* ---- lifecycle ------------------------------------------------------- */
+int dprk_init(dprk_ctx **out, int device_id);
+void dprk_free(dprk_ctx *ctx);
+const char *dprk_last_error(void);
+/* engine/version marker so the host can assert the native path is live */
+const char *dprk_build_info(void);
+
+/* Reserve device capacity for up to max_n pairs (with_vals=0: keys only).
+ * Must be called before put/gen; re-reservation allowed. */
+int dprk_reserve(dprk_ctx *ctx, uint64_t max_n, int with_vals);
+
+/* ---- data in/out ----------------------------------------------------- */
+/* Device-side PRNG per the synth spec (oracle/synth.py is the CPU mirror):
+ * dist 0 = uniform over key_universe (power of two; 0 => full 64-bit),
+ * dist 1 = zipf(1.1) via inverse-CDF table (reserved, not yet implemented).
+ */
+int dprk_gen_pairs(dprk_ctx *ctx, uint64_t seed, uint64_t n, int dist,
+                   uint64_t key_universe);
+int dprk_put_pairs(dprk_ctx *ctx, const int64_t *keys, const double *vals,
+                   uint64_t n); /* vals may be NULL for keys-only */
+int dprk_get_pairs(dprk_ctx *ctx, int64_t *keys, double *vals, uint64_t cap,
+                   uint64_t *n_out); /* D2H of the current dataset */
+/* current dataset size */
+int dprk_count(dprk_ctx *ctx, uint64_t *n_out);
+
+/* ---- the hot path ---------------------------------------------------- */
+/* Radix-partition the current dataset into nbuckets contiguous segments by
+ * the TOP bits of splitmix64(key) (the GPU HashPartitioner — SURVEY.md
+ * §8(a2)); bucket_counts (host, nbuckets entries) receives the segment
+ * sizes.  nbuckets must be a power of two <= 256. */
+int dprk_partition(dprk_ctx *ctx, int nbuckets, uint64_t *bucket_counts);
+
+/* LSD radix sort of the current dataset by key, ascending (int64 order),
+ * values carried.  Passes over constant digit positions are skipped (the
+ * pass count actually executed is reported via dprk_stats). */
+int dprk_sort_pairs(dprk_ctx *ctx);
+
+/* reduceByKey(+): sort (if needed) + segmented reduce.  Replaces the
+ * current dataset with (unique_keys_sorted, per-key float64 sums); n_out
+ * receives the number of unique keys.  Deterministic fixed-tree float64
+ * reduction: run-to-run bit-identical; vs a sequential CPU oracle the
+ * contract is <=1e-6 relative. */
+int dprk_reduce_by_key_sum(dprk_ctx *ctx, uint64_t *n_out);
+
+/* groupByKey structure: sort (if needed), then expose grouped layout:
+ * unique keys, per-key counts and start offsets into the (sorted) values
+ * array.  D2H accessors below. */
+int dprk_group_by_key(dprk_ctx *ctx, uint64_t *n_groups_out);
+int dprk_get_groups(dprk_ctx *ctx, int64_t *unique_keys, uint64_t *starts,
+                    uint64_t *counts, uint64_t cap, uint64_t *n_out);
+
+/* Inner join of ctx_a's and ctx_b's datasets (same device): sorts both,
+ * then per-key cross product (dpark join semantics — SURVEY.md §3.3).
+ * Result (k, va, vb) is held by ctx_a; fetch with dprk_get_join. */
+int dprk_join(dprk_ctx *ctx_a, dprk_ctx *ctx_b, uint64_t *n_out);
+int dprk_get_join(dprk_ctx *ctx, int64_t *keys, double *va, double *vb,
+                  uint64_t cap, uint64_t *n_out);
+
+/* ---- multi-GPU exchange (RCCL over xGMI) ------------------------------ */
+/* 128-byte opaque ncclUniqueId; rank 0 creates it, host control plane
+ * distributes it (e.g. torch.distributed gloo broadcast — plumbing only) */
+int dprk_comm_unique_id(char uid[128]);
+int dprk_comm_init(dprk_ctx *ctx, int nranks, int rank, const char uid[128]);
+/* All-to-allv of the current dataset, whose rank segments must have been
+ * laid out by dprk_partition(ctx, nranks): one grouped ncclSend/ncclRecv
+ * per peer (pairwise-direct over xGMI links — SURVEY.md §5).  send_counts/
+ * recv_counts are host arrays of nranks entries (recv side from the
+ * host-allgathered count matrix).  Replaces the current dataset with the
+ * received one. */
+int dprk_exchange(dprk_ctx *ctx, const uint64_t *send_counts,
+                  const uint64_t *recv_counts, int nranks);
+
+/* ---- measurement ------------------------------------------------------ */
+typedef struct {
+  /* accumulated HIP-event time (ms) and launch counts per kernel class,
+   * plus algorithmic bytes attributed */
+  double ms_gen, ms_hist, ms_tilehist, ms_scan, ms_scatter, ms_reduce,
+      ms_runext, ms_join, ms_exchange, ms_other;
+  uint64_t n_gen, n_hist, n_tilehist, n_scan, n_scatter, n_reduce, n_runext,
+      n_join, n_exchange, n_other;
+  uint64_t bytes_scatter;      /* algorithmic bytes over scatter launches */
+  uint64_t bytes_total;        /* algorithmic bytes over the whole op mix */
+  uint64_t sort_passes;        /* digit passes actually executed (last sort) */
+} dprk_stats_t;
+
+int dprk_stats(dprk_ctx *ctx, dprk_stats_t *out);
+int dprk_stats_reset(dprk_ctx *ctx);
+
+#ifdef __cplusplus
+}
+#endif
+
+#endif /* DPRK_H */
