@@ -1,0 +1,238 @@
+"""GPU parity tests: the HIP engine vs the dpark-semantics oracle and the
+independent numpy/pandas second oracle, through the C-ABI (tier rule ③).
+Bit-exact for int64 keys/counts/order; <=1e-6 relative for float64 sums
+(the tolerance the north star states — deterministic fixed-tree GPU
+reduction vs sequential CPU order)."""
+
+import glob
+import os
+
+import numpy as np
+import pytest
+
+pytestmark = pytest.mark.gpu
+
+from dpark_amd.engine import Engine  # noqa: E402
+from oracle import second_oracle as so  # noqa: E402
+from oracle.synth import gen_pairs  # noqa: E402
+
+GOLDEN = glob.glob(os.path.join(os.path.dirname(__file__), "golden", "*.npz"))
+RTOL = 1e-6
+
+
+@pytest.fixture(scope="module")
+def eng():
+    e = Engine(device=0)
+    yield e
+    e.close()
+
+
+def _put(eng, keys, vals=None):
+    eng.reserve(max(len(keys), 1), with_vals=True)
+    if vals is None:
+        vals = np.zeros(len(keys), dtype=np.float64)
+    eng.put(keys, vals)
+
+
+# ---------------------------------------------------------------------------
+# golden fixtures (pinned against both CPU oracles at generation time)
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize("path", GOLDEN, ids=[os.path.basename(p) for p in GOLDEN])
+def test_golden_reduce(eng, path):
+    g = np.load(path)
+    _put(eng, g["keys"], g["vals"])
+    n = eng.reduce_by_key_sum()
+    assert n == len(g["rbk_keys"])
+    if n == 0:
+        return
+    k, v = eng.get()
+    assert np.array_equal(k, g["rbk_keys"])          # bit-exact int
+    np.testing.assert_allclose(v, g["rbk_sums"], rtol=RTOL)
+
+
+@pytest.mark.parametrize("path", GOLDEN, ids=[os.path.basename(p) for p in GOLDEN])
+def test_golden_sort(eng, path):
+    g = np.load(path)
+    _put(eng, g["keys"], g["vals"])
+    eng.sort()
+    if len(g["keys"]) == 0:
+        return
+    k, v = eng.get()
+    assert np.array_equal(k, g["sorted_keys"])       # bit-exact order
+    # values must be a valid permutation accompanying their keys:
+    # per-key value multisets must match (duplicate order unspecified)
+    sk, sv = so.sort_by_key(g["keys"], g["vals"])
+    assert np.array_equal(k, sk)
+    # compare multiset per key via double lexsort
+    got = np.lexsort((v, k))
+    want = np.lexsort((sv, sk))
+    np.testing.assert_allclose(v[got], sv[want], rtol=0)  # exact: permutation
+
+
+@pytest.mark.parametrize("path", GOLDEN, ids=[os.path.basename(p) for p in GOLDEN])
+def test_golden_group(eng, path):
+    g = np.load(path)
+    _put(eng, g["keys"], g["vals"])
+    uk, starts, counts = eng.group_by_key()
+    assert len(uk) == len(g["group_keys"])
+    if len(uk) == 0:
+        return
+    assert np.array_equal(uk, g["group_keys"])
+    assert np.array_equal(counts.astype(np.int64), g["group_counts"])
+    # starts must partition [0, n)
+    assert starts[0] == 0
+    assert np.all(starts[1:] == starts[:-1] + counts[:-1])
+
+
+@pytest.mark.parametrize("path", GOLDEN, ids=[os.path.basename(p) for p in GOLDEN])
+def test_golden_join(eng, path):
+    g = np.load(path)
+    nb = int(g["join_b_n"])
+    a = Engine(device=0)
+    b = Engine(device=0)
+    try:
+        a.reserve(max(len(g["keys"]), 1), with_vals=True)
+        a.put(g["keys"], g["vals"])
+        b.reserve(max(nb, 1), with_vals=True)
+        b.put(g["keys"][:nb], g["vals"][:nb])
+        jk, jva, jvb = a.join(b)
+        assert len(jk) == len(g["join_keys"])
+        if len(jk) == 0:
+            return
+        # canonical multiset order
+        order = np.lexsort((jvb, jva, jk))
+        assert np.array_equal(jk[order], g["join_keys"])
+        np.testing.assert_allclose(jva[order], g["join_va"], rtol=0)
+        np.testing.assert_allclose(jvb[order], g["join_vb"], rtol=0)
+    finally:
+        a.close()
+        b.close()
+
+
+# ---------------------------------------------------------------------------
+# device PRNG parity (byte-identical to oracle/synth.py)
+# ---------------------------------------------------------------------------
+
+def test_gen_pairs_matches_cpu_mirror(eng):
+    for seed, n, uni in [(42, 100_000, 1 << 24), (7, 65_536, 0),
+                         (11, 12_345, 1 << 8)]:
+        eng.reserve(n, with_vals=True)
+        eng.gen_pairs(seed=seed, n=n, key_universe=uni)
+        gk, gv = eng.get()
+        mk, mv = gen_pairs(seed=seed, n=n, key_universe=uni if uni else 1 << 64)
+        assert np.array_equal(gk, mk), f"seed={seed}"
+        assert np.array_equal(gv, mv), f"seed={seed}"  # bit-identical f64
+
+
+# ---------------------------------------------------------------------------
+# randomized medium-size parity vs the second oracle
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize("n,uni,seed", [
+    (1_000_000, 1 << 16, 1),     # heavy duplicates
+    (1_000_000, 1 << 20, 2),
+    (4_000_000, 1 << 24, 3),     # config-2 shape scaled down
+    (1_000_000, 0, 4),           # full 64-bit incl. negatives, ~all unique
+    (4_096, 1, 5),               # all keys equal (universe 1 => key 0)
+    (1, 1 << 24, 6),
+])
+def test_reduce_parity_random(eng, n, uni, seed):
+    keys, vals = gen_pairs(seed, n, uni if uni else 1 << 64)
+    _put(eng, keys, vals)
+    got_n = eng.reduce_by_key_sum()
+    k, v = eng.get()
+    wk, wv = so.reduce_by_key_sum(keys, vals)
+    assert got_n == len(wk)
+    assert np.array_equal(k, wk)
+    np.testing.assert_allclose(v, wv, rtol=RTOL)
+
+
+@pytest.mark.parametrize("n,uni,seed", [
+    (1_000_000, 0, 10),
+    (2_000_000, 1 << 12, 11),    # long duplicate runs
+    (333_333, 1 << 63, 12),      # only positive half
+])
+def test_sort_parity_random(eng, n, uni, seed):
+    keys, _ = gen_pairs(seed, n, uni if uni else 1 << 64)
+    _put(eng, keys)
+    eng.sort()
+    k, _ = eng.get()
+    assert np.array_equal(k, np.sort(keys))
+
+
+def test_sort_determinism(eng):
+    """same seed => bit-identical output (SURVEY.md §5 race-detection
+    analogue: determinism test)"""
+    keys, vals = gen_pairs(99, 2_000_000, 1 << 24)
+    _put(eng, keys, vals)
+    eng.reduce_by_key_sum()
+    k1, v1 = eng.get()
+    _put(eng, keys, vals)
+    eng.reduce_by_key_sum()
+    k2, v2 = eng.get()
+    assert np.array_equal(k1, k2)
+    assert np.array_equal(v1, v2)  # bit-identical f64: fixed reduction tree
+
+
+def test_join_parity_random(eng):
+    ka, va = gen_pairs(21, 300_000, 1 << 16)
+    kb, vb = gen_pairs(22, 200_000, 1 << 16)
+    a = Engine(device=0)
+    b = Engine(device=0)
+    try:
+        a.reserve(len(ka), with_vals=True)
+        a.put(ka, va)
+        b.reserve(len(kb), with_vals=True)
+        b.put(kb, vb)
+        jk, jva, jvb = a.join(b)
+        wk, wva, wvb = so.join_multiset(ka, va, kb, vb)
+        assert len(jk) == len(wk)
+        order = np.lexsort((jvb, jva, jk))
+        assert np.array_equal(jk[order], wk)
+        np.testing.assert_allclose(jva[order], wva, rtol=0)
+        np.testing.assert_allclose(jvb[order], wvb, rtol=0)
+    finally:
+        a.close()
+        b.close()
+
+
+# ---------------------------------------------------------------------------
+# partition + size-independent properties at larger sizes
+# ---------------------------------------------------------------------------
+
+def test_partition_counts_and_content(eng):
+    keys, vals = gen_pairs(31, 1_000_000, 1 << 24)
+    _put(eng, keys, vals)
+    counts = eng.partition(8)
+    assert counts.sum() == len(keys)
+    k, v = eng.get()
+    # partition is a permutation of the input
+    assert np.array_equal(np.sort(k), np.sort(keys))
+    # segments must agree with the device hash (splitmix64 top bits)
+    from oracle.synth import splitmix64
+    ranks = (splitmix64(k.view(np.uint64)) >> np.uint64(61)).astype(np.int64)
+    off = 0
+    for r, c in enumerate(counts):
+        assert np.all(ranks[off:off + int(c)] == r), f"bucket {r}"
+        off += int(c)
+
+
+def test_large_properties(eng):
+    """Size-independent properties at a size the oracle can't check row by
+    row: sortedness, multiset preservation via checksum, sum conservation."""
+    n = 50_000_000
+    eng.reserve(n, with_vals=True)
+    eng.gen_pairs(seed=1234, n=n, key_universe=1 << 24)
+    k0, v0 = eng.get()
+    key_checksum = np.bitwise_xor.reduce(k0.view(np.uint64))
+    val_sum = v0.sum()
+    n_unique_expected = len(np.unique(k0))
+    got = eng.reduce_by_key_sum()
+    k, v = eng.get()
+    assert got == n_unique_expected
+    assert np.all(k[1:] > k[:-1])                       # strictly sorted
+    assert np.bitwise_xor.reduce(k.view(np.uint64)) == \
+        np.bitwise_xor.reduce(np.unique(k0).view(np.uint64))
+    np.testing.assert_allclose(v.sum(), val_sum, rtol=1e-9)  # conservation
+    assert key_checksum == key_checksum  # keep var referenced
