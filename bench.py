@@ -154,6 +154,9 @@ def main():
         elapsed = float(t[0])
 
     s = eng.stats()
+    log("engine stats over timed region: " + json.dumps(
+        {k: round(v, 3) if isinstance(v, float) else v
+         for k, v in s.items()}))
     if rank != 0:
         return
 

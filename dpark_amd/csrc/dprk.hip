@@ -274,7 +274,10 @@ __global__ __launch_bounds__(NTHREADS) void k_minmax_or(
 
 /* per-tile 256-bin histogram; one workgroup per tile; per-wave LDS
  * replicas to cut same-digit ds_add serialization (SURVEY.md §7 hard
- * part 2) */
+ * part 2).  Histograms are stored TILE-MAJOR (hist[t*RADIX + d]) so each
+ * workgroup writes one contiguous 1 KiB block — digit-major layout made
+ * every 4 B store a partial-line write that each XCD's L2 evicted as a
+ * full 64 B line (measured 10x write amplification, profiles/r01). */
 __global__ __launch_bounds__(NTHREADS) void k_tile_hist(
     const int64_t *__restrict__ keys, uint64_t n,
     uint32_t *__restrict__ tile_hist, uint64_t ntiles, int mode, int shift,
@@ -293,37 +296,31 @@ __global__ __launch_bounds__(NTHREADS) void k_tile_hist(
     for (int d = threadIdx.x; d < RADIX; d += NTHREADS) {
       uint32_t s = 0;
       for (int w = 0; w < NTHREADS / 64; w++) s += h[w][d];
-      tile_hist[(uint64_t)d * ntiles + t] = s; /* digit-major for the scan */
+      tile_hist[t * RADIX + d] = s; /* tile-major: coalesced full lines */
     }
     __syncthreads();
   }
 }
 
-/* scan level 1: per-(digit, segment) sums. grid.x = nseg, grid.y = RADIX */
+/* scan level 1: per-(segment, digit) sums.  One WG per segment, thread =
+ * digit, sequential walk over the segment's tiles (coalesced 1 KiB loads;
+ * the 25 MB hist array is L2/L3-resident right after k_tile_hist). */
 __global__ __launch_bounds__(NTHREADS) void k_seg_sums(
-    const uint32_t *__restrict__ tile_hist, uint64_t ntiles, uint64_t nseg,
-    uint32_t *__restrict__ seg_sums) {
-  __shared__ uint32_t sh[NTHREADS];
-  uint64_t d = blockIdx.y, s = blockIdx.x;
-  uint64_t t0 = s * (uint64_t)SEG_TILES;
-  uint64_t t1 = min(t0 + SEG_TILES, ntiles);
+    const uint32_t *__restrict__ tile_hist, uint64_t ntiles,
+    uint64_t seg_tiles, uint32_t *__restrict__ seg_sums) {
+  uint64_t s = blockIdx.x;
+  uint64_t t0 = s * seg_tiles;
+  uint64_t t1 = min(t0 + seg_tiles, ntiles);
   uint32_t acc = 0;
-  for (uint64_t t = t0 + threadIdx.x; t < t1; t += NTHREADS)
-    acc += tile_hist[d * ntiles + t];
-  sh[threadIdx.x] = acc;
-  __syncthreads();
-  for (int off = NTHREADS / 2; off; off >>= 1) {
-    if (threadIdx.x < off) sh[threadIdx.x] += sh[threadIdx.x + off];
-    __syncthreads();
-  }
-  if (threadIdx.x == 0) seg_sums[d * nseg + s] = sh[0];
+#pragma unroll 4
+  for (uint64_t t = t0; t < t1; t++) acc += tile_hist[t * RADIX + threadIdx.x];
+  seg_sums[s * RADIX + threadIdx.x] = acc;
 }
 
-/* scan level 2 (single workgroup): exclusive-scan each digit row of
- * seg_sums in place; compute digit base offsets (exclusive over digit
- * rows' totals) and fold them in; export per-digit totals (u64) for the
- * host/partition path.  nseg <= 256 and RADIX == NTHREADS == 256: thread
- * d owns digit row d. */
+/* scan level 2 (single workgroup): per-digit exclusive scan over segment
+ * sums (nseg <= 256), fold in the cross-digit exclusive base, export
+ * per-digit totals (u64) for the partition path.  Thread = digit;
+ * all accesses coalesced in the seg-major layout. */
 __global__ __launch_bounds__(NTHREADS) void k_scan_mid(
     uint32_t *__restrict__ seg_sums, uint64_t nseg,
     uint64_t *__restrict__ row_tot) {
@@ -331,14 +328,13 @@ __global__ __launch_bounds__(NTHREADS) void k_scan_mid(
   int d = threadIdx.x;
   uint64_t run = 0;
   for (uint64_t s = 0; s < nseg; s++) {
-    uint32_t v = seg_sums[(uint64_t)d * nseg + s];
-    seg_sums[(uint64_t)d * nseg + s] = (uint32_t)run;
+    uint32_t v = seg_sums[s * RADIX + d];
+    seg_sums[s * RADIX + d] = (uint32_t)run;
     run += v;
   }
   tot[d] = run;
   if (row_tot) row_tot[d] = run;
   __syncthreads();
-  /* exclusive scan of tot[256] -> base, Hillis-Steele in LDS */
   __shared__ uint64_t sc[RADIX];
   sc[d] = tot[d];
   __syncthreads();
@@ -350,72 +346,71 @@ __global__ __launch_bounds__(NTHREADS) void k_scan_mid(
   }
   uint64_t base = sc[d] - tot[d]; /* inclusive -> exclusive */
   for (uint64_t s = 0; s < nseg; s++)
-    seg_sums[(uint64_t)d * nseg + s] += (uint32_t)base;
+    seg_sums[s * RADIX + d] += (uint32_t)base;
   /* NOTE: offsets fit u32 because n < 2^32 (enforced in dprk_reserve) */
 }
 
-/* scan level 3: within each (digit, segment) produce exclusive tile
- * offsets (+ segment base).  grid.x = nseg, grid.y = RADIX. */
+/* scan level 3: per-(segment) exclusive tile offsets.  One WG per
+ * segment, thread = digit, sequential over tiles; loads and stores both
+ * coalesced (tile-major). */
 __global__ __launch_bounds__(NTHREADS) void k_tile_off(
-    const uint32_t *__restrict__ tile_hist, uint64_t ntiles, uint64_t nseg,
-    const uint32_t *__restrict__ seg_sums, uint32_t *__restrict__ tile_off) {
-  __shared__ uint32_t sh[NTHREADS];
-  uint64_t d = blockIdx.y, s = blockIdx.x;
-  uint64_t t0 = s * (uint64_t)SEG_TILES;
-  uint64_t t1 = min(t0 + SEG_TILES, ntiles);
-  uint32_t seg_base = seg_sums[d * nseg + s];
-  /* chunks of NTHREADS tiles: thread-per-tile load, LDS scan, running base */
-  uint32_t run = seg_base;
-  for (uint64_t c = t0; c < t1; c += NTHREADS) {
-    uint64_t t = c + threadIdx.x;
-    uint32_t v = (t < t1) ? tile_hist[d * ntiles + t] : 0;
-    sh[threadIdx.x] = v;
-    __syncthreads();
-    /* inclusive scan */
-    for (int off = 1; off < NTHREADS; off <<= 1) {
-      uint32_t add = (threadIdx.x >= (unsigned)off) ? sh[threadIdx.x - off] : 0;
-      __syncthreads();
-      sh[threadIdx.x] += add;
-      __syncthreads();
-    }
-    if (t < t1) tile_off[d * ntiles + t] = run + sh[threadIdx.x] - v;
-    uint32_t chunk_tot = sh[NTHREADS - 1];
-    __syncthreads();
-    run += chunk_tot;
+    const uint32_t *__restrict__ tile_hist, uint64_t ntiles,
+    uint64_t seg_tiles, const uint32_t *__restrict__ seg_sums,
+    uint32_t *__restrict__ tile_off) {
+  uint64_t s = blockIdx.x;
+  uint64_t t0 = s * seg_tiles;
+  uint64_t t1 = min(t0 + seg_tiles, ntiles);
+  uint32_t run = seg_sums[s * RADIX + threadIdx.x];
+#pragma unroll 2
+  for (uint64_t t = t0; t < t1; t++) {
+    uint32_t v = tile_hist[t * RADIX + threadIdx.x];
+    tile_off[t * RADIX + threadIdx.x] = run;
+    run += v;
   }
 }
 
 /* the scatter: stable rank within tile + LDS exchange + coalesced write.
- * One workgroup per tile (grid-stride), NTHREADS threads, TILE elements.
- * LDS: keys 32 KiB + vals 32 KiB + control ~6 KiB => 2 blocks/CU. */
+ *
+ * Wave-private structure (v2 — v1 paid ~80 barriers/tile and 64 KiB of
+ * LDS counter zeroing per tile, measured 2.0 TB/s; see git history):
+ * each of the 4 waves owns a contiguous 1024-element quarter of the
+ * tile, holds its 16 elements/lane in REGISTERS, builds a private LDS
+ * histogram, and ranks with 8-ballot same-digit masks against private
+ * running counters — no cross-wave traffic between two barriers.
+ * Digit-sorted staging in LDS, then a sequential sweep writes
+ * equal-digit runs to contiguous global addresses. */
+template <bool WITH_VALS>
 __global__ __launch_bounds__(NTHREADS, 2) void k_scatter(
     const int64_t *__restrict__ keys_in, const double *__restrict__ vals_in,
     int64_t *__restrict__ keys_out, double *__restrict__ vals_out,
     uint64_t n, const uint32_t *__restrict__ tile_hist,
     const uint32_t *__restrict__ tile_off, uint64_t ntiles, int mode,
     int shift, uint32_t mask) {
+  constexpr int NW = NTHREADS / 64;        /* waves per WG                */
+  constexpr int WBLK = TILE / NW;          /* 1024 elements per wave      */
+  constexpr int EPL = WBLK / 64;           /* 16 elements per lane        */
   __shared__ int64_t lk[TILE];
-  __shared__ double lv[TILE];
-  __shared__ uint32_t excl[RADIX];      /* tile-local digit exclusive scan */
-  __shared__ uint32_t goff[RADIX];      /* global write base per digit     */
-  __shared__ uint32_t wcnt[NTHREADS / 64][RADIX];
-  __shared__ uint32_t drun[RADIX];      /* running per-digit count         */
+  __shared__ double lv[WITH_VALS ? TILE : 1];
+  __shared__ uint32_t excl[RADIX];     /* tile-local digit excl scan  */
+  __shared__ uint32_t goff[RADIX];     /* global write base per digit */
+  __shared__ uint32_t h[NW][RADIX];    /* per-wave hist, then running */
+  __shared__ uint32_t wbase[NW][RADIX];
 
   const int wave = threadIdx.x / 64, lane = threadIdx.x % 64;
+  const uint64_t lowmask = ((uint64_t)1 << lane) - 1;
 
   for (uint64_t t = blockIdx.x; t < ntiles; t += gridDim.x) {
     const uint64_t start = t * (uint64_t)TILE;
     const uint64_t end = min(start + TILE, n);
     const uint32_t tile_n = (uint32_t)(end - start);
+    const uint32_t wbeg = wave * WBLK;
 
-    /* tile-local exclusive digit scan from the raw tile histogram */
+    /* phase 0: per-tile digit metadata (coalesced tile-major loads) */
     {
-      uint32_t v = tile_hist[(uint64_t)threadIdx.x * ntiles + t];
+      uint32_t v = tile_hist[t * RADIX + threadIdx.x];
       excl[threadIdx.x] = v;
-      goff[threadIdx.x] = tile_off[(uint64_t)threadIdx.x * ntiles + t];
-      drun[threadIdx.x] = 0;
+      goff[threadIdx.x] = tile_off[t * RADIX + threadIdx.x];
       __syncthreads();
-      /* inclusive scan then shift */
       for (int off = 1; off < RADIX; off <<= 1) {
         uint32_t add = (threadIdx.x >= (unsigned)off) ? excl[threadIdx.x - off] : 0;
         __syncthreads();
@@ -425,65 +420,70 @@ __global__ __launch_bounds__(NTHREADS, 2) void k_scatter(
       uint32_t inc = excl[threadIdx.x];
       __syncthreads();
       excl[threadIdx.x] = inc - v;
+      /* zero this wave's histogram row (4 entries per lane) */
+      for (int j = 0; j < RADIX / 64; j++) h[wave][lane * (RADIX / 64) + j] = 0;
       __syncthreads();
     }
 
-    /* rank + LDS scatter, NTHREADS elements per sub-iteration (order =
-     * element order => stable) */
-    for (uint32_t s = 0; s < tile_n; s += NTHREADS) {
-      uint32_t idx = s + threadIdx.x;
-      int64_t key = 0;
-      double val = 0.0;
-      uint32_t dig = 0;
-      bool active = idx < tile_n;
-      /* zero wave counters (4 per thread) */
-      for (int i = threadIdx.x; i < (NTHREADS / 64) * RADIX; i += NTHREADS)
-        ((uint32_t *)wcnt)[i] = 0;
-      __syncthreads();
-      uint32_t pos = 0;
-      if (active) {
-        key = keys_in[start + idx];
-        if (vals_in) val = vals_in[start + idx];
-        dig = d_digit(key, mode, shift, mask);
-        /* wave-level same-digit mask via 8 ballots */
-        uint64_t same = __ballot(1); /* active lanes */
+    /* phase 1: load into registers + wave-private histogram */
+    int64_t rk[EPL];
+    double rv[EPL];
+    uint32_t rd[EPL];
+#pragma unroll
+    for (int i = 0; i < EPL; i++) {
+      uint32_t idx = wbeg + i * 64 + lane;
+      if (idx < tile_n) {
+        rk[i] = keys_in[start + idx];
+        if (WITH_VALS) rv[i] = vals_in[start + idx];
+        rd[i] = d_digit(rk[i], mode, shift, mask);
+        atomicAdd(&h[wave][rd[i]], 1u);
+      }
+    }
+    __syncthreads();
+
+    /* phase 2: cross-wave digit bases; reset h rows to 0 as the running
+     * counters for phase 3.  Thread = digit. */
+    {
+      uint32_t d = threadIdx.x;
+      uint32_t b = excl[d];
+      for (int w = 0; w < NW; w++) {
+        wbase[w][d] = b;
+        b += h[w][d];
+        h[w][d] = 0;
+      }
+    }
+    __syncthreads();
+
+    /* phase 3: stable rank (wave-local, no barriers) + LDS scatter */
+#pragma unroll
+    for (int i = 0; i < EPL; i++) {
+      uint32_t idx = wbeg + i * 64 + lane;
+      if (idx < tile_n) {
+        uint32_t dig = rd[i];
+        uint64_t same = __ballot(1);
         for (int b = 0; b < RADIX_BITS; b++) {
           uint64_t bb = __ballot((dig >> b) & 1);
           same &= ((dig >> b) & 1) ? bb : ~bb;
         }
-        uint64_t lowmask = ((uint64_t)1 << lane) - 1;
         uint32_t rank_in_wave = (uint32_t)__popcll(same & lowmask);
-        if (rank_in_wave == 0)
-          wcnt[wave][dig] = (uint32_t)__popcll(same);
-        pos = rank_in_wave;
+        uint32_t runv = h[wave][dig];            /* broadcast LDS read  */
+        if (rank_in_wave == 0)                   /* leader bumps counter */
+          h[wave][dig] = runv + (uint32_t)__popcll(same);
+        uint32_t pos = wbase[wave][dig] + runv + rank_in_wave;
+        lk[pos] = rk[i];
+        if (WITH_VALS) lv[pos] = rv[i];
       }
-      __syncthreads();
-      if (active) {
-        uint32_t before = drun[dig];
-        for (int w = 0; w < wave; w++) before += wcnt[w][dig];
-        uint32_t p = excl[dig] + before + pos;
-        lk[p] = key;
-        if (vals_in) lv[p] = val;
-      }
-      __syncthreads();
-      /* fold this sub-iteration's counts into the running totals */
-      {
-        uint32_t d = threadIdx.x;
-        uint32_t add = 0;
-        for (int w = 0; w < NTHREADS / 64; w++) add += wcnt[w][d];
-        drun[d] += add;
-      }
-      __syncthreads();
     }
+    __syncthreads();
 
-    /* write out in digit-sorted tile order: runs of equal digit land in
-     * consecutive global addresses => coalesced segments */
+    /* phase 4: write out in digit-sorted tile order: equal-digit runs
+     * land in consecutive global addresses => coalesced segments */
     for (uint32_t idx = threadIdx.x; idx < tile_n; idx += NTHREADS) {
       int64_t key = lk[idx];
       uint32_t dig = d_digit(key, mode, shift, mask);
       uint32_t dst = goff[dig] + (idx - excl[dig]);
       keys_out[dst] = key;
-      if (vals_in) vals_out[dst] = lv[idx];
+      if (WITH_VALS) vals_out[dst] = lv[idx];
     }
     __syncthreads();
   }
@@ -499,13 +499,17 @@ __global__ __launch_bounds__(NTHREADS, 2) void k_scatter(
  * tile, compacted into scratch at tile*TILE; record per-tile trailing open
  * sum T, any-end flag and run count for the stitch. Input must be sorted.
  * Run ends are globally exact (next-key lookahead crosses tiles). */
+/* per-thread chunks are CONTIGUOUS 16-element runs, so LDS indices are
+ * padded (PIDX: +1 slot per 16) — the unpadded layout put lanes l and
+ * l+2 on the same bank (16-way conflict, measured 1.26 ms/launch). */
+#define PIDX(j) ((j) + ((j) >> 4))
 __global__ __launch_bounds__(NTHREADS, 2) void k_reduce_tiles(
     const int64_t *__restrict__ keys, const double *__restrict__ vals,
     uint64_t n, int64_t *__restrict__ keys_scr, double *__restrict__ sums_scr,
     double *__restrict__ tile_T, uint32_t *__restrict__ tile_anyend,
     uint32_t *__restrict__ tile_nruns, uint64_t ntiles) {
-  __shared__ int64_t lk[TILE + 1];   /* +1: next tile's first key        */
-  __shared__ double lv[TILE];
+  __shared__ int64_t lk[TILE + TILE / 16 + 1];   /* padded; +1: lookahead */
+  __shared__ double lv[TILE + TILE / 16];
   __shared__ double t_carry[NTHREADS];   /* per-thread trailing open sum */
   __shared__ uint32_t t_flag[NTHREADS];  /* thread chunk has a head      */
   __shared__ uint32_t t_ends[NTHREADS];  /* run ends in thread chunk     */
@@ -515,11 +519,11 @@ __global__ __launch_bounds__(NTHREADS, 2) void k_reduce_tiles(
     const uint64_t end = min(start + TILE, n);
     const uint32_t tile_n = (uint32_t)(end - start);
     for (uint32_t i = threadIdx.x; i < tile_n; i += NTHREADS) {
-      lk[i] = keys[start + i];
-      lv[i] = vals[start + i];
+      lk[PIDX(i)] = keys[start + i];
+      lv[PIDX(i)] = vals[start + i];
     }
     if (threadIdx.x == 0)
-      lk[tile_n] = (end < n) ? keys[end] : 0; /* lookahead (unused if end==n) */
+      lk[PIDX(tile_n)] = (end < n) ? keys[end] : 0; /* lookahead (unused if end==n) */
     __syncthreads();
 
     const uint32_t c0 = threadIdx.x * ELEMS_PER_THREAD;         /* chunk */
@@ -534,16 +538,16 @@ __global__ __launch_bounds__(NTHREADS, 2) void k_reduce_tiles(
       if (c0 == 0) {
         if (start == 0) { have_prev = false; }
         else { prev_key = keys[start - 1]; have_prev = true; }
-      } else { prev_key = lk[c0 - 1]; have_prev = true; }
+      } else { prev_key = lk[PIDX(c0 - 1)]; have_prev = true; }
       for (uint32_t i = c0; i < c1; i++) {
-        bool head = !have_prev || lk[i] != prev_key;
-        if (head) { open = lv[i]; has_head = 1; }
-        else open += lv[i];
+        bool head = !have_prev || lk[PIDX(i)] != prev_key;
+        if (head) { open = lv[PIDX(i)]; has_head = 1; }
+        else open += lv[PIDX(i)];
         bool is_end = (start + i == n - 1) ||
-                      ((i + 1 < tile_n) ? lk[i + 1] != lk[i]
-                                        : lk[tile_n] != lk[i]);
+                      ((i + 1 < tile_n) ? lk[PIDX(i + 1)] != lk[PIDX(i)]
+                                        : lk[PIDX(tile_n)] != lk[PIDX(i)]);
         if (is_end) n_ends++;
-        prev_key = lk[i];
+        prev_key = lk[PIDX(i)];
         have_prev = true;
       }
     }
@@ -588,27 +592,27 @@ __global__ __launch_bounds__(NTHREADS, 2) void k_reduce_tiles(
       bool have_prev2 = false;
       if (c0 == 0) {
         if (start != 0) { prev_key2 = keys[start - 1]; have_prev2 = true; }
-      } else { prev_key2 = lk[c0 - 1]; have_prev2 = true; }
+      } else { prev_key2 = lk[PIDX(c0 - 1)]; have_prev2 = true; }
       uint32_t w = ends_excl;
       for (uint32_t i = c0; i < c1; i++) {
-        bool head = !have_prev2 || lk[i] != prev_key2;
-        if (head) running = lv[i];
-        else running += lv[i];
+        bool head = !have_prev2 || lk[PIDX(i)] != prev_key2;
+        if (head) running = lv[PIDX(i)];
+        else running += lv[PIDX(i)];
         bool is_end = (start + i == n - 1) ||
-                      ((i + 1 < tile_n) ? lk[i + 1] != lk[i]
-                                        : lk[tile_n] != lk[i]);
+                      ((i + 1 < tile_n) ? lk[PIDX(i + 1)] != lk[PIDX(i)]
+                                        : lk[PIDX(tile_n)] != lk[PIDX(i)]);
         if (is_end) {
-          keys_scr[start + w] = lk[i];
+          keys_scr[start + w] = lk[PIDX(i)];
           sums_scr[start + w] = running;
           w++;
         }
-        prev_key2 = lk[i];
+        prev_key2 = lk[PIDX(i)];
         have_prev2 = true;
       }
       /* last thread with data records the tile trailing state */
       if (c1 == tile_n) {
         bool last_is_end = (start + tile_n - 1 == n - 1) ||
-                           (lk[tile_n] != lk[tile_n - 1]);
+                           (lk[PIDX(tile_n)] != lk[PIDX(tile_n - 1)]);
         tile_T[t] = last_is_end ? 0.0 : running;
         /* any_end for the tile = total ends > 0 */
         tile_anyend[t] = (t_ends[NTHREADS - 1] > 0) ? 1u : 0u;
@@ -1129,7 +1133,8 @@ static int radix_pass(dprk_ctx *c, int mode, int shift, uint32_t mask,
                       int want_row_tot) {
   uint64_t n = c->n;
   uint64_t ntiles = ceil_div(n, TILE);
-  uint64_t nseg = ceil_div(ntiles, SEG_TILES);
+  uint64_t seg_tiles = ceil_div(ntiles, RADIX); /* nseg <= 256 */
+  uint64_t nseg = ceil_div(ntiles, seg_tiles);
   int src = c->cur, dst = c->cur ^ 1;
   int thgrid = (int)min(ntiles, (uint64_t)2048);
   TIMED(c, K_TILEHIST,
@@ -1137,24 +1142,30 @@ static int radix_pass(dprk_ctx *c, int mode, int shift, uint32_t mask,
                            c->stream, c->keys[src], n, c->tile_hist, ntiles,
                            mode, shift, mask));
   TIMED(c, K_SCAN,
-        hipLaunchKernelGGL(k_seg_sums, dim3((uint32_t)nseg, RADIX),
-                           dim3(NTHREADS), 0, c->stream, c->tile_hist,
-                           ntiles, nseg, c->seg_sums));
+        hipLaunchKernelGGL(k_seg_sums, dim3((uint32_t)nseg), dim3(NTHREADS),
+                           0, c->stream, c->tile_hist, ntiles, seg_tiles,
+                           c->seg_sums));
   TIMED(c, K_SCAN,
         hipLaunchKernelGGL(k_scan_mid, dim3(1), dim3(NTHREADS), 0, c->stream,
                            c->seg_sums, nseg,
                            want_row_tot ? c->row_tot : nullptr));
   TIMED(c, K_SCAN,
-        hipLaunchKernelGGL(k_tile_off, dim3((uint32_t)nseg, RADIX),
-                           dim3(NTHREADS), 0, c->stream, c->tile_hist,
-                           ntiles, nseg, c->seg_sums, c->tile_off));
-  TIMED(c, K_SCATTER,
-        hipLaunchKernelGGL(k_scatter, dim3(thgrid), dim3(NTHREADS), 0,
-                           c->stream, c->keys[src],
-                           c->with_vals ? c->vals[src] : nullptr,
-                           c->keys[dst], c->with_vals ? c->vals[dst] : nullptr,
-                           n, c->tile_hist, c->tile_off, ntiles, mode, shift,
-                           mask));
+        hipLaunchKernelGGL(k_tile_off, dim3((uint32_t)nseg), dim3(NTHREADS),
+                           0, c->stream, c->tile_hist, ntiles, seg_tiles,
+                           c->seg_sums, c->tile_off));
+  if (c->with_vals) {
+    TIMED(c, K_SCATTER,
+          hipLaunchKernelGGL(k_scatter<true>, dim3(thgrid), dim3(NTHREADS),
+                             0, c->stream, c->keys[src], c->vals[src],
+                             c->keys[dst], c->vals[dst], n, c->tile_hist,
+                             c->tile_off, ntiles, mode, shift, mask));
+  } else {
+    TIMED(c, K_SCATTER,
+          hipLaunchKernelGGL(k_scatter<false>, dim3(thgrid), dim3(NTHREADS),
+                             0, c->stream, c->keys[src], nullptr,
+                             c->keys[dst], nullptr, n, c->tile_hist,
+                             c->tile_off, ntiles, mode, shift, mask));
+  }
   c->stats.bytes_scatter += n * (c->with_vals ? 32 : 16);
   c->stats.bytes_total += n * (c->with_vals ? 40 : 24); /* +hist read */
   c->cur = dst;

@@ -1,0 +1,75 @@
+"""Multi-rank shuffle plan — the host-side half of the exchange
+(SURVEY.md §8(e)).  The data path is dprk_partition + dprk_exchange (RCCL
+all-to-allv inside the engine); this module holds the pure-plan logic
+(rank function mirror, count matrix -> recv counts) so the N>1 semantics
+are testable on CPU with gloo (tests/test_dist_cpu.py), world_size > 1,
+no GPU.
+"""
+
+from __future__ import annotations
+
+import numpy as np
+
+
+def rank_of_keys(keys, nranks: int):
+    """Numpy mirror of the engine's rank digit: TOP log2(nranks) bits of
+    splitmix64(key) (d_digit mode 1 in dpark_amd/csrc/dprk.hip).  The GPU
+    HashPartitioner — layout intentionally differs from reference
+    portable_hash."""
+    from oracle.synth import splitmix64  # test-infra import is fine here:
+    # the splitmix64 SPEC is normative and shared; the engine never calls
+    # this function (it has its own device implementation).
+    assert nranks >= 1 and (nranks & (nranks - 1)) == 0
+    if nranks == 1:
+        return np.zeros(len(keys), dtype=np.int64)
+    bits = nranks.bit_length() - 1
+    mix = splitmix64(np.asarray(keys, dtype=np.int64).view(np.uint64))
+    return (mix >> np.uint64(64 - bits)).astype(np.int64)
+
+
+def recv_counts(count_matrix, rank: int):
+    """count_matrix[r] = rank r's send counts (length nranks).  Rank r
+    receives column r."""
+    return [int(count_matrix[src][rank]) for src in range(len(count_matrix))]
+
+
+def exchange_with_torch_dist(dist, keys, vals, nranks, rank):
+    """CPU/gloo reference exchange used by the world_size>1 CPU tests:
+    partition locally by rank_of_keys, all-to-allv via isend/irecv pairs
+    (gloo has no all_to_all).  Returns the received (keys, vals)."""
+    import torch
+    r = rank_of_keys(keys, nranks)
+    order = np.argsort(r, kind="stable")
+    keys, vals = keys[order], vals[order]
+    send = np.bincount(r, minlength=nranks)
+    cm = [torch.zeros(nranks, dtype=torch.int64) for _ in range(nranks)]
+    dist.all_gather(cm, torch.tensor(send, dtype=torch.int64))
+    recv = recv_counts(cm, rank)
+    offs = np.concatenate([[0], np.cumsum(send)]).astype(np.int64)
+    out_k, out_v = [], []
+    reqs = []
+    send_bufs = []
+    for peer in range(nranks):
+        kb = torch.tensor(keys[offs[peer]:offs[peer + 1]], dtype=torch.int64)
+        vb = torch.tensor(vals[offs[peer]:offs[peer + 1]],
+                          dtype=torch.float64)
+        send_bufs.append((kb, vb))
+    recv_bufs = [(torch.zeros(recv[p], dtype=torch.int64),
+                  torch.zeros(recv[p], dtype=torch.float64))
+                 for p in range(nranks)]
+    for peer in range(nranks):
+        if peer == rank:
+            continue
+        if len(send_bufs[peer][0]):
+            reqs.append(dist.isend(send_bufs[peer][0], dst=peer, tag=0))
+            reqs.append(dist.isend(send_bufs[peer][1], dst=peer, tag=1))
+        if recv[peer]:
+            reqs.append(dist.irecv(recv_bufs[peer][0], src=peer, tag=0))
+            reqs.append(dist.irecv(recv_bufs[peer][1], src=peer, tag=1))
+    for q in reqs:
+        q.wait()
+    recv_bufs[rank] = send_bufs[rank]
+    for p in range(nranks):
+        out_k.append(recv_bufs[p][0].numpy())
+        out_v.append(recv_bufs[p][1].numpy())
+    return np.concatenate(out_k), np.concatenate(out_v)

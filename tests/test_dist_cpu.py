@@ -1,0 +1,103 @@
+"""Multi-process (world_size 2) CPU tests of the exchange-plan semantics:
+gloo backend, numpy payloads — the N>1 path minus the GPU kernels
+(gpurun has one GPU, so this is where the distributed
+logic is kept correct-by-construction)."""
+
+import os
+import sys
+
+import numpy as np
+import pytest
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+from dpark_amd.distshuffle import rank_of_keys, recv_counts  # noqa: E402
+
+
+def test_rank_function_properties():
+    from oracle.synth import gen_pairs
+    keys, _ = gen_pairs(3, 100_000, 1 << 24)
+    for nranks in (1, 2, 4, 8):
+        r = rank_of_keys(keys, nranks)
+        assert r.min() >= 0 and r.max() < nranks
+        # deterministic
+        assert np.array_equal(r, rank_of_keys(keys, nranks))
+        # same key -> same rank (the property reduceByKey correctness
+        # rests on: all copies of a key land on one rank)
+        uk, inv = np.unique(keys, return_inverse=True)
+        per_key = rank_of_keys(uk, nranks)
+        assert np.array_equal(r, per_key[inv])
+        # roughly balanced for uniform keys
+        if nranks > 1:
+            counts = np.bincount(r, minlength=nranks)
+            assert counts.max() / counts.min() < 1.05
+
+
+def test_recv_counts_matrix():
+    cm = [[5, 3], [2, 7]]
+    assert recv_counts(cm, 0) == [5, 2]
+    assert recv_counts(cm, 1) == [3, 7]
+
+
+def _dist_worker(rank, world, port, fail_q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        os.environ["RANK"] = str(rank)
+        os.environ["WORLD_SIZE"] = str(world)
+        import torch.distributed as dist
+        from dpark_amd.distshuffle import exchange_with_torch_dist
+        from oracle.synth import gen_pairs
+        from oracle import second_oracle as so
+        import torch
+
+        dist.init_process_group(backend="gloo", rank=rank, world_size=world)
+        n_local = 50_000
+        keys, vals = gen_pairs(seed=100 + rank, n=n_local,
+                               key_universe=1 << 12)
+        rk, rv = exchange_with_torch_dist(dist, keys, vals, world, rank)
+        # every received key must belong to this rank
+        assert np.all(rank_of_keys(rk, world) == rank)
+        # local reduce of received shard
+        uk, sums = so.reduce_by_key_sum(rk, rv)
+        # gather all shards' results and compare with the global oracle
+        obj = [None] * world
+        dist.all_gather_object(obj, (uk, sums))
+        if rank == 0:
+            gk = np.concatenate([o[0] for o in obj])
+            gs = np.concatenate([o[1] for o in obj])
+            order = np.argsort(gk)
+            gk, gs = gk[order], gs[order]
+            all_k = np.concatenate([gen_pairs(100 + r, n_local, 1 << 12)[0]
+                                    for r in range(world)])
+            all_v = np.concatenate([gen_pairs(100 + r, n_local, 1 << 12)[1]
+                                    for r in range(world)])
+            wk, ws = so.reduce_by_key_sum(all_k, all_v)
+            assert np.array_equal(gk, wk)
+            np.testing.assert_allclose(gs, ws, rtol=1e-9)
+        dist.barrier()
+        dist.destroy_process_group()
+    except Exception as e:  # pragma: no cover
+        fail_q.put(f"rank {rank}: {type(e).__name__}: {e}")
+        raise
+
+
+@pytest.mark.timeout(180)
+def test_world2_exchange_reduce_matches_oracle():
+    torch = pytest.importorskip("torch")
+    assert torch  # noqa
+    import multiprocessing as mp
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29511
+    procs = [ctx.Process(target=_dist_worker, args=(r, 2, port, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=170)
+    errs = []
+    while not q.empty():
+        errs.append(q.get())
+    assert not errs, errs
+    assert all(p.exitcode == 0 for p in procs), [p.exitcode for p in procs]
