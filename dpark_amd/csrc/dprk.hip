@@ -103,6 +103,11 @@ struct dprk_ctx {
   uint64_t cap = 0;
   int with_vals = 0;
   int sorted = 0;
+  /* conservative OR/AND of biased keys known from generation (digit-skip
+   * without the k_minmax_or pass); valid only straight after gen /
+   * permutation-only ops */
+  int orand_valid = 0;
+  uint64_t known_or = 0, known_and = 0;
 
   /* radix scratch */
   uint32_t *tile_hist = nullptr;  /* [RADIX][ntiles] raw                 */
@@ -1090,6 +1095,16 @@ extern "C" int dprk_gen_pairs(dprk_ctx *c, uint64_t seed, uint64_t n,
                            seed << 32, keymask));
   c->n = n;
   c->sorted = 0;
+  /* keys = splitmix64 & keymask, so biased keys lie in
+   * [SIGNFLIP, SIGNFLIP | keymask]: conservative OR/AND bound — may only
+   * RUN an avoidable pass, never skip a needed one */
+  if (keymask != ~0ull) {
+    c->orand_valid = 1;
+    c->known_or = SIGNFLIP | keymask;
+    c->known_and = SIGNFLIP;
+  } else {
+    c->orand_valid = 0;
+  }
   return drain_timing(c);
 }
 
@@ -1107,6 +1122,7 @@ extern "C" int dprk_put_pairs(dprk_ctx *c, const int64_t *keys,
   HIP_CHECK(hipStreamSynchronize(c->stream));
   c->n = n;
   c->sorted = 0;
+  c->orand_valid = 0;
   return 0;
 }
 
@@ -1177,19 +1193,25 @@ extern "C" int dprk_sort_pairs(dprk_ctx *c) {
   HIP_CHECK(hipSetDevice(c->device));
   if (c->n == 0) { c->sorted = 1; return 0; }
   if (c->sorted) return 0;
-  /* digit-skip detection: OR/AND of biased keys */
-  uint64_t init[2] = {0, ~0ull};
-  HIP_CHECK(hipMemcpyAsync(c->d_orand, init, 16, hipMemcpyHostToDevice,
-                           c->stream));
-  TIMED(c, K_HIST,
-        hipLaunchKernelGGL(k_minmax_or, dim3(stream_grid(c->n)),
-                           dim3(NTHREADS), 0, c->stream, c->keys[c->cur],
-                           c->n, c->d_orand));
+  /* digit-skip detection: OR/AND of biased keys (or the conservative
+   * generation-time bound, saving one full read of the keys) */
   uint64_t orand[2];
-  HIP_CHECK(hipMemcpyAsync(orand, c->d_orand, 16, hipMemcpyDeviceToHost,
-                           c->stream));
-  HIP_CHECK(hipStreamSynchronize(c->stream));
-  c->stats.bytes_total += c->n * 8;
+  if (c->orand_valid) {
+    orand[0] = c->known_or;
+    orand[1] = c->known_and;
+  } else {
+    uint64_t init[2] = {0, ~0ull};
+    HIP_CHECK(hipMemcpyAsync(c->d_orand, init, 16, hipMemcpyHostToDevice,
+                             c->stream));
+    TIMED(c, K_HIST,
+          hipLaunchKernelGGL(k_minmax_or, dim3(stream_grid(c->n)),
+                             dim3(NTHREADS), 0, c->stream, c->keys[c->cur],
+                             c->n, c->d_orand));
+    HIP_CHECK(hipMemcpyAsync(orand, c->d_orand, 16, hipMemcpyDeviceToHost,
+                             c->stream));
+    HIP_CHECK(hipStreamSynchronize(c->stream));
+    c->stats.bytes_total += c->n * 8;
+  }
   c->stats.sort_passes = 0;
   for (int p = 0; p < 8; p++) {
     uint32_t ob = (uint32_t)(orand[0] >> (8 * p)) & 0xFF;
@@ -1503,6 +1525,7 @@ extern "C" int dprk_exchange(dprk_ctx *c, const uint64_t *send_counts,
   c->cur = dst;
   c->n = total_recv;
   c->sorted = 0;
+  c->orand_valid = 0;
   return drain_timing(c);
 }
 
