@@ -58,6 +58,13 @@ class DparkContext:
     def textFile(self, path, numSplits=None):
         return TextFileRDD(self, path, numSplits)
 
+    def vectorPairs(self, keys, vals, numSlices=None):
+        """Engine-eligible pair RDD (numpy int64 keys / float64 vals):
+        the hot-path operators route to the HIP engine on a GPU machine
+        (dpark_amd/vector.py)."""
+        from .vector import VectorPairsRDD
+        return VectorPairsRDD(self, keys, vals, numSlices)
+
     def union(self, rdds):
         return UnionRDD(self, list(rdds))
 

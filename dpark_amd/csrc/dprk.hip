@@ -34,7 +34,10 @@
 #define RADIX_BITS 8
 #define TILE 4096          /* elements per scatter/reduce workgroup      */
 #define NTHREADS 256       /* 4 waves of 64                              */
-#define SEG_TILES 4096     /* tiles per mid-level scan segment           */
+#define NTHREADS_SC 512    /* scatter workgroup: 8 waves                 */
+#define RTILE 2048         /* elements per reduce/head workgroup (4 WG/CU
+                              occupancy: the reduce is latency-bound)    */
+#define EPT_R (RTILE / NTHREADS) /* 8 consecutive elements per thread    */
 #define ELEMS_PER_THREAD (TILE / NTHREADS) /* 16 */
 #define SCAN_CHUNK 16384   /* generic u64 scan: elements per workgroup   */
 #define MAX_EVTS 16384
@@ -232,15 +235,30 @@ __device__ __forceinline__ uint32_t d_digit(int64_t key, int mode, int shift,
 __global__ __launch_bounds__(NTHREADS) void k_gen(
     int64_t *__restrict__ keys, double *__restrict__ vals, uint64_t n,
     uint64_t seed_hi, uint64_t keymask) {
-  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
-  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += stride) {
-    uint64_t x = seed_hi + i;
-    uint64_t k = d_splitmix64(x) & keymask;
-    keys[i] = (int64_t)k;
+  uint64_t stride = 2 * (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = 2 * ((uint64_t)blockIdx.x * blockDim.x + threadIdx.x);
+       i < n; i += stride) {
+    uint64_t x0 = seed_hi + i, x1 = seed_hi + i + 1;
+    int64_t k0 = (int64_t)(d_splitmix64(x0) & keymask);
+    if (i + 1 < n) { /* 16-B vectorized stores */
+      longlong2 kk;
+      kk.x = k0;
+      kk.y = (int64_t)(d_splitmix64(x1) & keymask);
+      *reinterpret_cast<longlong2 *>(&keys[i]) = kk;
+    } else {
+      keys[i] = k0;
+    }
     if (vals) {
-      uint64_t vb = d_splitmix64(x ^ 0xDEADBEEFCAFEBABEull);
-      vals[i] = (double)(vb >> 11) * (1.0 / 9007199254740992.0); /* 2^-53 */
+      uint64_t v0 = d_splitmix64(x0 ^ 0xDEADBEEFCAFEBABEull);
+      if (i + 1 < n) {
+        uint64_t v1 = d_splitmix64(x1 ^ 0xDEADBEEFCAFEBABEull);
+        double2 vv;
+        vv.x = (double)(v0 >> 11) * (1.0 / 9007199254740992.0);
+        vv.y = (double)(v1 >> 11) * (1.0 / 9007199254740992.0);
+        *reinterpret_cast<double2 *>(&vals[i]) = vv;
+      } else {
+        vals[i] = (double)(v0 >> 11) * (1.0 / 9007199254740992.0);
+      }
     }
   }
 }
@@ -295,8 +313,15 @@ __global__ __launch_bounds__(NTHREADS) void k_tile_hist(
     __syncthreads();
     uint64_t start = t * (uint64_t)TILE;
     uint64_t end = min(start + TILE, n);
-    for (uint64_t i = start + threadIdx.x; i < end; i += NTHREADS)
-      atomicAdd(&h[wave][d_digit(keys[i], mode, shift, mask)], 1u);
+    /* 16-B loads: two keys per lane (8-B streams run at 0.54-0.70x the
+     * 16-B rate on gfx950 — MI355X_MICROARCH.md §visibility table) */
+    for (uint64_t i = start + 2 * threadIdx.x; i + 1 < end; i += 2 * NTHREADS) {
+      longlong2 kk = *reinterpret_cast<const longlong2 *>(&keys[i]);
+      atomicAdd(&h[wave][d_digit(kk.x, mode, shift, mask)], 1u);
+      atomicAdd(&h[wave][d_digit(kk.y, mode, shift, mask)], 1u);
+    }
+    if (end == n && (n & 1) && threadIdx.x == 0 && end - start > 0)
+      atomicAdd(&h[wave][d_digit(keys[end - 1], mode, shift, mask)], 1u);
     __syncthreads();
     for (int d = threadIdx.x; d < RADIX; d += NTHREADS) {
       uint32_t s = 0;
@@ -385,21 +410,21 @@ __global__ __launch_bounds__(NTHREADS) void k_tile_off(
  * Digit-sorted staging in LDS, then a sequential sweep writes
  * equal-digit runs to contiguous global addresses. */
 template <bool WITH_VALS>
-__global__ __launch_bounds__(NTHREADS, 2) void k_scatter(
+__global__ __launch_bounds__(NTHREADS_SC, 2) void k_scatter(
     const int64_t *__restrict__ keys_in, const double *__restrict__ vals_in,
     int64_t *__restrict__ keys_out, double *__restrict__ vals_out,
     uint64_t n, const uint32_t *__restrict__ tile_hist,
     const uint32_t *__restrict__ tile_off, uint64_t ntiles, int mode,
     int shift, uint32_t mask) {
-  constexpr int NW = NTHREADS / 64;        /* waves per WG                */
-  constexpr int WBLK = TILE / NW;          /* 1024 elements per wave      */
-  constexpr int EPL = WBLK / 64;           /* 16 elements per lane        */
+  constexpr int NW = NTHREADS_SC / 64;     /* 8 waves per WG              */
+  constexpr int WBLK = TILE / NW;          /* 512 elements per wave       */
+  constexpr int EPL = WBLK / 64;           /* 8 elements per lane         */
   __shared__ int64_t lk[TILE];
   __shared__ double lv[WITH_VALS ? TILE : 1];
   __shared__ uint32_t excl[RADIX];     /* tile-local digit excl scan  */
   __shared__ uint32_t goff[RADIX];     /* global write base per digit */
-  __shared__ uint32_t h[NW][RADIX];    /* per-wave hist, then running */
-  __shared__ uint32_t wbase[NW][RADIX];
+  __shared__ uint32_t h[NW][RADIX];    /* per-wave hist -> wave bases
+                                          -> running counters (in place) */
 
   const int wave = threadIdx.x / 64, lane = threadIdx.x % 64;
   const uint64_t lowmask = ((uint64_t)1 << lane) - 1;
@@ -410,25 +435,29 @@ __global__ __launch_bounds__(NTHREADS, 2) void k_scatter(
     const uint32_t tile_n = (uint32_t)(end - start);
     const uint32_t wbeg = wave * WBLK;
 
-    /* phase 0: per-tile digit metadata (coalesced tile-major loads) */
-    {
-      uint32_t v = tile_hist[t * RADIX + threadIdx.x];
-      excl[threadIdx.x] = v;
+    /* phase 0: per-tile digit metadata (coalesced tile-major loads);
+     * zero this wave's histogram row */
+    uint32_t myv = 0;
+    if (threadIdx.x < RADIX) {
+      myv = tile_hist[t * RADIX + threadIdx.x];
+      excl[threadIdx.x] = myv;
       goff[threadIdx.x] = tile_off[t * RADIX + threadIdx.x];
+    }
+    for (int j = 0; j < RADIX / 64; j++) h[wave][lane * (RADIX / 64) + j] = 0;
+    __syncthreads();
+    for (int off = 1; off < RADIX; off <<= 1) {
+      uint32_t add = 0;
+      if (threadIdx.x < RADIX && threadIdx.x >= (unsigned)off)
+        add = excl[threadIdx.x - off];
       __syncthreads();
-      for (int off = 1; off < RADIX; off <<= 1) {
-        uint32_t add = (threadIdx.x >= (unsigned)off) ? excl[threadIdx.x - off] : 0;
-        __syncthreads();
-        excl[threadIdx.x] += add;
-        __syncthreads();
-      }
-      uint32_t inc = excl[threadIdx.x];
-      __syncthreads();
-      excl[threadIdx.x] = inc - v;
-      /* zero this wave's histogram row (4 entries per lane) */
-      for (int j = 0; j < RADIX / 64; j++) h[wave][lane * (RADIX / 64) + j] = 0;
+      if (threadIdx.x < RADIX) excl[threadIdx.x] += add;
       __syncthreads();
     }
+    if (threadIdx.x < RADIX) {
+      uint32_t inc = excl[threadIdx.x];
+      excl[threadIdx.x] = inc - myv; /* own slot only: no race */
+    }
+    __syncthreads();
 
     /* phase 1: load into registers + wave-private histogram */
     int64_t rk[EPL];
@@ -446,15 +475,16 @@ __global__ __launch_bounds__(NTHREADS, 2) void k_scatter(
     }
     __syncthreads();
 
-    /* phase 2: cross-wave digit bases; reset h rows to 0 as the running
-     * counters for phase 3.  Thread = digit. */
-    {
+    /* phase 2: turn h into per-wave digit BASES in place (thread=digit):
+     * h[w][d] := excl[d] + sum_{w'<w} h[w'][d]; phase 3 then uses h as
+     * the running counters directly. */
+    if (threadIdx.x < RADIX) {
       uint32_t d = threadIdx.x;
       uint32_t b = excl[d];
       for (int w = 0; w < NW; w++) {
-        wbase[w][d] = b;
-        b += h[w][d];
-        h[w][d] = 0;
+        uint32_t cnt = h[w][d];
+        h[w][d] = b;
+        b += cnt;
       }
     }
     __syncthreads();
@@ -471,10 +501,10 @@ __global__ __launch_bounds__(NTHREADS, 2) void k_scatter(
           same &= ((dig >> b) & 1) ? bb : ~bb;
         }
         uint32_t rank_in_wave = (uint32_t)__popcll(same & lowmask);
-        uint32_t runv = h[wave][dig];            /* broadcast LDS read  */
-        if (rank_in_wave == 0)                   /* leader bumps counter */
-          h[wave][dig] = runv + (uint32_t)__popcll(same);
-        uint32_t pos = wbase[wave][dig] + runv + rank_in_wave;
+        uint32_t base = h[wave][dig];             /* broadcast LDS read  */
+        if (rank_in_wave == 0)                    /* leader advances     */
+          h[wave][dig] = base + (uint32_t)__popcll(same);
+        uint32_t pos = base + rank_in_wave;
         lk[pos] = rk[i];
         if (WITH_VALS) lv[pos] = rv[i];
       }
@@ -483,7 +513,7 @@ __global__ __launch_bounds__(NTHREADS, 2) void k_scatter(
 
     /* phase 4: write out in digit-sorted tile order: equal-digit runs
      * land in consecutive global addresses => coalesced segments */
-    for (uint32_t idx = threadIdx.x; idx < tile_n; idx += NTHREADS) {
+    for (uint32_t idx = threadIdx.x; idx < tile_n; idx += NTHREADS_SC) {
       int64_t key = lk[idx];
       uint32_t dig = d_digit(key, mode, shift, mask);
       uint32_t dst = goff[dig] + (idx - excl[dig]);
@@ -507,21 +537,21 @@ __global__ __launch_bounds__(NTHREADS, 2) void k_scatter(
 /* per-thread chunks are CONTIGUOUS 16-element runs, so LDS indices are
  * padded (PIDX: +1 slot per 16) — the unpadded layout put lanes l and
  * l+2 on the same bank (16-way conflict, measured 1.26 ms/launch). */
-#define PIDX(j) ((j) + ((j) >> 4))
+#define PIDX(j) ((j) + ((j) >> 3))
 __global__ __launch_bounds__(NTHREADS, 2) void k_reduce_tiles(
     const int64_t *__restrict__ keys, const double *__restrict__ vals,
     uint64_t n, int64_t *__restrict__ keys_scr, double *__restrict__ sums_scr,
     double *__restrict__ tile_T, uint32_t *__restrict__ tile_anyend,
     uint32_t *__restrict__ tile_nruns, uint64_t ntiles) {
-  __shared__ int64_t lk[TILE + TILE / 16 + 1];   /* padded; +1: lookahead */
-  __shared__ double lv[TILE + TILE / 16];
+  __shared__ int64_t lk[RTILE + RTILE / 8 + 1];  /* padded; +1: lookahead */
+  __shared__ double lv[RTILE + RTILE / 8];
   __shared__ double t_carry[NTHREADS];   /* per-thread trailing open sum */
   __shared__ uint32_t t_flag[NTHREADS];  /* thread chunk has a head      */
   __shared__ uint32_t t_ends[NTHREADS];  /* run ends in thread chunk     */
 
   for (uint64_t t = blockIdx.x; t < ntiles; t += gridDim.x) {
-    const uint64_t start = t * (uint64_t)TILE;
-    const uint64_t end = min(start + TILE, n);
+    const uint64_t start = t * (uint64_t)RTILE;
+    const uint64_t end = min(start + RTILE, n);
     const uint32_t tile_n = (uint32_t)(end - start);
     for (uint32_t i = threadIdx.x; i < tile_n; i += NTHREADS) {
       lk[PIDX(i)] = keys[start + i];
@@ -531,8 +561,8 @@ __global__ __launch_bounds__(NTHREADS, 2) void k_reduce_tiles(
       lk[PIDX(tile_n)] = (end < n) ? keys[end] : 0; /* lookahead (unused if end==n) */
     __syncthreads();
 
-    const uint32_t c0 = threadIdx.x * ELEMS_PER_THREAD;         /* chunk */
-    const uint32_t c1 = min(c0 + (uint32_t)ELEMS_PER_THREAD, tile_n);
+    const uint32_t c0 = threadIdx.x * EPT_R;                    /* chunk */
+    const uint32_t c1 = min(c0 + (uint32_t)EPT_R, tile_n);
 
     /* pass 1: trailing open sum + has-head + end count per thread chunk */
     double open = 0.0;
@@ -715,7 +745,7 @@ __global__ __launch_bounds__(NTHREADS) void k_compact_runs(
   for (uint64_t t = blockIdx.x; t < ntiles; t += gridDim.x) {
     uint32_t nr = tile_nruns[t];
     uint64_t base = tile_outoff[t];
-    uint64_t src = t * (uint64_t)TILE;
+    uint64_t src = t * (uint64_t)RTILE;
     for (uint32_t j = threadIdx.x; j < nr; j += NTHREADS) {
       double s = sums_scr[src + j];
       if (j == 0) s += tile_C[t];
@@ -736,11 +766,11 @@ __global__ __launch_bounds__(NTHREADS) void k_head_tiles(
     uint32_t *__restrict__ tile_nheads, uint64_t ntiles) {
   __shared__ uint32_t cnt[NTHREADS];
   for (uint64_t t = blockIdx.x; t < ntiles; t += gridDim.x) {
-    const uint64_t start = t * (uint64_t)TILE;
-    const uint64_t end = min(start + TILE, n);
+    const uint64_t start = t * (uint64_t)RTILE;
+    const uint64_t end = min(start + RTILE, n);
     /* count heads per thread (strided chunks of ELEMS_PER_THREAD) */
-    const uint64_t c0 = start + threadIdx.x * ELEMS_PER_THREAD;
-    const uint64_t c1 = min(c0 + ELEMS_PER_THREAD, end);
+    const uint64_t c0 = start + threadIdx.x * EPT_R;
+    const uint64_t c1 = min(c0 + EPT_R, end);
     uint32_t nh = 0;
     for (uint64_t i = c0; i < c1 && i < end; i++) {
       bool head = (i == 0) || keys[i] != keys[i - 1];
@@ -777,7 +807,7 @@ __global__ __launch_bounds__(NTHREADS) void k_compact_heads(
   for (uint64_t t = blockIdx.x; t < ntiles; t += gridDim.x) {
     uint32_t nh = tile_nheads[t];
     uint64_t base = tile_outoff[t];
-    uint64_t src = t * (uint64_t)TILE;
+    uint64_t src = t * (uint64_t)RTILE;
     for (uint32_t j = threadIdx.x; j < nh; j += NTHREADS) {
       uk[base + j] = keys_scr[src + j];
       starts[base + j] = idx_scr[src + j];
@@ -1042,11 +1072,7 @@ extern "C" int dprk_reserve(dprk_ctx *c, uint64_t max_n, int with_vals) {
   free_all(c);
   if (max_n == 0) max_n = 1;
   uint64_t ntiles = ceil_div(max_n, TILE);
-  uint64_t nseg = ceil_div(ntiles, SEG_TILES);
-  if (nseg > RADIX) {
-    set_err("ntiles too large");
-    return 1;
-  }
+  uint64_t ntiles_r = ceil_div(max_n, RTILE); /* reduce/head tiling */
   HIP_CHECK(hipMalloc(&c->keys[0], max_n * 8));
   HIP_CHECK(hipMalloc(&c->keys[1], max_n * 8));
   if (with_vals) {
@@ -1058,11 +1084,11 @@ extern "C" int dprk_reserve(dprk_ctx *c, uint64_t max_n, int with_vals) {
   HIP_CHECK(hipMalloc(&c->seg_sums, (uint64_t)RADIX * RADIX * 4));
   HIP_CHECK(hipMalloc(&c->row_tot, RADIX * 8));
   HIP_CHECK(hipMalloc(&c->d_orand, 2 * 8));
-  HIP_CHECK(hipMalloc(&c->tile_T, ntiles * 8));
-  HIP_CHECK(hipMalloc(&c->tile_anyend, ntiles * 4));
-  HIP_CHECK(hipMalloc(&c->tile_nruns, ntiles * 4));
-  HIP_CHECK(hipMalloc(&c->tile_outoff, ntiles * 8));
-  HIP_CHECK(hipMalloc(&c->tile_C, ntiles * 8));
+  HIP_CHECK(hipMalloc(&c->tile_T, ntiles_r * 8));
+  HIP_CHECK(hipMalloc(&c->tile_anyend, ntiles_r * 4));
+  HIP_CHECK(hipMalloc(&c->tile_nruns, ntiles_r * 4));
+  HIP_CHECK(hipMalloc(&c->tile_outoff, ntiles_r * 8));
+  HIP_CHECK(hipMalloc(&c->tile_C, ntiles_r * 8));
   HIP_CHECK(hipMalloc(&c->d_scalar, 4 * 8));
   c->cap = max_n;
   c->max_tiles = ntiles;
@@ -1171,13 +1197,13 @@ static int radix_pass(dprk_ctx *c, int mode, int shift, uint32_t mask,
                            c->seg_sums, c->tile_off));
   if (c->with_vals) {
     TIMED(c, K_SCATTER,
-          hipLaunchKernelGGL(k_scatter<true>, dim3(thgrid), dim3(NTHREADS),
+          hipLaunchKernelGGL(k_scatter<true>, dim3(thgrid), dim3(NTHREADS_SC),
                              0, c->stream, c->keys[src], c->vals[src],
                              c->keys[dst], c->vals[dst], n, c->tile_hist,
                              c->tile_off, ntiles, mode, shift, mask));
   } else {
     TIMED(c, K_SCATTER,
-          hipLaunchKernelGGL(k_scatter<false>, dim3(thgrid), dim3(NTHREADS),
+          hipLaunchKernelGGL(k_scatter<false>, dim3(thgrid), dim3(NTHREADS_SC),
                              0, c->stream, c->keys[src], nullptr,
                              c->keys[dst], nullptr, n, c->tile_hist,
                              c->tile_off, ntiles, mode, shift, mask));
@@ -1266,7 +1292,7 @@ extern "C" int dprk_reduce_by_key_sum(dprk_ctx *c, uint64_t *n_out) {
   if (c->n == 0) { if (n_out) *n_out = 0; return 0; }
   if (!c->sorted && dprk_sort_pairs(c)) return 1;
   uint64_t n = c->n;
-  uint64_t ntiles = ceil_div(n, TILE);
+  uint64_t ntiles = ceil_div(n, RTILE);
   int src = c->cur, scr = c->cur ^ 1;
   int grid = (int)min(ntiles, (uint64_t)2048);
   TIMED(c, K_REDUCE,
@@ -1299,7 +1325,7 @@ extern "C" int dprk_reduce_by_key_sum(dprk_ctx *c, uint64_t *n_out) {
 /* run extraction into ctx group arrays (assumes sorted) */
 static int extract_runs(dprk_ctx *c) {
   uint64_t n = c->n;
-  uint64_t ntiles = ceil_div(n, TILE);
+  uint64_t ntiles = ceil_div(n, RTILE);
   int src = c->cur, scr = c->cur ^ 1;
   int grid = (int)min(ntiles, (uint64_t)2048);
   /* idx scratch reuses the alt VALS buffer as u64 (same 8B stride);

@@ -1,0 +1,168 @@
+"""Engine-eligible RDDs: int64-key / float64-value pair datasets held as
+numpy arrays, routed through the HIP engine for the hot-path operators
+("reduceByKey on an int64/f64 RDD routes to the engine;
+everything else falls back to the M0 CPU path — that fallback IS dpark's
+behavioral contract for config 1").
+
+Routing policy (tier rule ③ — no silent fallback ON the engine path):
+  - GPU present (/dev/kfd) -> the HIP engine is THE path; a missing or
+    unloadable libdprk.so raises EngineUnavailable.
+  - no GPU (CPU-only container) -> the generic CPU path serves the same
+    API (how the reference behaves everywhere).
+"""
+
+from __future__ import annotations
+
+import operator
+import os
+
+import numpy as np
+
+from . import engine as _engine
+from .rdd import RDD, Split
+
+
+def gpu_present() -> bool:
+    return os.path.exists("/dev/kfd")
+
+
+def engine_required() -> bool:
+    """True iff the engine MUST be used (GPU present).  Raises loudly if
+    the native library is unavailable on a GPU machine."""
+    if not gpu_present():
+        return False
+    _engine.load_library()  # raises EngineUnavailable with the cause
+    return True
+
+
+_ADD_FUNCS = None
+
+
+def _is_add(f) -> bool:
+    global _ADD_FUNCS
+    if _ADD_FUNCS is None:
+        import builtins  # noqa
+        _ADD_FUNCS = {operator.add, operator.iadd}
+    return f in _ADD_FUNCS
+
+
+class VectorPairsRDD(RDD):
+    """Pair RDD backed by numpy int64 keys / float64 vals.  Iterating a
+    split yields python (k, v) tuples, so every generic RDD operator
+    composes; the hot-path operators below route to the engine."""
+
+    def __init__(self, ctx, keys, vals, numSlices=None):
+        super().__init__(ctx)
+        self.keys = np.ascontiguousarray(keys, dtype=np.int64)
+        self.vals = np.ascontiguousarray(vals, dtype=np.float64)
+        assert len(self.keys) == len(self.vals)
+        numSlices = numSlices or ctx.defaultParallelism
+        n = len(self.keys)
+        self._bounds = [(n * i) // numSlices for i in range(numSlices + 1)]
+        self._splits = [Split(i) for i in range(numSlices)]
+
+    def splits(self):
+        return self._splits
+
+    def compute(self, split):
+        lo, hi = self._bounds[split.index], self._bounds[split.index + 1]
+        return zip(self.keys[lo:hi].tolist(), self.vals[lo:hi].tolist())
+
+    # -- engine-routed hot path ------------------------------------------
+    def _engine_run(self, op, other=None):
+        eng = _engine.Engine(device=0)
+        try:
+            eng.reserve(max(len(self.keys), 1), with_vals=True)
+            eng.put(self.keys, self.vals)
+            if op == "reduce":
+                eng.reduce_by_key_sum()
+                k, v = eng.get()
+                return VectorPairsRDD(self.ctx, k, v, len(self._splits))
+            if op == "sort":
+                eng.sort()
+                k, v = eng.get()
+                return VectorPairsRDD(self.ctx, k, v, len(self._splits))
+            if op == "group":
+                uk, starts, counts = eng.group_by_key()
+                _, vals = eng.get()
+                return uk, starts, counts, vals
+            if op == "join":
+                eng_b = _engine.Engine(device=0)
+                try:
+                    eng_b.reserve(max(len(other.keys), 1), with_vals=True)
+                    eng_b.put(other.keys, other.vals)
+                    jk, jva, jvb = eng.join(eng_b)
+                    return jk, jva, jvb
+                finally:
+                    eng_b.close()
+            raise ValueError(op)
+        finally:
+            eng.close()
+
+    def reduceByKey(self, func, numSplits=None):
+        if _is_add(func) and engine_required():
+            return self._engine_run("reduce")
+        return super().reduceByKey(func, numSplits)
+
+    def sortByKey(self, key=None, reverse=False, numSplits=None):
+        if key is None and not reverse and engine_required():
+            return self._engine_run("sort")
+        return super().sortByKey(key, reverse, numSplits)
+
+    def groupByKey(self, numSplits=None):
+        if engine_required():
+            uk, starts, counts, vals = self._engine_run("group")
+            return GroupedVectorRDD(self.ctx, uk, starts, counts, vals)
+        return super().groupByKey(numSplits)
+
+    def join(self, other, numSplits=None):
+        if isinstance(other, VectorPairsRDD) and engine_required():
+            jk, jva, jvb = self._engine_run("join", other)
+            return JoinedVectorRDD(self.ctx, jk, jva, jvb)
+        return super().join(other, numSplits)
+
+    # -- array-level accessors (bulk results without tuple boxing) -------
+    def collectArrays(self):
+        return self.keys, self.vals
+
+
+class GroupedVectorRDD(RDD):
+    """groupByKey result in grouped-array layout: iterating yields
+    (key, [values]) like the reference; arrays exposed directly too."""
+
+    def __init__(self, ctx, uk, starts, counts, vals):
+        super().__init__(ctx)
+        self.unique_keys, self.starts, self.counts = uk, starts, counts
+        self.grouped_vals = vals
+        self._splits = [Split(0)]
+
+    def splits(self):
+        return self._splits
+
+    def compute(self, split):
+        for i, k in enumerate(self.unique_keys.tolist()):
+            s = int(self.starts[i])
+            c = int(self.counts[i])
+            yield k, self.grouped_vals[s:s + c].tolist()
+
+    def collectArrays(self):
+        return (self.unique_keys, self.starts, self.counts,
+                self.grouped_vals)
+
+
+class JoinedVectorRDD(RDD):
+    def __init__(self, ctx, k, va, vb):
+        super().__init__(ctx)
+        self.join_keys, self.va, self.vb = k, va, vb
+        self._splits = [Split(0)]
+
+    def splits(self):
+        return self._splits
+
+    def compute(self, split):
+        for k, a, b in zip(self.join_keys.tolist(), self.va.tolist(),
+                           self.vb.tolist()):
+            yield k, (a, b)
+
+    def collectArrays(self):
+        return self.join_keys, self.va, self.vb

@@ -127,3 +127,33 @@ def test_accumulator_broadcast(ctx):
     ctx.parallelize(range(10), 2).map(
         lambda v: acc.add(v * bc.value["x"]) or v).collect()
     assert acc.value == sum(range(10)) * 2
+
+
+def test_vector_pairs_cpu_path(ctx):
+    """VectorPairsRDD on a no-GPU machine serves the same API via the CPU
+    scheduler (reference behavior); results match the oracle."""
+    import operator
+    from dpark_amd.vector import VectorPairsRDD, gpu_present
+    rng = np.random.default_rng(9)
+    keys = rng.integers(-30, 30, 400).astype(np.int64)
+    vals = rng.random(400)
+    r = ctx.vectorPairs(keys, vals, 4)
+    assert isinstance(r, VectorPairsRDD)
+    got = dict(r.reduceByKey(operator.add, 4).collect())
+    uk, s = so.reduce_by_key_sum(keys, vals)
+    assert sorted(got.keys()) == uk.tolist()
+    np.testing.assert_allclose([got[int(k)] for k in uk], s,
+                               rtol=1e-6 if gpu_present() else 1e-9)
+    # sortByKey via either path: total order on keys
+    sk = [k for k, _ in r.sortByKey(numSplits=4).collect()]
+    assert sk == sorted(keys.tolist())
+    # join with itself
+    j = sorted(r.join(ctx.vectorPairs(keys[:100], vals[:100], 2)).collect())
+    want = sorted(dpo.join(zip(keys.tolist(), vals.tolist()),
+                           zip(keys[:100].tolist(), vals[:100].tolist())))
+    assert len(j) == len(want)
+    gb = dict(r.groupByKey(3).collect())
+    og = dpo.group_by_key(zip(keys.tolist(), vals.tolist()))
+    assert set(gb) == set(og)
+    for k in og:
+        np.testing.assert_allclose(sorted(gb[k]), sorted(og[k]), rtol=1e-9)
