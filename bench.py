@@ -71,8 +71,12 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=10)
     ap.add_argument("--warmup", type=int, default=3)
-    ap.add_argument("--n-pairs", type=int, default=N_PAIRS_PER_GPU,
-                    help="pairs per GPU (config 2 default)")
+    ap.add_argument("--n-pairs", type=int, default=None,
+                    help="pairs per GPU (default: workload-specific)")
+    ap.add_argument("--op", choices=["reduce", "sort", "join"],
+                    default="reduce",
+                    help="reduce=config 2 (headline), sort=config 3 "
+                         "(1B keys-only LSD sort), join=config 4 shape")
     ap.add_argument("--config1", action="store_true",
                     help="run the CPU plumbing check instead")
     ap.add_argument("--skip-cpu-baseline", action="store_true")
@@ -99,9 +103,16 @@ def main():
 
     from dpark_amd.engine import Engine  # fails loudly if .so missing
 
-    n_local = args.n_pairs
+    defaults = {"reduce": N_PAIRS_PER_GPU, "sort": 1_000_000_000,
+                "join": 500_000_000}
+    n_local = args.n_pairs or defaults[args.op]
+    with_vals = args.op != "sort"
     eng = Engine(device=local_rank)
-    eng.reserve(n_local, with_vals=True)
+    eng.reserve(n_local, with_vals=with_vals)
+    eng_b = None
+    if args.op == "join":
+        eng_b = Engine(device=local_rank)
+        eng_b.reserve(n_local, with_vals=True)
 
     if nranks > 1:
         # RCCL init: rank 0 creates the unique id, gloo broadcasts it
@@ -117,19 +128,35 @@ def main():
         if dist is not None:
             dist.barrier()
 
-    def step(i):
-        # regenerate the resident input (seed fixed per rank; deterministic)
-        eng.gen_pairs(seed=SEED + rank, n=n_local, key_universe=KEY_UNIVERSE)
+    def exchange_if_dist(e):
         if nranks > 1:
-            send = eng.partition(nranks)              # rank-digit layout
+            send = e.partition(nranks)                # rank-digit layout
             counts = torch.tensor(send.astype("int64"))
             mat = [torch.zeros(nranks, dtype=torch.int64)
                    for _ in range(nranks)]
             dist.all_gather(mat, counts)              # count matrix (tiny)
             recv = [int(mat[r][rank]) for r in range(nranks)]
-            eng.exchange([int(x) for x in send], recv)
-        n_unique = eng.reduce_by_key_sum()
-        return n_unique
+            e.exchange([int(x) for x in send], recv)
+
+    def step(i):
+        # regenerate the resident input (seed fixed per rank; deterministic)
+        if args.op == "reduce":
+            eng.gen_pairs(seed=SEED + rank, n=n_local,
+                          key_universe=KEY_UNIVERSE)
+            exchange_if_dist(eng)
+            return eng.reduce_by_key_sum()
+        if args.op == "sort":
+            # config 3: 1B int64 keys uniform over full 64 bits, keys only
+            eng.gen_pairs(seed=7 + rank, n=n_local, key_universe=0)
+            eng.sort()
+            return n_local
+        # config 4 shape: two sides, keys from 2^26 universe, co-partition
+        # (shared hash digits) -> sort both -> merge join
+        eng.gen_pairs(seed=13 + rank, n=n_local, key_universe=1 << 26)
+        eng_b.gen_pairs(seed=17 + rank, n=n_local, key_universe=1 << 26)
+        exchange_if_dist(eng)
+        exchange_if_dist(eng_b)
+        return eng.join_count(eng_b)
 
     # warmup
     nu = None
@@ -161,6 +188,8 @@ def main():
         return
 
     total_pairs = n_local * nranks * args.steps
+    if args.op == "join":
+        total_pairs *= 2  # pairs processed across both sides
     value = total_pairs / elapsed / 1e6
     ms_per_step = elapsed * 1000.0 / args.steps
     gen_ms = s["ms_gen"] / args.steps
@@ -194,7 +223,9 @@ def main():
                                   key_universe=KEY_UNIVERSE)
 
     out = {
-        "metric": "M KV-pairs/sec through reduceByKey shuffle",
+        "metric": {"reduce": "M KV-pairs/sec through reduceByKey shuffle",
+                   "sort": "M keys/sec through sortByKey (LSD radix)",
+                   "join": "M input KV-pairs/sec through join"}[args.op],
         "value": round(value, 2),
         "unit": "M KV-pairs/s",
         "n_gpus": nranks,
@@ -207,10 +238,18 @@ def main():
         "dtype": "int64+f64",
         "data": "synthetic",
         "config": {
-            "workload": "reduceByKey-100M-int64-f64-uniform-2^24-per-gpu "
-                        "(BASELINE.json configs[1])",
+            "workload": {
+                "reduce": "reduceByKey-100M-int64-f64-uniform-2^24-per-gpu "
+                          "(BASELINE.json configs[1])",
+                "sort": "sortByKey-1B-int64-keys-full64 "
+                        "(BASELINE.json configs[2])",
+                "join": "join-2x500M-int64-f64-universe-2^26 "
+                        "(BASELINE.json configs[3] shape)",
+            }[args.op],
+            "op": args.op,
             "n_pairs_per_gpu": n_local,
-            "key_universe": "2^24",
+            "key_universe": {"reduce": "2^24", "sort": "2^64",
+                             "join": "2^26"}[args.op],
             "seed": SEED,
             "parallelism": f"dp{nranks}" if nranks > 1 else "single-gpu",
             "step_includes_gen_ms": round(gen_ms, 3),
