@@ -119,6 +119,8 @@ struct dprk_ctx {
   uint32_t *seg_sums = nullptr;   /* [RADIX][nseg]                       */
   uint64_t *row_tot = nullptr;    /* [RADIX] per-bucket totals (partition)*/
   uint64_t *d_orand = nullptr;    /* {OR, AND} of biased keys            */
+  uint64_t *alias_table = nullptr; /* packed (thresh<<32)|alias           */
+  uint64_t alias_universe = 0;
   uint64_t max_tiles = 0;
 
   /* segmented-reduce / run-extract scratch */
@@ -267,6 +269,32 @@ __global__ __launch_bounds__(NTHREADS) void k_gen(
       } else {
         vals[i] = (double)(v0 >> 11) * (1.0 / 9007199254740992.0);
       }
+    }
+  }
+}
+
+/* Zipf (or any discrete distribution) sampling via a caller-provided
+ * alias table: one random 8-B table load per sample.
+ * table[slot] = (thresh << 32) | alias; take slot if u2_hi32 < thresh
+ * else alias.  Bit-exact mirror: oracle/synth.py gen_zipf (the table is
+ * BUILT host-side in numpy and shipped via dprk_set_alias_table, so no
+ * cross-libm drift can break parity). */
+__global__ __launch_bounds__(NTHREADS) void k_gen_alias(
+    int64_t *__restrict__ keys, double *__restrict__ vals, uint64_t n,
+    uint64_t seed_hi, const uint64_t *__restrict__ table, uint64_t universe) {
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    uint64_t x = seed_hi + i;
+    uint64_t u1 = d_splitmix64(x ^ 0x5851F42D4C957F2Dull);
+    uint64_t slot = __umul64hi(u1, universe);
+    uint64_t e = table[slot];
+    uint64_t u2 = d_splitmix64(x ^ 0x14057B7EF767814Full);
+    uint64_t k = ((u2 >> 32) < (e >> 32)) ? slot : (e & 0xFFFFFFFFull);
+    keys[i] = (int64_t)k;
+    if (vals) {
+      uint64_t vb = d_splitmix64(x ^ 0xDEADBEEFCAFEBABEull);
+      vals[i] = (double)(vb >> 11) * (1.0 / 9007199254740992.0); /* 2^-53 */
     }
   }
 }
@@ -1010,8 +1038,11 @@ __global__ __launch_bounds__(NTHREADS) void k_join_count(
   }
 }
 
-/* emit: one wave per A-run (grid-stride); writes (k, va, vb) tuples.
- * dpark join semantics: per-key cross product. */
+/* emit: per-key cross products (dpark join semantics — SURVEY.md §3.3).
+ * One 16-lane sub-wave group per A-run (avg products are small — config
+ * 4: ~14 — so a full wave per run left 78% of lanes idle; measured 37 ms
+ * for 9.3e8 outputs).  Consecutive groups emit consecutive runs, whose
+ * output ranges are adjacent => coalesced-ish stores. */
 __global__ __launch_bounds__(NTHREADS) void k_join_emit(
     const int64_t *__restrict__ uka, const uint64_t *__restrict__ starta,
     const uint64_t *__restrict__ cnta, uint64_t ga,
@@ -1020,19 +1051,19 @@ __global__ __launch_bounds__(NTHREADS) void k_join_emit(
     const uint64_t *__restrict__ bidx, const double *__restrict__ va,
     const double *__restrict__ vb, int64_t *__restrict__ out_k,
     double *__restrict__ out_a, double *__restrict__ out_b) {
-  const int wave = threadIdx.x / 64, lane = threadIdx.x % 64;
-  const uint64_t nwaves = (uint64_t)gridDim.x * (NTHREADS / 64);
-  for (uint64_t r = (uint64_t)blockIdx.x * (NTHREADS / 64) + wave; r < ga;
-       r += nwaves) {
+  const uint32_t lane16 = threadIdx.x & 15;
+  const uint64_t ngroups = (uint64_t)gridDim.x * (NTHREADS / 16);
+  const uint64_t gid = (uint64_t)blockIdx.x * (NTHREADS / 16) + threadIdx.x / 16;
+  for (uint64_t r = gid; r < ga; r += ngroups) {
     uint64_t p = prod[r];
     if (p == 0) continue;
     int64_t k = uka[r];
     uint64_t sa = starta[r];
     uint64_t b = bidx[r];
     uint64_t sb = startb[b];
-    uint64_t cb = cntb[b];
+    uint32_t cb = (uint32_t)cntb[b];
     uint64_t base = prod_off[r];
-    for (uint64_t j = lane; j < p; j += 64) {
+    for (uint64_t j = lane16; j < p; j += 16) {
       out_k[base + j] = k;
       out_a[base + j] = va[sa + j / cb];
       out_b[base + j] = vb[sb + j % cb];
@@ -1127,6 +1158,7 @@ static int free_all(dprk_ctx *c) {
   if (c->seg_sums) HIP_CHECK(hipFree(c->seg_sums));
   if (c->row_tot) HIP_CHECK(hipFree(c->row_tot));
   if (c->d_orand) HIP_CHECK(hipFree(c->d_orand));
+  if (c->alias_table) HIP_CHECK(hipFree(c->alias_table));
   if (c->tile_T) HIP_CHECK(hipFree(c->tile_T));
   if (c->tile_anyend) HIP_CHECK(hipFree(c->tile_anyend));
   if (c->tile_nruns) HIP_CHECK(hipFree(c->tile_nruns));
@@ -1140,6 +1172,7 @@ static int free_all(dprk_ctx *c) {
   if (c->scan_part) HIP_CHECK(hipFree(c->scan_part));
   c->tile_hist = nullptr; c->tile_off = nullptr; c->seg_sums = nullptr;
   c->row_tot = nullptr; c->d_orand = nullptr; c->tile_T = nullptr;
+  c->alias_table = nullptr; c->alias_universe = 0;
   c->tile_anyend = nullptr; c->tile_nruns = nullptr; c->tile_outoff = nullptr;
   c->tile_C = nullptr; c->d_scalar = nullptr; c->scan_part = nullptr;
   c->chain_open = nullptr; c->chunk_T = nullptr; c->chunk_F = nullptr;
@@ -1203,15 +1236,41 @@ extern "C" int dprk_count(dprk_ctx *c, uint64_t *n_out) {
   return 0;
 }
 
+extern "C" int dprk_set_alias_table(dprk_ctx *c, const uint64_t *table,
+                                    uint64_t universe) {
+  if (!c) { set_err("null ctx"); return 1; }
+  HIP_CHECK(hipSetDevice(c->device));
+  if (c->alias_table) { HIP_CHECK(hipFree(c->alias_table));
+                        c->alias_table = nullptr; }
+  HIP_CHECK(hipMalloc(&c->alias_table, universe * 8));
+  HIP_CHECK(hipMemcpyAsync(c->alias_table, table, universe * 8,
+                           hipMemcpyHostToDevice, c->stream));
+  HIP_CHECK(hipStreamSynchronize(c->stream));
+  c->alias_universe = universe;
+  return 0;
+}
+
 extern "C" int dprk_gen_pairs(dprk_ctx *c, uint64_t seed, uint64_t n,
                               int dist, uint64_t key_universe) {
   if (!c) { set_err("null ctx"); return 1; }
-  if (dist != 0) { set_err("dist %d not implemented (zipf is a later "
-                           "milestone)", dist); return 1; }
+  if (dist != 0 && dist != 1) { set_err("gen: unknown dist %d", dist);
+                                return 1; }
+  if (dist == 1 && (!c->alias_table || c->alias_universe != key_universe)) {
+    set_err("gen dist=1: call dprk_set_alias_table(universe=%llu) first",
+            (unsigned long long)key_universe);
+    return 1;
+  }
   if (n > c->cap) { set_err("gen: n > reserved cap"); return 1; }
   HIP_CHECK(hipSetDevice(c->device));
   uint64_t keymask =
       (key_universe == 0) ? ~0ull : (key_universe - 1); /* power of two */
+  if (dist == 1) {
+    TIMED(c, K_GEN,
+          hipLaunchKernelGGL(k_gen_alias, dim3(stream_grid(n)), dim3(NTHREADS),
+                             0, c->stream, c->keys[c->cur],
+                             c->with_vals ? c->vals[c->cur] : nullptr, n,
+                             seed << 32, c->alias_table, key_universe));
+  } else
   TIMED(c, K_GEN,
         hipLaunchKernelGGL(k_gen, dim3(stream_grid(n)), dim3(NTHREADS), 0,
                            c->stream, c->keys[c->cur],
@@ -1222,9 +1281,13 @@ extern "C" int dprk_gen_pairs(dprk_ctx *c, uint64_t seed, uint64_t n,
   /* keys = splitmix64 & keymask, so biased keys lie in
    * [SIGNFLIP, SIGNFLIP | keymask]: conservative OR/AND bound — may only
    * RUN an avoidable pass, never skip a needed one */
-  if (keymask != ~0ull) {
+  if (dist == 0 && keymask != ~0ull) {
     c->orand_valid = 1;
     c->known_or = SIGNFLIP | keymask;
+    c->known_and = SIGNFLIP;
+  } else if (dist == 1 && (key_universe & (key_universe - 1)) == 0) {
+    c->orand_valid = 1;   /* alias keys are < universe (power of two) */
+    c->known_or = SIGNFLIP | (key_universe - 1);
     c->known_and = SIGNFLIP;
   } else {
     c->orand_valid = 0;

@@ -53,7 +53,8 @@ class DprkStats(ctypes.Structure):
 # every symbol include/dprk.h declares — tests/test_cabi.py asserts these
 SYMBOLS = [
     "dprk_init", "dprk_free", "dprk_last_error", "dprk_build_info",
-    "dprk_reserve", "dprk_gen_pairs", "dprk_put_pairs", "dprk_get_pairs",
+    "dprk_reserve", "dprk_gen_pairs", "dprk_set_alias_table",
+    "dprk_put_pairs", "dprk_get_pairs",
     "dprk_count", "dprk_partition", "dprk_sort_pairs",
     "dprk_reduce_by_key_sum", "dprk_group_by_key", "dprk_get_groups",
     "dprk_join", "dprk_get_join", "dprk_comm_unique_id", "dprk_comm_init",
@@ -91,6 +92,7 @@ def load_library():
     lib.dprk_free.restype = None
     lib.dprk_reserve.argtypes = [p, u64, ctypes.c_int]
     lib.dprk_gen_pairs.argtypes = [p, u64, u64, ctypes.c_int, u64]
+    lib.dprk_set_alias_table.argtypes = [p, ctypes.c_void_p, u64]
     lib.dprk_put_pairs.argtypes = [p, ctypes.c_void_p, ctypes.c_void_p, u64]
     lib.dprk_get_pairs.argtypes = [p, ctypes.c_void_p, ctypes.c_void_p, u64,
                                    ctypes.POINTER(u64)]
@@ -151,6 +153,15 @@ class Engine:
         _check(self._lib, self._lib.dprk_gen_pairs(self._ctx, seed, n, dist,
                                                    key_universe),
                "dprk_gen_pairs")
+
+    def set_alias_table(self, table):
+        import numpy as _np
+        table = _np.ascontiguousarray(table, dtype=_np.uint64)
+        _check(self._lib,
+               self._lib.dprk_set_alias_table(
+                   self._ctx, table.ctypes.data_as(ctypes.c_void_p),
+                   len(table)),
+               "dprk_set_alias_table")
 
     def put(self, keys, vals=None):
         keys = np.ascontiguousarray(keys, dtype=np.int64)

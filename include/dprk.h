@@ -52,10 +52,15 @@ int dprk_reserve(dprk_ctx *ctx, uint64_t max_n, int with_vals);
 /* ---- data in/out ----------------------------------------------------- */
 /* Device-side PRNG per the synth spec (oracle/synth.py is the CPU mirror):
  * dist 0 = uniform over key_universe (power of two; 0 => full 64-bit),
- * dist 1 = zipf(1.1) via inverse-CDF table (reserved, not yet implemented).
+ * dist 1 = alias-table sampling (zipf etc.; dprk_set_alias_table).
  */
 int dprk_gen_pairs(dprk_ctx *ctx, uint64_t seed, uint64_t n, int dist,
                    uint64_t key_universe);
+/* dist 1 (zipf etc.) samples a caller-provided alias table (built host
+ * side — oracle/synth.py build_alias_table is the normative builder, so
+ * the CPU mirror is bit-identical): table[slot] = (thresh<<32)|alias. */
+int dprk_set_alias_table(dprk_ctx *ctx, const uint64_t *table,
+                         uint64_t universe);
 int dprk_put_pairs(dprk_ctx *ctx, const int64_t *keys, const double *vals,
                    uint64_t n); /* vals may be NULL for keys-only */
 int dprk_get_pairs(dprk_ctx *ctx, int64_t *keys, double *vals, uint64_t cap,
