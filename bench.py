@@ -73,10 +73,11 @@ def main():
     ap.add_argument("--warmup", type=int, default=3)
     ap.add_argument("--n-pairs", type=int, default=None,
                     help="pairs per GPU (default: workload-specific)")
-    ap.add_argument("--op", choices=["reduce", "sort", "join"],
+    ap.add_argument("--op", choices=["reduce", "sort", "join", "zipf"],
                     default="reduce",
                     help="reduce=config 2 (headline), sort=config 3 "
-                         "(1B keys-only LSD sort), join=config 4 shape")
+                         "(1B keys-only LSD sort), join=config 4 shape, "
+                         "zipf=config 5 (1B Zipf(1.1) skewed reduceByKey)")
     ap.add_argument("--config1", action="store_true",
                     help="run the CPU plumbing check instead")
     ap.add_argument("--skip-cpu-baseline", action="store_true")
@@ -104,11 +105,17 @@ def main():
     from dpark_amd.engine import Engine  # fails loudly if .so missing
 
     defaults = {"reduce": N_PAIRS_PER_GPU, "sort": 1_000_000_000,
-                "join": 500_000_000}
+                "join": 500_000_000, "zipf": 1_000_000_000}
     n_local = args.n_pairs or defaults[args.op]
     with_vals = args.op != "sort"
     eng = Engine(device=local_rank)
     eng.reserve(n_local, with_vals=with_vals)
+    if args.op == "zipf":
+        from oracle.synth import build_alias_table, zipf_weights
+        log("building Zipf(1.1) alias table over 2^24 keys ...")
+        t0 = time.perf_counter()
+        eng.set_alias_table(build_alias_table(zipf_weights(1 << 24, 1.1)))
+        log(f"alias table built+uploaded in {time.perf_counter()-t0:.1f}s")
     eng_b = None
     if args.op == "join":
         eng_b = Engine(device=local_rank)
@@ -150,6 +157,12 @@ def main():
             eng.gen_pairs(seed=7 + rank, n=n_local, key_universe=0)
             eng.sort()
             return n_local
+        if args.op == "zipf":
+            # config 5: Zipf(alpha=1.1) skewed keys over 2^24
+            eng.gen_pairs(seed=11 + rank, n=n_local, key_universe=1 << 24,
+                          dist=1)
+            exchange_if_dist(eng)
+            return eng.reduce_by_key_sum()
         # config 4 shape: two sides, keys from 2^26 universe, co-partition
         # (shared hash digits) -> sort both -> merge join
         eng.gen_pairs(seed=13 + rank, n=n_local, key_universe=1 << 26)
@@ -225,7 +238,8 @@ def main():
     out = {
         "metric": {"reduce": "M KV-pairs/sec through reduceByKey shuffle",
                    "sort": "M keys/sec through sortByKey (LSD radix)",
-                   "join": "M input KV-pairs/sec through join"}[args.op],
+                   "join": "M input KV-pairs/sec through join",
+                   "zipf": "M KV-pairs/sec through skewed reduceByKey"}[args.op],
         "value": round(value, 2),
         "unit": "M KV-pairs/s",
         "n_gpus": nranks,
@@ -245,11 +259,13 @@ def main():
                         "(BASELINE.json configs[2])",
                 "join": "join-2x500M-int64-f64-universe-2^26 "
                         "(BASELINE.json configs[3] shape)",
+                "zipf": "reduceByKey-1B-zipf1.1-2^24 "
+                        "(BASELINE.json configs[4])",
             }[args.op],
             "op": args.op,
             "n_pairs_per_gpu": n_local,
             "key_universe": {"reduce": "2^24", "sort": "2^64",
-                             "join": "2^26"}[args.op],
+                             "join": "2^26", "zipf": "2^24"}[args.op],
             "seed": SEED,
             "parallelism": f"dp{nranks}" if nranks > 1 else "single-gpu",
             "step_includes_gen_ms": round(gen_ms, 3),

@@ -65,8 +65,12 @@ def build_alias_table(weights):
     p = w / w.sum() * n          # scaled probabilities, mean 1
     alias = np.zeros(n, dtype=np.uint64)
     prob = np.ones(n, dtype=np.float64)
-    small = [i for i in range(n) if p[i] < 1.0]
-    large = [i for i in range(n) if p[i] >= 1.0]
+    # preallocated stacks (a python-list Vose loop costs ~1 us/entry; this
+    # runs once per bench over a 2^24-entry table)
+    idx = np.arange(n, dtype=np.int64)
+    is_small = p < 1.0
+    small = idx[is_small].tolist()
+    large = idx[~is_small].tolist()
     p = p.copy()
     while small and large:
         s = small.pop()

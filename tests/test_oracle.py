@@ -96,3 +96,24 @@ def test_group_by_key_multiset():
     g = dpo.group_by_key(pairs)
     assert sorted(g[1]) == [10.0, 10.0, 30.0]
     assert g[2] == [20.0]
+
+
+def test_alias_table_distribution():
+    """The alias table reproduces the target distribution and the zipf
+    mirror is deterministic."""
+    from oracle.synth import build_alias_table, zipf_weights, gen_zipf
+    uni = 1 << 10
+    w = zipf_weights(uni, 1.1)
+    table = build_alias_table(w)
+    assert len(table) == uni
+    k1, v1 = gen_zipf(3, 200_000, table, uni)
+    k2, _ = gen_zipf(3, 200_000, table, uni)
+    assert np.array_equal(k1, k2)
+    assert k1.min() >= 0 and k1.max() < uni
+    # empirical head frequencies close to target
+    p = w / w.sum()
+    for key in range(3):
+        emp = float(np.mean(k1 == key))
+        assert abs(emp - p[key]) < 0.01, (key, emp, p[key])
+    # heavy skew: top key much hotter than median key
+    assert np.mean(k1 == 0) > 50 * np.mean(k1 == uni // 2)
