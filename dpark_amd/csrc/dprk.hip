@@ -445,6 +445,16 @@ __global__ __launch_bounds__(NTHREADS) void k_tile_off(
  * running counters — no cross-wave traffic between two barriers.
  * Digit-sorted staging in LDS, then a sequential sweep writes
  * equal-digit runs to contiguous global addresses. */
+/* wave-level inclusive scan over 64 lanes (no barriers) */
+__device__ __forceinline__ uint32_t d_wave_incl_scan(uint32_t v, int lane) {
+#pragma unroll
+  for (int d = 1; d < 64; d <<= 1) {
+    uint32_t up = (uint32_t)__shfl_up((int)v, d, 64);
+    if (lane >= d) v += up;
+  }
+  return v;
+}
+
 template <bool WITH_VALS>
 __global__ __launch_bounds__(NTHREADS_SC, 2) void k_scatter(
     const int64_t *__restrict__ keys_in, const double *__restrict__ vals_in,
@@ -461,59 +471,70 @@ __global__ __launch_bounds__(NTHREADS_SC, 2) void k_scatter(
   __shared__ uint32_t goff[RADIX];     /* global write base per digit */
   __shared__ uint32_t h[NW][RADIX];    /* per-wave hist -> wave bases
                                           -> running counters (in place) */
+  __shared__ uint32_t wtot[NW > 4 ? NW : 4]; /* cross-wave scan carries  */
 
   const int wave = threadIdx.x / 64, lane = threadIdx.x % 64;
   const uint64_t lowmask = ((uint64_t)1 << lane) - 1;
+  const uint64_t stride = gridDim.x;
 
-  for (uint64_t t = blockIdx.x; t < ntiles; t += gridDim.x) {
+  /* v5: cross-tile software pipeline — the NEXT tile's key/val loads are
+   * issued before the current tile's write phase, so HBM reads overlap
+   * the write phase instead of serializing behind it. */
+  int64_t rk[EPL], rk2[EPL];
+  double rv[EPL], rv2[EPL];
+  uint32_t rd[EPL];
+
+  uint64_t t = blockIdx.x;
+  if (t < ntiles) {
+    /* prologue: load tile t */
     const uint64_t start = t * (uint64_t)TILE;
-    const uint64_t end = min(start + TILE, n);
-    const uint32_t tile_n = (uint32_t)(end - start);
+    const uint32_t tile_n = (uint32_t)(min(start + TILE, n) - start);
     const uint32_t wbeg = wave * WBLK;
-
-    /* phase 0: per-tile digit metadata (coalesced tile-major loads);
-     * zero this wave's histogram row */
-    uint32_t myv = 0;
-    if (threadIdx.x < RADIX) {
-      myv = tile_hist[t * RADIX + threadIdx.x];
-      excl[threadIdx.x] = myv;
-      goff[threadIdx.x] = tile_off[t * RADIX + threadIdx.x];
-    }
-    for (int j = 0; j < RADIX / 64; j++) h[wave][lane * (RADIX / 64) + j] = 0;
-    __syncthreads();
-    for (int off = 1; off < RADIX; off <<= 1) {
-      uint32_t add = 0;
-      if (threadIdx.x < RADIX && threadIdx.x >= (unsigned)off)
-        add = excl[threadIdx.x - off];
-      __syncthreads();
-      if (threadIdx.x < RADIX) excl[threadIdx.x] += add;
-      __syncthreads();
-    }
-    if (threadIdx.x < RADIX) {
-      uint32_t inc = excl[threadIdx.x];
-      excl[threadIdx.x] = inc - myv; /* own slot only: no race */
-    }
-    __syncthreads();
-
-    /* phase 1: load into registers + wave-private histogram */
-    int64_t rk[EPL];
-    double rv[EPL];
-    uint32_t rd[EPL];
 #pragma unroll
     for (int i = 0; i < EPL; i++) {
       uint32_t idx = wbeg + i * 64 + lane;
       if (idx < tile_n) {
         rk[i] = keys_in[start + idx];
         if (WITH_VALS) rv[i] = vals_in[start + idx];
+      }
+    }
+  }
+
+  for (; t < ntiles; t += stride) {
+    const uint64_t start = t * (uint64_t)TILE;
+    const uint64_t end = min(start + TILE, n);
+    const uint32_t tile_n = (uint32_t)(end - start);
+    const uint32_t wbeg = wave * WBLK;
+
+    /* phase 0: per-tile digit metadata; excl scan via wave shfl scans
+     * (2 barriers, vs 16 for the Hillis-Steele form); zero h rows */
+    uint32_t myv = 0;
+    if (threadIdx.x < RADIX) {
+      myv = tile_hist[t * RADIX + threadIdx.x];
+      goff[threadIdx.x] = tile_off[t * RADIX + threadIdx.x];
+      uint32_t inc = d_wave_incl_scan(myv, lane);
+      if (lane == 63) wtot[wave] = inc;
+      excl[threadIdx.x] = inc - myv; /* within-wave exclusive */
+    }
+    for (int j = 0; j < RADIX / 64; j++) h[wave][lane * (RADIX / 64) + j] = 0;
+    __syncthreads();
+    if (threadIdx.x < RADIX && wave > 0) {
+      uint32_t add = 0;
+      for (int w = 0; w < wave; w++) add += wtot[w];
+      excl[threadIdx.x] += add;
+    }
+    /* phase 1 (no loads — regs were prefetched): digits + wave hist */
+#pragma unroll
+    for (int i = 0; i < EPL; i++) {
+      uint32_t idx = wbeg + i * 64 + lane;
+      if (idx < tile_n) {
         rd[i] = d_digit(rk[i], mode, shift, mask);
         atomicAdd(&h[wave][rd[i]], 1u);
       }
     }
     __syncthreads();
 
-    /* phase 2: turn h into per-wave digit BASES in place (thread=digit):
-     * h[w][d] := excl[d] + sum_{w'<w} h[w'][d]; phase 3 then uses h as
-     * the running counters directly. */
+    /* phase 2: turn h into per-wave digit BASES in place (thread=digit) */
     if (threadIdx.x < RADIX) {
       uint32_t d = threadIdx.x;
       uint32_t b = excl[d];
@@ -545,10 +566,25 @@ __global__ __launch_bounds__(NTHREADS_SC, 2) void k_scatter(
         if (WITH_VALS) lv[pos] = rv[i];
       }
     }
+    /* issue NEXT tile's loads now — they fly during phase 4 */
+    {
+      uint64_t tn = t + stride;
+      if (tn < ntiles) {
+        const uint64_t s2 = tn * (uint64_t)TILE;
+        const uint32_t n2 = (uint32_t)(min(s2 + TILE, n) - s2);
+#pragma unroll
+        for (int i = 0; i < EPL; i++) {
+          uint32_t idx = wbeg + i * 64 + lane;
+          if (idx < n2) {
+            rk2[i] = keys_in[s2 + idx];
+            if (WITH_VALS) rv2[i] = vals_in[s2 + idx];
+          }
+        }
+      }
+    }
     __syncthreads();
 
-    /* phase 4: write out in digit-sorted tile order: equal-digit runs
-     * land in consecutive global addresses => coalesced segments */
+    /* phase 4: write out in digit-sorted tile order */
     for (uint32_t idx = threadIdx.x; idx < tile_n; idx += NTHREADS_SC) {
       int64_t key = lk[idx];
       uint32_t dig = d_digit(key, mode, shift, mask);
@@ -557,6 +593,11 @@ __global__ __launch_bounds__(NTHREADS_SC, 2) void k_scatter(
       if (WITH_VALS) vals_out[dst] = lv[idx];
     }
     __syncthreads();
+#pragma unroll
+    for (int i = 0; i < EPL; i++) {
+      rk[i] = rk2[i];
+      if (WITH_VALS) rv[i] = rv2[i];
+    }
   }
 }
 
