@@ -21,6 +21,7 @@
 #include <rccl/rccl.h>
 
 #include <cstdarg>
+#include <cstdlib>
 #include <cstdio>
 #include <cstdint>
 #include <cstring>
@@ -1729,7 +1730,10 @@ extern "C" int dprk_comm_init(dprk_ctx *c, int nranks, int rank,
 extern "C" int dprk_exchange(dprk_ctx *c, const uint64_t *send_counts,
                              const uint64_t *recv_counts, int nranks) {
   if (!c) { set_err("null ctx"); return 1; }
-  if (nranks == 1) return 0;
+  /* N=1 is a no-op — unless DPRK_FORCE_SELF_EXCHANGE is set, which runs
+   * the real grouped ncclSend/ncclRecv self path so the RCCL exchange is
+   * exercised on the single-GPU test box. */
+  if (nranks == 1 && !getenv("DPRK_FORCE_SELF_EXCHANGE")) return 0;
   if (!c->comm) { set_err("exchange: comm not initialized"); return 1; }
   if (nranks != c->nranks) { set_err("exchange: nranks mismatch"); return 1; }
   HIP_CHECK(hipSetDevice(c->device));

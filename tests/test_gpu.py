@@ -285,3 +285,33 @@ def test_zipf_gen_matches_cpu_mirror(eng):
     wk, wv = so.reduce_by_key_sum(mk, mv)
     assert got_n == len(wk) and np.array_equal(k, wk)
     np.testing.assert_allclose(v, wv, rtol=RTOL)
+
+
+def test_rccl_self_exchange():
+    """RCCL path on the 1-GPU box: comm init (nranks=1) + the real grouped
+    ncclSend/ncclRecv all-to-allv forced through the self path
+    (DPRK_FORCE_SELF_EXCHANGE).  The N>1 plan semantics are covered by
+    tests/test_dist_cpu.py (gloo, world 2); real 2/4/8-GPU runs are the
+    driver's round-end job (single-GPU gpurun — SURVEY.md §0b)."""
+    os.environ["DPRK_FORCE_SELF_EXCHANGE"] = "1"
+    try:
+        e = Engine(device=0)
+        uid = Engine.comm_unique_id()
+        e.comm_init(1, 0, uid)
+        keys, vals = gen_pairs(77, 1_000_000, 1 << 20)
+        e.reserve(len(keys), with_vals=True)
+        e.put(keys, vals)
+        counts = e.partition(1)
+        assert counts[0] == len(keys)
+        e.exchange([len(keys)], [len(keys)])     # real ncclSend/Recv to self
+        k, v = e.get()
+        assert np.array_equal(np.sort(k), np.sort(keys))
+        n = e.reduce_by_key_sum()
+        wk, wv = so.reduce_by_key_sum(keys, vals)
+        assert n == len(wk)
+        k, v = e.get()
+        assert np.array_equal(k, wk)
+        np.testing.assert_allclose(v, wv, rtol=RTOL)
+        e.close()
+    finally:
+        del os.environ["DPRK_FORCE_SELF_EXCHANGE"]
