@@ -109,7 +109,12 @@ def main():
     n_local = args.n_pairs or defaults[args.op]
     with_vals = args.op != "sort"
     eng = Engine(device=local_rank)
-    eng.reserve(n_local, with_vals=with_vals)
+    # multi-rank: a rank receives a hash-dependent share — reserve slack
+    # (uniform keys: ~+0.03%; generous 4% margin. Heavy zipf skew can
+    # exceed any margin — hot-bucket splitting is a next-row item,
+    # SURVEY.md §7 M4 — dprk_exchange then fails with a clear error.)
+    cap = n_local if nranks == 1 else int(n_local * 1.04) + 4096
+    eng.reserve(cap, with_vals=with_vals)
     if args.op == "zipf":
         from oracle.synth import build_alias_table, zipf_weights
         log("building Zipf(1.1) alias table over 2^24 keys ...")

@@ -181,12 +181,12 @@ static int timed_begin(dprk_ctx *c, int kind) {
   r.a = c->evts[2 * c->n_recs];
   r.b = c->evts[2 * c->n_recs + 1];
   r.kind = kind;
-  hipEventRecord(r.a, c->stream);
+  (void)hipEventRecord(r.a, c->stream);
   return 1;
 }
 static void timed_end(dprk_ctx *c, int began) {
   if (!began) return;
-  hipEventRecord(c->recs[c->n_recs].b, c->stream);
+  (void)hipEventRecord(c->recs[c->n_recs].b, c->stream);
   c->n_recs++;
 }
 
@@ -1117,11 +1117,12 @@ extern "C" int dprk_init(dprk_ctx **out, int device_id) {
 
 extern "C" void dprk_free(dprk_ctx *c) {
   if (!c) return;
-  hipSetDevice(c->device);
+  (void)hipSetDevice(c->device);
   free_all(c);
-  for (int i = 0; i < c->n_evts_created; i++) hipEventDestroy(c->evts[i]);
+  for (int i = 0; i < c->n_evts_created; i++)
+    (void)hipEventDestroy(c->evts[i]);
   if (c->comm) ncclCommDestroy(c->comm);
-  if (c->stream) hipStreamDestroy(c->stream);
+  if (c->stream) (void)hipStreamDestroy(c->stream);
   delete c;
 }
 

@@ -315,3 +315,41 @@ def test_rccl_self_exchange():
         e.close()
     finally:
         del os.environ["DPRK_FORCE_SELF_EXCHANGE"]
+
+
+@pytest.mark.parametrize("case", ["sorted", "reverse", "two_keys",
+                                  "alternating", "nan_inf"])
+def test_adversarial_inputs(eng, case):
+    """Inputs the reference's own tests would cover:
+    pre-sorted, reverse-sorted, two-key, alternating, non-finite values."""
+    n = 300_000
+    if case == "sorted":
+        keys = np.arange(n, dtype=np.int64) // 7
+        vals = np.linspace(0, 1, n)
+    elif case == "reverse":
+        keys = (np.arange(n, dtype=np.int64) // 7)[::-1].copy()
+        vals = np.linspace(0, 1, n)
+    elif case == "two_keys":
+        keys = np.where(np.arange(n) % 2 == 0, -5, 9).astype(np.int64)
+        vals = np.full(n, 0.25)
+    elif case == "alternating":
+        keys = (np.arange(n, dtype=np.int64) % 1000) - 500
+        vals = np.ones(n)
+    else:  # nan_inf: non-finite values must propagate like the oracle
+        keys = np.arange(n, dtype=np.int64) % 97
+        vals = np.ones(n)
+        vals[100] = np.inf
+        vals[200] = -np.inf  # key 6: inf + -inf -> nan (both paths)
+        vals[300] = np.nan
+    _put(eng, keys, vals)
+    got_n = eng.reduce_by_key_sum()
+    k, v = eng.get()
+    wk, wv = so.reduce_by_key_sum(keys, vals)
+    assert got_n == len(wk)
+    assert np.array_equal(k, wk)
+    finite = np.isfinite(wv)
+    np.testing.assert_allclose(v[finite], wv[finite], rtol=RTOL)
+    # non-finite classes must match exactly (nan==nan, inf signs)
+    assert np.array_equal(np.isnan(v), np.isnan(wv))
+    assert np.array_equal(np.isposinf(v), np.isposinf(wv))
+    assert np.array_equal(np.isneginf(v), np.isneginf(wv))
