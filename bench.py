@@ -23,6 +23,8 @@ import os
 import sys
 import time
 
+import numpy as np
+
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
 N_PAIRS_PER_GPU = 100_000_000
@@ -160,6 +162,24 @@ def main():
         if args.op == "sort":
             # config 3: 1B int64 keys uniform over full 64 bits, keys only
             eng.gen_pairs(seed=7 + rank, n=n_local, key_universe=0)
+            if nranks > 1:
+                # distributed sortByKey: sampled range splitters (dpark
+                # RangePartitioner semantics) -> all-to-allv -> local sort
+                sample = eng.sample_keys(4096)
+                obj = [None] * nranks
+                dist.all_gather_object(obj, sample)
+                allsamp = np.sort(np.concatenate(obj))
+                step_w = len(allsamp) / nranks
+                splitters = np.array(
+                    [allsamp[int(step_w * r)] for r in range(1, nranks)],
+                    dtype=np.int64)
+                send = eng.partition_range(splitters, nranks)
+                counts = torch.tensor(send.astype("int64"))
+                mat = [torch.zeros(nranks, dtype=torch.int64)
+                       for _ in range(nranks)]
+                dist.all_gather(mat, counts)
+                recv = [int(mat[r][rank]) for r in range(nranks)]
+                eng.exchange([int(x) for x in send], recv)
             eng.sort()
             return n_local
         if args.op == "zipf":

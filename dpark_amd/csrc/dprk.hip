@@ -233,11 +233,39 @@ __device__ __forceinline__ uint64_t d_splitmix64(uint64_t x) {
 /* digit extraction.  mode 0: radix digit of the sign-biased key (sort);
  * mode 1: TOP bits of splitmix64(key) (the GPU HashPartitioner,
  * dpark/dependency.py :: HashPartitioner equivalent [U] — layout freedom
- * per SURVEY.md §8(a2)). */
+ * per SURVEY.md §8(a2)); mode 2: range partition over sorted splitters
+ * (dpark/dependency.py :: RangePartitioner [U], bisect_left semantics:
+ * rank = #{j : splitters[j] <= key}... bisect_left gives
+ * #{j : splitters[j] < key} + (key present ? its first index) — we use
+ * rank = #{j : key >= splitters[j]}, which equals bisect_right; any
+ * consistent monotone rule yields a valid total order across ranks). */
+__device__ __forceinline__ uint32_t d_digit_s(
+    int64_t key, int mode, int shift, uint32_t mask,
+    const int64_t *__restrict__ splitters, int nsplit) {
+  if (mode == 2) {
+    uint32_t r = 0;
+    for (int j = 0; j < nsplit; j++) r += (key >= splitters[j]) ? 1u : 0u;
+    return r;
+  }
+  uint64_t u = mode ? d_splitmix64((uint64_t)key) : ((uint64_t)key ^ SIGNFLIP);
+  return (uint32_t)(u >> shift) & mask;
+}
+
 __device__ __forceinline__ uint32_t d_digit(int64_t key, int mode, int shift,
                                             uint32_t mask) {
   uint64_t u = mode ? d_splitmix64((uint64_t)key) : ((uint64_t)key ^ SIGNFLIP);
   return (uint32_t)(u >> shift) & mask;
+}
+
+/* strided sample of the current keys (sortByKey splitter sampling —
+ * dpark samples partitions to the driver [U]) */
+__global__ __launch_bounds__(NTHREADS) void k_sample(
+    const int64_t *__restrict__ keys, uint64_t n, uint64_t stride_elems,
+    int64_t *__restrict__ out, uint64_t n_out) {
+  uint64_t gstride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < n_out; i += gstride)
+    out[i] = keys[min(i * stride_elems, n - 1)];
 }
 
 /* ------------------------------------------------------------------ */
@@ -341,7 +369,7 @@ __global__ __launch_bounds__(NTHREADS) void k_minmax_or(
 __global__ __launch_bounds__(NTHREADS) void k_tile_hist(
     const int64_t *__restrict__ keys, uint64_t n,
     uint32_t *__restrict__ tile_hist, uint64_t ntiles, int mode, int shift,
-    uint32_t mask) {
+    uint32_t mask, const int64_t *__restrict__ splitters, int nsplit) {
   __shared__ uint32_t h[NTHREADS / 64][RADIX];
   for (uint64_t t = blockIdx.x; t < ntiles; t += gridDim.x) {
     int wave = threadIdx.x / 64;
@@ -354,11 +382,11 @@ __global__ __launch_bounds__(NTHREADS) void k_tile_hist(
      * 16-B rate on gfx950 — MI355X_MICROARCH.md §visibility table) */
     for (uint64_t i = start + 2 * threadIdx.x; i + 1 < end; i += 2 * NTHREADS) {
       longlong2 kk = *reinterpret_cast<const longlong2 *>(&keys[i]);
-      atomicAdd(&h[wave][d_digit(kk.x, mode, shift, mask)], 1u);
-      atomicAdd(&h[wave][d_digit(kk.y, mode, shift, mask)], 1u);
+      atomicAdd(&h[wave][d_digit_s(kk.x, mode, shift, mask, splitters, nsplit)], 1u);
+      atomicAdd(&h[wave][d_digit_s(kk.y, mode, shift, mask, splitters, nsplit)], 1u);
     }
     if (end == n && (n & 1) && threadIdx.x == 0 && end - start > 0)
-      atomicAdd(&h[wave][d_digit(keys[end - 1], mode, shift, mask)], 1u);
+      atomicAdd(&h[wave][d_digit_s(keys[end - 1], mode, shift, mask, splitters, nsplit)], 1u);
     __syncthreads();
     for (int d = threadIdx.x; d < RADIX; d += NTHREADS) {
       uint32_t s = 0;
@@ -394,10 +422,19 @@ __global__ __launch_bounds__(NTHREADS) void k_scan_mid(
   __shared__ uint64_t tot[RADIX];
   int d = threadIdx.x;
   uint64_t run = 0;
-  for (uint64_t s = 0; s < nseg; s++) {
-    uint32_t v = seg_sums[s * RADIX + d];
-    seg_sums[s * RADIX + d] = (uint32_t)run;
-    run += v;
+  /* register-blocked: 8 loads in flight before any store — the naive
+   * in-place loop serializes on (unprovable) store->load aliasing
+   * (measured 77 us for a single-WG pass over 96 segments) */
+  for (uint64_t s0 = 0; s0 < nseg; s0 += 8) {
+    uint32_t v[8];
+#pragma unroll
+    for (int j = 0; j < 8; j++)
+      v[j] = (s0 + j < nseg) ? seg_sums[(s0 + j) * RADIX + d] : 0;
+#pragma unroll
+    for (int j = 0; j < 8; j++) {
+      if (s0 + j < nseg) seg_sums[(s0 + j) * RADIX + d] = (uint32_t)run;
+      run += v[j];
+    }
   }
   tot[d] = run;
   if (row_tot) row_tot[d] = run;
@@ -412,8 +449,16 @@ __global__ __launch_bounds__(NTHREADS) void k_scan_mid(
     __syncthreads();
   }
   uint64_t base = sc[d] - tot[d]; /* inclusive -> exclusive */
-  for (uint64_t s = 0; s < nseg; s++)
-    seg_sums[s * RADIX + d] += (uint32_t)base;
+  for (uint64_t s0 = 0; s0 < nseg; s0 += 8) {
+    uint32_t v[8];
+#pragma unroll
+    for (int j = 0; j < 8; j++)
+      v[j] = (s0 + j < nseg) ? seg_sums[(s0 + j) * RADIX + d] : 0;
+#pragma unroll
+    for (int j = 0; j < 8; j++)
+      if (s0 + j < nseg)
+        seg_sums[(s0 + j) * RADIX + d] = v[j] + (uint32_t)base;
+  }
   /* NOTE: offsets fit u32 because n < 2^32 (enforced in dprk_reserve) */
 }
 
@@ -462,7 +507,8 @@ __global__ __launch_bounds__(NTHREADS_SC, 2) void k_scatter(
     int64_t *__restrict__ keys_out, double *__restrict__ vals_out,
     uint64_t n, const uint32_t *__restrict__ tile_hist,
     const uint32_t *__restrict__ tile_off, uint64_t ntiles, int mode,
-    int shift, uint32_t mask) {
+    int shift, uint32_t mask, const int64_t *__restrict__ splitters,
+    int nsplit) {
   constexpr int NW = NTHREADS_SC / 64;     /* 8 waves per WG              */
   constexpr int WBLK = TILE / NW;          /* 512 elements per wave       */
   constexpr int EPL = WBLK / 64;           /* 8 elements per lane         */
@@ -511,7 +557,7 @@ __global__ __launch_bounds__(NTHREADS_SC, 2) void k_scatter(
       if (idx < tile_n) {
         rk[i] = keys_in[start + idx];
         if (WITH_VALS) rv[i] = vals_in[start + idx];
-        rd[i] = d_digit(rk[i], mode, shift, mask);
+        rd[i] = d_digit_s(rk[i], mode, shift, mask, splitters, nsplit);
         atomicAdd(&h[wave][rd[i]], 1u);
       }
     }
@@ -554,7 +600,7 @@ __global__ __launch_bounds__(NTHREADS_SC, 2) void k_scatter(
     /* phase 4: write out in digit-sorted tile order */
     for (uint32_t idx = threadIdx.x; idx < tile_n; idx += NTHREADS_SC) {
       int64_t key = lk[idx];
-      uint32_t dig = d_digit(key, mode, shift, mask);
+      uint32_t dig = d_digit_s(key, mode, shift, mask, splitters, nsplit);
       uint32_t dst = goff[dig] + (idx - excl[dig]);
       keys_out[dst] = key;
       if (WITH_VALS) vals_out[dst] = lv[idx];
@@ -1346,7 +1392,8 @@ extern "C" int dprk_get_pairs(dprk_ctx *c, int64_t *keys, double *vals,
 
 /* ---- one radix pass: tile_hist -> 3-level scan -> scatter ---------- */
 static int radix_pass(dprk_ctx *c, int mode, int shift, uint32_t mask,
-                      int want_row_tot) {
+                      int want_row_tot, const int64_t *d_split = nullptr,
+                      int nsplit = 0) {
   uint64_t n = c->n;
   uint64_t ntiles = ceil_div(n, TILE);
   uint64_t seg_tiles = ceil_div(ntiles, RADIX); /* nseg <= 256 */
@@ -1356,7 +1403,7 @@ static int radix_pass(dprk_ctx *c, int mode, int shift, uint32_t mask,
   TIMED(c, K_TILEHIST,
         hipLaunchKernelGGL(k_tile_hist, dim3(thgrid), dim3(NTHREADS), 0,
                            c->stream, c->keys[src], n, c->tile_hist, ntiles,
-                           mode, shift, mask));
+                           mode, shift, mask, d_split, nsplit));
   TIMED(c, K_SCAN,
         hipLaunchKernelGGL(k_seg_sums, dim3((uint32_t)nseg), dim3(NTHREADS),
                            0, c->stream, c->tile_hist, ntiles, seg_tiles,
@@ -1374,18 +1421,78 @@ static int radix_pass(dprk_ctx *c, int mode, int shift, uint32_t mask,
           hipLaunchKernelGGL(k_scatter<true>, dim3(thgrid), dim3(NTHREADS_SC),
                              0, c->stream, c->keys[src], c->vals[src],
                              c->keys[dst], c->vals[dst], n, c->tile_hist,
-                             c->tile_off, ntiles, mode, shift, mask));
+                             c->tile_off, ntiles, mode, shift, mask,
+                             d_split, nsplit));
   } else {
     TIMED(c, K_SCATTER,
           hipLaunchKernelGGL(k_scatter<false>, dim3(thgrid), dim3(NTHREADS_SC),
                              0, c->stream, c->keys[src], nullptr,
                              c->keys[dst], nullptr, n, c->tile_hist,
-                             c->tile_off, ntiles, mode, shift, mask));
+                             c->tile_off, ntiles, mode, shift, mask,
+                             d_split, nsplit));
   }
   c->stats.bytes_scatter += n * (c->with_vals ? 32 : 16);
   c->stats.bytes_total += n * (c->with_vals ? 40 : 24); /* +hist read */
   c->cur = dst;
   return 0;
+}
+
+extern "C" int dprk_sample_keys(dprk_ctx *c, uint64_t n_samples,
+                                int64_t *out_host) {
+  if (!c) { set_err("null ctx"); return 1; }
+  HIP_CHECK(hipSetDevice(c->device));
+  if (c->n == 0 || n_samples == 0) return 0;
+  if (n_samples > c->n) n_samples = c->n;
+  uint64_t stride_elems = c->n / n_samples;
+  int64_t *d_out = nullptr;
+  HIP_CHECK(hipMalloc(&d_out, n_samples * 8));
+  TIMED(c, K_OTHER,
+        hipLaunchKernelGGL(k_sample, dim3(stream_grid(n_samples)),
+                           dim3(NTHREADS), 0, c->stream, c->keys[c->cur],
+                           c->n, stride_elems, d_out, n_samples));
+  HIP_CHECK(hipMemcpyAsync(out_host, d_out, n_samples * 8,
+                           hipMemcpyDeviceToHost, c->stream));
+  HIP_CHECK(hipStreamSynchronize(c->stream));
+  HIP_CHECK(hipFree(d_out));
+  return drain_timing(c);
+}
+
+/* range partition by sorted splitters (nranks-1 of them): bucket r =
+ * #{j : key >= splitters[j]} — the engine RangePartitioner
+ * (dpark/dependency.py :: RangePartitioner [U]); order-preserving across
+ * ranks, used by distributed sortByKey. */
+extern "C" int dprk_partition_range(dprk_ctx *c, const int64_t *splitters,
+                                    int nbuckets, uint64_t *bucket_counts) {
+  if (!c) { set_err("null ctx"); return 1; }
+  if (nbuckets < 1 || nbuckets > 256) {
+    set_err("partition_range: nbuckets must be in [1,256]");
+    return 1;
+  }
+  HIP_CHECK(hipSetDevice(c->device));
+  if (nbuckets == 1 || c->n == 0) {
+    if (bucket_counts) {
+      for (int b = 0; b < nbuckets; b++) bucket_counts[b] = 0;
+      if (nbuckets >= 1) bucket_counts[0] = c->n;
+    }
+    return 0;
+  }
+  int64_t *d_split = nullptr;
+  HIP_CHECK(hipMalloc(&d_split, (nbuckets - 1) * 8));
+  HIP_CHECK(hipMemcpyAsync(d_split, splitters, (nbuckets - 1) * 8,
+                           hipMemcpyHostToDevice, c->stream));
+  int rc = radix_pass(c, /*mode=*/2, 0, 0, 1, d_split, nbuckets - 1);
+  c->sorted = 0;
+  c->orand_valid = 0;
+  if (!rc && bucket_counts) {
+    uint64_t rt[RADIX];
+    HIP_CHECK(hipMemcpyAsync(rt, c->row_tot, RADIX * 8,
+                             hipMemcpyDeviceToHost, c->stream));
+    HIP_CHECK(hipStreamSynchronize(c->stream));
+    for (int b = 0; b < nbuckets; b++) bucket_counts[b] = rt[b];
+  }
+  int rc2 = drain_timing(c);
+  HIP_CHECK(hipFree(d_split));
+  return rc ? rc : rc2;
 }
 
 extern "C" int dprk_sort_pairs(dprk_ctx *c) {

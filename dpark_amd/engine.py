@@ -55,7 +55,8 @@ SYMBOLS = [
     "dprk_init", "dprk_free", "dprk_last_error", "dprk_build_info",
     "dprk_reserve", "dprk_gen_pairs", "dprk_set_alias_table",
     "dprk_put_pairs", "dprk_get_pairs",
-    "dprk_count", "dprk_partition", "dprk_sort_pairs",
+    "dprk_count", "dprk_partition", "dprk_partition_range",
+    "dprk_sample_keys", "dprk_sort_pairs",
     "dprk_reduce_by_key_sum", "dprk_group_by_key", "dprk_get_groups",
     "dprk_join", "dprk_get_join", "dprk_comm_unique_id", "dprk_comm_init",
     "dprk_exchange", "dprk_stats", "dprk_stats_reset",
@@ -98,6 +99,9 @@ def load_library():
                                    ctypes.POINTER(u64)]
     lib.dprk_count.argtypes = [p, ctypes.POINTER(u64)]
     lib.dprk_partition.argtypes = [p, ctypes.c_int, ctypes.POINTER(u64)]
+    lib.dprk_partition_range.argtypes = [p, ctypes.c_void_p, ctypes.c_int,
+                                         ctypes.POINTER(u64)]
+    lib.dprk_sample_keys.argtypes = [p, u64, ctypes.c_void_p]
     lib.dprk_sort_pairs.argtypes = [p]
     lib.dprk_reduce_by_key_sum.argtypes = [p, ctypes.POINTER(u64)]
     lib.dprk_group_by_key.argtypes = [p, ctypes.POINTER(u64)]
@@ -200,6 +204,25 @@ class Engine:
         counts = (ctypes.c_uint64 * nbuckets)()
         _check(self._lib, self._lib.dprk_partition(self._ctx, nbuckets, counts),
                "dprk_partition")
+        return np.array(counts, dtype=np.uint64)
+
+    def sample_keys(self, n_samples):
+        out = np.empty(n_samples, dtype=np.int64)
+        _check(self._lib,
+               self._lib.dprk_sample_keys(self._ctx, n_samples,
+                                          out.ctypes.data_as(ctypes.c_void_p)),
+               "dprk_sample_keys")
+        return out
+
+    def partition_range(self, splitters, nbuckets):
+        splitters = np.ascontiguousarray(splitters, dtype=np.int64)
+        assert len(splitters) == nbuckets - 1
+        counts = (ctypes.c_uint64 * nbuckets)()
+        _check(self._lib,
+               self._lib.dprk_partition_range(
+                   self._ctx, splitters.ctypes.data_as(ctypes.c_void_p),
+                   nbuckets, counts),
+               "dprk_partition_range")
         return np.array(counts, dtype=np.uint64)
 
     def sort(self):

@@ -75,6 +75,16 @@ int dprk_count(dprk_ctx *ctx, uint64_t *n_out);
  * sizes.  nbuckets must be a power of two <= 256. */
 int dprk_partition(dprk_ctx *ctx, int nbuckets, uint64_t *bucket_counts);
 
+/* Range partition by nbuckets-1 sorted splitters (bucket r = #{j : key >=
+ * splitters[j]}) — the engine RangePartitioner (dpark/dependency.py ::
+ * RangePartitioner [U]); order-preserving across ranks, used by the
+ * distributed sortByKey (splitters from dprk_sample_keys + host merge,
+ * mirroring dpark's driver-side sampling). */
+int dprk_partition_range(dprk_ctx *ctx, const int64_t *splitters,
+                         int nbuckets, uint64_t *bucket_counts);
+/* Evenly strided sample of the current keys (D2H). */
+int dprk_sample_keys(dprk_ctx *ctx, uint64_t n_samples, int64_t *out_host);
+
 /* LSD radix sort of the current dataset by key, ascending (int64 order),
  * values carried.  Passes over constant digit positions are skipped (the
  * pass count actually executed is reported via dprk_stats). */

@@ -353,3 +353,27 @@ def test_adversarial_inputs(eng, case):
     assert np.array_equal(np.isnan(v), np.isnan(wv))
     assert np.array_equal(np.isposinf(v), np.isposinf(wv))
     assert np.array_equal(np.isneginf(v), np.isneginf(wv))
+
+
+def test_partition_range(eng):
+    """Device RangePartitioner: counts + order-preserving segments match
+    the splitter semantics (#{j: key >= splitters[j]})."""
+    keys, vals = gen_pairs(41, 1_000_000, 0)
+    _put(eng, keys, vals)
+    sample = eng.sample_keys(4096)
+    assert len(sample) == 4096
+    allsamp = np.sort(sample)
+    nb = 8
+    splitters = np.array([allsamp[int(len(allsamp) / nb * r)]
+                          for r in range(1, nb)], dtype=np.int64)
+    counts = eng.partition_range(splitters, nb)
+    assert counts.sum() == len(keys)
+    k, v = eng.get()
+    assert np.array_equal(np.sort(k), np.sort(keys))  # permutation
+    ranks = np.searchsorted(splitters, k, side="right")
+    off = 0
+    for r, c in enumerate(counts):
+        assert np.all(ranks[off:off + int(c)] == r), f"bucket {r}"
+        off += int(c)
+    # reasonably balanced with 4096 samples over uniform keys
+    assert counts.max() / max(counts.min(), 1) < 1.3
