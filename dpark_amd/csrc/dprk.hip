@@ -379,14 +379,29 @@ __global__ __launch_bounds__(NTHREADS) void k_tile_hist(
     uint64_t start = t * (uint64_t)TILE;
     uint64_t end = min(start + TILE, n);
     /* 16-B loads: two keys per lane (8-B streams run at 0.54-0.70x the
-     * 16-B rate on gfx950 — MI355X_MICROARCH.md §visibility table) */
-    for (uint64_t i = start + 2 * threadIdx.x; i + 1 < end; i += 2 * NTHREADS) {
-      longlong2 kk = *reinterpret_cast<const longlong2 *>(&keys[i]);
-      atomicAdd(&h[wave][d_digit_s(kk.x, mode, shift, mask, splitters, nsplit)], 1u);
-      atomicAdd(&h[wave][d_digit_s(kk.y, mode, shift, mask, splitters, nsplit)], 1u);
+     * 16-B rate on gfx950).  Full tiles take the constant-trip unrolled
+     * path so all 8 loads issue up front (the variable-bound loop kept
+     * only one load in flight: 95% wave-parked in the SQ profile). */
+    if (end - start == TILE) {
+      longlong2 kk[TILE / (2 * NTHREADS)];
+#pragma unroll
+      for (int j = 0; j < TILE / (2 * NTHREADS); j++)
+        kk[j] = *reinterpret_cast<const longlong2 *>(
+            &keys[start + 2 * threadIdx.x + (uint64_t)j * 2 * NTHREADS]);
+#pragma unroll
+      for (int j = 0; j < TILE / (2 * NTHREADS); j++) {
+        atomicAdd(&h[wave][d_digit_s(kk[j].x, mode, shift, mask, splitters, nsplit)], 1u);
+        atomicAdd(&h[wave][d_digit_s(kk[j].y, mode, shift, mask, splitters, nsplit)], 1u);
+      }
+    } else {
+      for (uint64_t i = start + 2 * threadIdx.x; i + 1 < end; i += 2 * NTHREADS) {
+        longlong2 kk = *reinterpret_cast<const longlong2 *>(&keys[i]);
+        atomicAdd(&h[wave][d_digit_s(kk.x, mode, shift, mask, splitters, nsplit)], 1u);
+        atomicAdd(&h[wave][d_digit_s(kk.y, mode, shift, mask, splitters, nsplit)], 1u);
+      }
+      if (end == n && (n & 1) && threadIdx.x == 0 && end - start > 0)
+        atomicAdd(&h[wave][d_digit_s(keys[end - 1], mode, shift, mask, splitters, nsplit)], 1u);
     }
-    if (end == n && (n & 1) && threadIdx.x == 0 && end - start > 0)
-      atomicAdd(&h[wave][d_digit_s(keys[end - 1], mode, shift, mask, splitters, nsplit)], 1u);
     __syncthreads();
     for (int d = threadIdx.x; d < RADIX; d += NTHREADS) {
       uint32_t s = 0;
