@@ -33,9 +33,13 @@
 /* ------------------------------------------------------------------ */
 #define RADIX 256
 #define RADIX_BITS 8
-#define TILE 4096          /* elements per scatter/reduce workgroup      */
+#ifndef TILE
+#define TILE 4096          /* elements per radix-pass workgroup (tunable) */
+#endif
 #define NTHREADS 256       /* 4 waves of 64                              */
-#define NTHREADS_SC 512    /* scatter workgroup: 8 waves                 */
+#ifndef NTHREADS_SC
+#define NTHREADS_SC 512    /* scatter workgroup: 8 waves (tunable)       */
+#endif
 #define RTILE 2048         /* elements per reduce/head workgroup (4 WG/CU
                               occupancy: the reduce is latency-bound)    */
 #define EPT_R (RTILE / NTHREADS) /* 8 consecutive elements per thread    */
@@ -531,8 +535,14 @@ __global__ __launch_bounds__(NTHREADS_SC, 2) void k_scatter(
   __shared__ double lv[WITH_VALS ? TILE : 1];
   __shared__ uint32_t excl[RADIX];     /* tile-local digit excl scan  */
   __shared__ uint32_t goff[RADIX];     /* global write base per digit */
+#ifdef DPRK_PACKED_H
+  /* two u16 counters per word (digits 2d, 2d+1): halves the per-wave
+   * histogram LDS so 16-wave workgroups still fit 2 blocks/CU */
+  __shared__ uint32_t h[NW][RADIX / 2];
+#else
   __shared__ uint32_t h[NW][RADIX];    /* per-wave hist -> wave bases
                                           -> running counters (in place) */
+#endif
   __shared__ uint32_t wtot[NW > 4 ? NW : 4]; /* cross-wave scan carries  */
 
   const int wave = threadIdx.x / 64, lane = threadIdx.x % 64;
@@ -554,7 +564,12 @@ __global__ __launch_bounds__(NTHREADS_SC, 2) void k_scatter(
       if (lane == 63) wtot[wave] = inc;
       excl[threadIdx.x] = inc - myv; /* within-wave exclusive */
     }
+#ifdef DPRK_PACKED_H
+    for (int j = threadIdx.x; j < NW * RADIX / 2; j += NTHREADS_SC)
+      ((uint32_t *)h)[j] = 0;
+#else
     for (int j = 0; j < RADIX / 64; j++) h[wave][lane * (RADIX / 64) + j] = 0;
+#endif
     __syncthreads();
     if (threadIdx.x < RADIX && wave > 0) {
       uint32_t add = 0;
@@ -573,12 +588,28 @@ __global__ __launch_bounds__(NTHREADS_SC, 2) void k_scatter(
         rk[i] = keys_in[start + idx];
         if (WITH_VALS) rv[i] = vals_in[start + idx];
         rd[i] = d_digit_s(rk[i], mode, shift, mask, splitters, nsplit);
+#ifdef DPRK_PACKED_H
+        atomicAdd(&h[wave][rd[i] >> 1], 1u << (16 * (rd[i] & 1)));
+#else
         atomicAdd(&h[wave][rd[i]], 1u);
+#endif
       }
     }
     __syncthreads();
 
-    /* phase 2: per-wave digit bases in place (thread=digit) */
+    /* phase 2: per-wave digit bases in place */
+#ifdef DPRK_PACKED_H
+    if (threadIdx.x < RADIX / 2) {      /* thread = digit pair */
+      uint32_t d = threadIdx.x;
+      uint32_t b0 = excl[2 * d], b1 = excl[2 * d + 1];
+      for (int w = 0; w < NW; w++) {
+        uint32_t word = h[w][d];
+        h[w][d] = (b0 & 0xFFFFu) | (b1 << 16);
+        b0 += word & 0xFFFFu;
+        b1 += word >> 16;
+      }
+    }
+#else
     if (threadIdx.x < RADIX) {
       uint32_t d = threadIdx.x;
       uint32_t b = excl[d];
@@ -588,6 +619,7 @@ __global__ __launch_bounds__(NTHREADS_SC, 2) void k_scatter(
         b += cnt;
       }
     }
+#endif
     __syncthreads();
 
     /* phase 3: stable rank (wave-local, no barriers) + LDS scatter */
@@ -602,9 +634,17 @@ __global__ __launch_bounds__(NTHREADS_SC, 2) void k_scatter(
           same &= ((dig >> b) & 1) ? bb : ~bb;
         }
         uint32_t rank_in_wave = (uint32_t)__popcll(same & lowmask);
+#ifdef DPRK_PACKED_H
+        uint32_t word = h[wave][dig >> 1];        /* read before leader  */
+        uint32_t base = (word >> (16 * (dig & 1))) & 0xFFFFu;
+        if (rank_in_wave == 0)                    /* leader advances     */
+          atomicAdd(&h[wave][dig >> 1],
+                    ((uint32_t)__popcll(same)) << (16 * (dig & 1)));
+#else
         uint32_t base = h[wave][dig];             /* broadcast LDS read  */
         if (rank_in_wave == 0)                    /* leader advances     */
           h[wave][dig] = base + (uint32_t)__popcll(same);
+#endif
         uint32_t pos = base + rank_in_wave;
         lk[pos] = rk[i];
         if (WITH_VALS) lv[pos] = rv[i];
