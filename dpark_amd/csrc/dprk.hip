@@ -693,18 +693,39 @@ __global__ __launch_bounds__(NTHREADS, 2) void k_reduce_tiles(
     const uint64_t start = t * (uint64_t)RTILE;
     const uint64_t end = min(start + RTILE, n);
     const uint32_t tile_n = (uint32_t)(end - start);
-    /* 16-B staging loads (8-B streams run at ~0.6x the 16-B rate) */
-    for (uint32_t i = 2 * threadIdx.x; i + 1 < tile_n; i += 2 * NTHREADS) {
-      longlong2 kk = *reinterpret_cast<const longlong2 *>(&keys[start + i]);
-      double2 vv = *reinterpret_cast<const double2 *>(&vals[start + i]);
-      lk[PIDX(i)] = kk.x;
-      lk[PIDX(i + 1)] = kk.y;
-      lv[PIDX(i)] = vv.x;
-      lv[PIDX(i + 1)] = vv.y;
-    }
-    if ((tile_n & 1) && threadIdx.x == 0) {
-      lk[PIDX(tile_n - 1)] = keys[start + tile_n - 1];
-      lv[PIDX(tile_n - 1)] = vals[start + tile_n - 1];
+    /* 16-B staging loads; full tiles take a constant-trip unrolled path
+     * so all loads issue up front (variable-bound loops keep ~1 load in
+     * flight; the kernel was 69% wave-parked) */
+    if (tile_n == RTILE) {
+      longlong2 kk[RTILE / (2 * NTHREADS)];
+      double2 vv[RTILE / (2 * NTHREADS)];
+#pragma unroll
+      for (int j = 0; j < RTILE / (2 * NTHREADS); j++) {
+        uint32_t i = 2 * threadIdx.x + (uint32_t)j * 2 * NTHREADS;
+        kk[j] = *reinterpret_cast<const longlong2 *>(&keys[start + i]);
+        vv[j] = *reinterpret_cast<const double2 *>(&vals[start + i]);
+      }
+#pragma unroll
+      for (int j = 0; j < RTILE / (2 * NTHREADS); j++) {
+        uint32_t i = 2 * threadIdx.x + (uint32_t)j * 2 * NTHREADS;
+        lk[PIDX(i)] = kk[j].x;
+        lk[PIDX(i + 1)] = kk[j].y;
+        lv[PIDX(i)] = vv[j].x;
+        lv[PIDX(i + 1)] = vv[j].y;
+      }
+    } else {
+      for (uint32_t i = 2 * threadIdx.x; i + 1 < tile_n; i += 2 * NTHREADS) {
+        longlong2 kk = *reinterpret_cast<const longlong2 *>(&keys[start + i]);
+        double2 vv = *reinterpret_cast<const double2 *>(&vals[start + i]);
+        lk[PIDX(i)] = kk.x;
+        lk[PIDX(i + 1)] = kk.y;
+        lv[PIDX(i)] = vv.x;
+        lv[PIDX(i + 1)] = vv.y;
+      }
+      if ((tile_n & 1) && threadIdx.x == 0) {
+        lk[PIDX(tile_n - 1)] = keys[start + tile_n - 1];
+        lv[PIDX(tile_n - 1)] = vals[start + tile_n - 1];
+      }
     }
     if (threadIdx.x == 0)
       lk[PIDX(tile_n)] = (end < n) ? keys[end] : 0; /* lookahead (unused if end==n) */

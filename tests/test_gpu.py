@@ -377,3 +377,33 @@ def test_partition_range(eng):
         off += int(c)
     # reasonably balanced with 4096 samples over uniform keys
     assert counts.max() / max(counts.min(), 1) < 1.3
+
+
+def test_error_paths():
+    """C-ABI error discipline: clear failures, no crashes (FetchFailed-
+    style exception surface)."""
+    from dpark_amd.engine import EngineError
+    e = Engine(device=0)
+    try:
+        e.reserve(1000, with_vals=False)
+        with pytest.raises(EngineError, match="cap"):
+            e.put(np.arange(2000, dtype=np.int64))
+        with pytest.raises(EngineError, match="keys-only"):
+            e.put(np.arange(10, dtype=np.int64), np.zeros(10))
+        e.put(np.arange(100, dtype=np.int64))
+        with pytest.raises(EngineError, match="keys-only"):
+            e.reduce_by_key_sum()
+        with pytest.raises(EngineError, match="power of two"):
+            e.partition(3)
+        with pytest.raises(EngineError, match="alias"):
+            e.gen_pairs(seed=1, n=10, key_universe=1 << 8, dist=1)
+        with pytest.raises(EngineError, match="unknown dist"):
+            e.gen_pairs(seed=1, n=10, key_universe=1 << 8, dist=7)
+        with pytest.raises(EngineError):
+            e.reserve(1 << 33, with_vals=False)  # > 2^32: out-of-core is next-row
+        # ctx still usable after errors
+        e.reserve(1000, with_vals=True)
+        e.put(np.arange(100, dtype=np.int64), np.ones(100))
+        assert e.reduce_by_key_sum() == 100
+    finally:
+        e.close()
