@@ -104,7 +104,12 @@ def main():
         dist = _dist
         dist.init_process_group(backend="gloo")  # control plane only
 
-    from dpark_amd.engine import Engine  # fails loudly if .so missing
+    if os.environ.get("DPRK_BENCH_SIM"):
+        # CPU control-plane dry run (tests only): exercises the exact
+        # torchrun launch path with no GPU; output is marked SIMULATED.
+        from dpark_amd.testing import SimEngine as Engine
+    else:
+        from dpark_amd.engine import Engine  # fails loudly if .so missing
 
     defaults = {"reduce": N_PAIRS_PER_GPU, "sort": 1_000_000_000,
                 "join": 500_000_000, "zipf": 1_000_000_000}
@@ -275,7 +280,7 @@ def main():
         "scaling": "weak",
         "vs_baseline": None,  # no published reference number
         "dtype": "int64+f64",
-        "data": "synthetic",
+        "data": ("SIMULATED-no-gpu" if os.environ.get("DPRK_BENCH_SIM") else "synthetic"),
         "config": {
             "workload": {
                 "reduce": "reduceByKey-100M-int64-f64-uniform-2^24-per-gpu "
