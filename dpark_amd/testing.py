@@ -80,10 +80,7 @@ class SimEngine:
         return len(uk)
 
     def join_count(self, other):
-        common, ia, ib = np.intersect1d(np.unique(self.keys),
-                                        np.unique(other.keys),
-                                        return_indices=True)
-        ca = np.bincount(np.searchsorted(np.unique(self.keys), self.keys))
+        common = np.intersect1d(np.unique(self.keys), np.unique(other.keys))
         return int(len(common))  # control-flow only
 
     def get(self, with_vals=True):

@@ -41,7 +41,6 @@ _ADD_FUNCS = None
 def _is_add(f) -> bool:
     global _ADD_FUNCS
     if _ADD_FUNCS is None:
-        import builtins  # noqa
         _ADD_FUNCS = {operator.add, operator.iadd}
     return f in _ADD_FUNCS
 
