@@ -426,3 +426,26 @@ def test_soak_repeatability(eng):
         else:
             assert np.array_equal(k, ref_k)
             assert np.array_equal(v, ref_v)
+
+
+def test_fused_reduce_opt_in():
+    """The fused last-pass reduce (DPRK_FUSED_REDUCE) matches the oracle
+    and the base path's keys exactly; sums within 1e-6 (both paths are
+    deterministic; trees differ in last ulp, which is why the path is an
+    explicit opt-in, never auto-selected)."""
+    os.environ["DPRK_FUSED_REDUCE"] = "1"
+    try:
+        e = Engine(device=0)
+        for n, uni, seed in [(2_000_000, 1 << 18, 61), (500_000, 1 << 8, 62),
+                             (4096, 1, 63), (100_000, 0, 64)]:
+            keys, vals = gen_pairs(seed, n, uni if uni else 1 << 64)
+            e.reserve(n, with_vals=True)
+            e.put(keys, vals)
+            got = e.reduce_by_key_sum()
+            k, v = e.get()
+            wk, wv = so.reduce_by_key_sum(keys, vals)
+            assert got == len(wk) and np.array_equal(k, wk), (n, uni)
+            np.testing.assert_allclose(v, wv, rtol=RTOL)
+        e.close()
+    finally:
+        del os.environ["DPRK_FUSED_REDUCE"]
