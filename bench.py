@@ -155,7 +155,15 @@ def main():
                    for _ in range(nranks)]
             dist.all_gather(mat, counts)              # count matrix (tiny)
             recv = [int(mat[r][rank]) for r in range(nranks)]
+            # report exchange imbalance from the count
+            # matrix (max over ranks of received total / mean)
+            totals = [sum(int(mat[s][r]) for s in range(nranks))
+                      for r in range(nranks)]
+            imbalance["max"] = max(imbalance.get("max", 0.0),
+                                   max(totals) / (sum(totals) / nranks))
             e.exchange([int(x) for x in send], recv)
+
+    imbalance = {}
 
     def step(i):
         # regenerate the resident input (seed fixed per rank; deterministic)
@@ -301,6 +309,8 @@ def main():
             "step_includes_gen_ms": round(gen_ms, 3),
             "pipeline_ms_per_step": round(pipeline_ms, 3),
             "sort_passes": int(s["sort_passes"]),
+            "exchange_imbalance": (round(imbalance["max"], 4)
+                                   if imbalance else None),
             "engine": "libdprk.so (HIP gfx950)",
         },
         "roofline": roofline,
