@@ -117,6 +117,7 @@ struct dprk_ctx {
    * permutation-only ops */
   int orand_valid = 0;
   uint64_t known_or = 0, known_and = 0;
+  int hist_shift_valid = -1; /* tile_hist currently holds this pass shift */
 
 
   /* radix scratch */
@@ -310,6 +311,60 @@ __global__ __launch_bounds__(NTHREADS) void k_gen(
         vals[i] = (double)(v0 >> 11) * (1.0 / 9007199254740992.0);
       }
     }
+  }
+}
+
+/* tile-structured generator that also produces the per-tile digit-0
+ * histogram (tile-major) — the first radix pass then skips its
+ * k_tile_hist read (~4% of the config-2 step).  One WG per tile. */
+__global__ __launch_bounds__(NTHREADS) void k_gen_hist(
+    int64_t *__restrict__ keys, double *__restrict__ vals, uint64_t n,
+    uint64_t seed_hi, uint64_t keymask, uint32_t *__restrict__ tile_hist,
+    uint64_t ntiles, int shift) {
+  __shared__ uint32_t h[NTHREADS / 64][RADIX];
+  const int wave = threadIdx.x / 64;
+  for (uint64_t t = blockIdx.x; t < ntiles; t += gridDim.x) {
+    for (int j = threadIdx.x; j < (NTHREADS / 64) * RADIX; j += NTHREADS)
+      ((uint32_t *)h)[j] = 0;
+    __syncthreads();
+    uint64_t start = t * (uint64_t)TILE;
+    uint64_t end = min(start + TILE, n);
+    for (uint64_t i = start + 2 * threadIdx.x; i + 1 < end;
+         i += 2 * NTHREADS) {
+      uint64_t x0 = seed_hi + i, x1 = seed_hi + i + 1;
+      longlong2 kk;
+      kk.x = (int64_t)(d_splitmix64(x0) & keymask);
+      kk.y = (int64_t)(d_splitmix64(x1) & keymask);
+      *reinterpret_cast<longlong2 *>(&keys[i]) = kk;
+      atomicAdd(&h[wave][d_digit(kk.x, 0, shift, 0xFF)], 1u);
+      atomicAdd(&h[wave][d_digit(kk.y, 0, shift, 0xFF)], 1u);
+      if (vals) {
+        double2 vv;
+        uint64_t v0 = d_splitmix64(x0 ^ 0xDEADBEEFCAFEBABEull);
+        uint64_t v1 = d_splitmix64(x1 ^ 0xDEADBEEFCAFEBABEull);
+        vv.x = (double)(v0 >> 11) * (1.0 / 9007199254740992.0);
+        vv.y = (double)(v1 >> 11) * (1.0 / 9007199254740992.0);
+        *reinterpret_cast<double2 *>(&vals[i]) = vv;
+      }
+    }
+    if ((end - start) & 1 && threadIdx.x == 0) {
+      uint64_t i = end - 1;
+      uint64_t x = seed_hi + i;
+      int64_t k = (int64_t)(d_splitmix64(x) & keymask);
+      keys[i] = k;
+      atomicAdd(&h[wave][d_digit(k, 0, shift, 0xFF)], 1u);
+      if (vals) {
+        uint64_t vb = d_splitmix64(x ^ 0xDEADBEEFCAFEBABEull);
+        vals[i] = (double)(vb >> 11) * (1.0 / 9007199254740992.0);
+      }
+    }
+    __syncthreads();
+    for (int d = threadIdx.x; d < RADIX; d += NTHREADS) {
+      uint32_t s = 0;
+      for (int w = 0; w < NTHREADS / 64; w++) s += h[w][d];
+      tile_hist[t * RADIX + d] = s;
+    }
+    __syncthreads();
   }
 }
 
@@ -1660,18 +1715,31 @@ extern "C" int dprk_gen_pairs(dprk_ctx *c, uint64_t seed, uint64_t n,
   HIP_CHECK(hipSetDevice(c->device));
   uint64_t keymask =
       (key_universe == 0) ? ~0ull : (key_universe - 1); /* power of two */
+  c->hist_shift_valid = -1;
   if (dist == 1) {
     TIMED(c, K_GEN,
           hipLaunchKernelGGL(k_gen_alias, dim3(stream_grid(n)), dim3(NTHREADS),
                              0, c->stream, c->keys[c->cur],
                              c->with_vals ? c->vals[c->cur] : nullptr, n,
                              seed << 32, c->alias_table, key_universe));
-  } else
-  TIMED(c, K_GEN,
-        hipLaunchKernelGGL(k_gen, dim3(stream_grid(n)), dim3(NTHREADS), 0,
-                           c->stream, c->keys[c->cur],
-                           c->with_vals ? c->vals[c->cur] : nullptr, n,
-                           seed << 32, keymask));
+  } else {
+    /* fused gen+hist: the first sort pass's digit shift for masked keys
+     * is the lowest nonzero mask byte (= 0 unless the low byte is
+     * masked out); full-range keys also start at shift 0 */
+    int shift0 = 0;
+    while (keymask != ~0ull && ((keymask >> shift0) & 0xFF) == 0 &&
+           shift0 < 56)
+      shift0 += 8;
+    uint64_t ntiles = ceil_div(n, (uint64_t)TILE);
+    TIMED(c, K_GEN,
+          hipLaunchKernelGGL(k_gen_hist,
+                             dim3((uint32_t)min(ntiles, (uint64_t)2048)),
+                             dim3(NTHREADS), 0, c->stream, c->keys[c->cur],
+                             c->with_vals ? c->vals[c->cur] : nullptr, n,
+                             seed << 32, keymask, c->tile_hist, ntiles,
+                             shift0));
+    c->hist_shift_valid = shift0;
+  }
   c->n = n;
   c->sorted = 0;
   /* keys = splitmix64 & keymask, so biased keys lie in
@@ -1706,6 +1774,7 @@ extern "C" int dprk_put_pairs(dprk_ctx *c, const int64_t *keys,
   c->n = n;
   c->sorted = 0;
   c->orand_valid = 0;
+  c->hist_shift_valid = -1;
   return 0;
 }
 
@@ -1737,10 +1806,14 @@ static int radix_pass(dprk_ctx *c, int mode, int shift, uint32_t mask,
   uint64_t nseg = ceil_div(ntiles, seg_tiles);
   int src = c->cur, dst = c->cur ^ 1;
   int thgrid = (int)min(ntiles, (uint64_t)2048);
-  TIMED(c, K_TILEHIST,
-        hipLaunchKernelGGL(k_tile_hist, dim3(thgrid), dim3(NTHREADS), 0,
-                           c->stream, c->keys[src], n, c->tile_hist, ntiles,
-                           mode, shift, mask, d_split, nsplit));
+  int hist_ran = !(mode == 0 && c->hist_shift_valid == shift);
+  if (hist_ran) {
+    TIMED(c, K_TILEHIST,
+          hipLaunchKernelGGL(k_tile_hist, dim3(thgrid), dim3(NTHREADS), 0,
+                             c->stream, c->keys[src], n, c->tile_hist,
+                             ntiles, mode, shift, mask, d_split, nsplit));
+  }
+  c->hist_shift_valid = -1; /* consumed; scatter reorders the data */
   TIMED(c, K_SCAN,
         hipLaunchKernelGGL(k_seg_sums, dim3((uint32_t)nseg), dim3(NTHREADS),
                            0, c->stream, c->tile_hist, ntiles, seg_tiles,
@@ -1769,7 +1842,7 @@ static int radix_pass(dprk_ctx *c, int mode, int shift, uint32_t mask,
                              d_split, nsplit));
   }
   c->stats.bytes_scatter += n * (c->with_vals ? 32 : 16);
-  c->stats.bytes_total += n * (c->with_vals ? 40 : 24); /* +hist read */
+  c->stats.bytes_total += n * ((c->with_vals ? 32 : 16) + (hist_ran ? 8 : 0));
   c->cur = dst;
   return 0;
 }
@@ -1944,10 +2017,13 @@ static int fused_reduce_last_pass(dprk_ctx *c, int shift, uint64_t *n_out) {
   int src = c->cur, dst = c->cur ^ 1;
   int grid = (int)min(ntiles, (uint64_t)2048);
   if (ensure_fused(c, ceil_div(c->cap, (uint64_t)TILE) * RADIX)) return 1;
-  TIMED(c, K_TILEHIST,
-        hipLaunchKernelGGL(k_tile_hist, dim3(grid), dim3(NTHREADS), 0,
-                           c->stream, c->keys[src], n, c->tile_hist, ntiles,
-                           0, shift, 0xFF, nullptr, 0));
+  if (c->hist_shift_valid != shift) {
+    TIMED(c, K_TILEHIST,
+          hipLaunchKernelGGL(k_tile_hist, dim3(grid), dim3(NTHREADS), 0,
+                             c->stream, c->keys[src], n, c->tile_hist,
+                             ntiles, 0, shift, 0xFF, nullptr, 0));
+  }
+  c->hist_shift_valid = -1;
   TIMED(c, K_REDUCE,
         hipLaunchKernelGGL(k_scatter_reduce<true>, dim3(grid),
                            dim3(NTHREADS_SC), 0, c->stream, c->keys[src],
@@ -2324,6 +2400,7 @@ extern "C" int dprk_exchange(dprk_ctx *c, const uint64_t *send_counts,
   c->n = total_recv;
   c->sorted = 0;
   c->orand_valid = 0;
+  c->hist_shift_valid = -1;
   return drain_timing(c);
 }
 
