@@ -148,20 +148,26 @@ def main():
             dist.barrier()
 
     def exchange_if_dist(e):
+        # Balanced bucket exchange: partition into 256
+        # hash buckets, all-gather the count matrix, deterministically
+        # greedy-assign buckets to ranks (hot buckets stay whole but get
+        # a rank largely to themselves), then one bucket-granular RCCL
+        # all-to-allv.  reduceByKey correctness is untouched: every
+        # bucket (hence every key) lands wholly on one rank.
         if nranks > 1:
-            send = e.partition(nranks)                # rank-digit layout
+            from dpark_amd.distshuffle import greedy_assign, recv_matrix_for
+            nbuckets = 256
+            send = e.partition(nbuckets)
             counts = torch.tensor(send.astype("int64"))
-            mat = [torch.zeros(nranks, dtype=torch.int64)
+            mat = [torch.zeros(nbuckets, dtype=torch.int64)
                    for _ in range(nranks)]
             dist.all_gather(mat, counts)              # count matrix (tiny)
-            recv = [int(mat[r][rank]) for r in range(nranks)]
-            # report exchange imbalance from the count
-            # matrix (max over ranks of received total / mean)
-            totals = [sum(int(mat[s][r]) for s in range(nranks))
-                      for r in range(nranks)]
+            m = np.stack([r.numpy() for r in mat]).astype(np.uint64)
+            bucket_rank, loads = greedy_assign(m.sum(axis=0), nranks)
             imbalance["max"] = max(imbalance.get("max", 0.0),
-                                   max(totals) / (sum(totals) / nranks))
-            e.exchange([int(x) for x in send], recv)
+                                   max(loads) / (sum(loads) / nranks))
+            rm = recv_matrix_for(m, bucket_rank, rank)
+            e.exchange_buckets(send, bucket_rank, rm, nranks)
 
     imbalance = {}
 

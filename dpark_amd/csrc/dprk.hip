@@ -2404,6 +2404,85 @@ extern "C" int dprk_exchange(dprk_ctx *c, const uint64_t *send_counts,
   return drain_timing(c);
 }
 
+/* Bucket-granular all-to-allv: the dataset laid out by
+ * dprk_partition(ctx, nbuckets) is exchanged under an arbitrary
+ * bucket->rank assignment — one grouped ncclSend/Recv per nonempty
+ * bucket, both sides iterating buckets ascending so point-to-point
+ * pairs match.  With the count matrix all-gathered on the host, a
+ * deterministic greedy assignment balances skewed buckets across ranks
+ * (SURVEY.md §8(e) hot buckets); each bucket stays whole, so reduceByKey
+ * needs no second combine.  Residual limit: a single bucket larger than
+ * a rank's capacity still fails (clearly) — splitting ONE key's partial
+ * sums across ranks is future work.
+ * recv_matrix[s*nbuckets + b] = elements source rank s sends this rank
+ * from bucket b (zero for buckets not assigned here). */
+extern "C" int dprk_exchange_buckets(dprk_ctx *c,
+                                     const uint64_t *bucket_counts,
+                                     const int *bucket_rank, int nbuckets,
+                                     const uint64_t *recv_matrix,
+                                     int nranks) {
+  if (!c) { set_err("null ctx"); return 1; }
+  if (!c->comm) { set_err("exchange_buckets: comm not initialized");
+                  return 1; }
+  if (nranks != c->nranks) { set_err("exchange_buckets: nranks mismatch");
+                             return 1; }
+  HIP_CHECK(hipSetDevice(c->device));
+  uint64_t total_send = 0, total_recv = 0;
+  for (int b = 0; b < nbuckets; b++) total_send += bucket_counts[b];
+  for (int s = 0; s < nranks; s++)
+    for (int b = 0; b < nbuckets; b++)
+      total_recv += recv_matrix[(uint64_t)s * nbuckets + b];
+  if (total_send != c->n) { set_err("exchange_buckets: counts != n");
+                            return 1; }
+  if (total_recv > c->cap) { set_err("exchange_buckets: recv %llu > cap %llu"
+                                     " (skew beyond balancing: a single "
+                                     "key exceeds a rank's share)",
+                                     (unsigned long long)total_recv,
+                                     (unsigned long long)c->cap);
+                             return 1; }
+  int src = c->cur, dst = c->cur ^ 1;
+  int began = timed_begin(c, K_EXCHANGE);
+  NCCL_CHECK(ncclGroupStart());
+  uint64_t soff = 0;
+  for (int b = 0; b < nbuckets; b++) {
+    uint64_t cnt = bucket_counts[b];
+    if (cnt) {
+      int peer = bucket_rank[b];
+      NCCL_CHECK(ncclSend(c->keys[src] + soff, cnt, ncclInt64, peer,
+                          c->comm, c->stream));
+      if (c->with_vals)
+        NCCL_CHECK(ncclSend(c->vals[src] + soff, cnt, ncclDouble, peer,
+                            c->comm, c->stream));
+    }
+    soff += cnt;
+  }
+  /* recvs: src-major, ascending bucket — matches each sender's ascending
+   * bucket order of sends to us */
+  uint64_t roff = 0;
+  for (int s = 0; s < nranks; s++) {
+    for (int b = 0; b < nbuckets; b++) {
+      uint64_t cnt = recv_matrix[(uint64_t)s * nbuckets + b];
+      if (cnt) {
+        NCCL_CHECK(ncclRecv(c->keys[dst] + roff, cnt, ncclInt64, s,
+                            c->comm, c->stream));
+        if (c->with_vals)
+          NCCL_CHECK(ncclRecv(c->vals[dst] + roff, cnt, ncclDouble, s,
+                              c->comm, c->stream));
+        roff += cnt;
+      }
+    }
+  }
+  NCCL_CHECK(ncclGroupEnd());
+  timed_end(c, began);
+  c->stats.bytes_total += (total_send + total_recv) * (c->with_vals ? 16 : 8);
+  c->cur = dst;
+  c->n = total_recv;
+  c->sorted = 0;
+  c->orand_valid = 0;
+  c->hist_shift_valid = -1;
+  return drain_timing(c);
+}
+
 /* ---- stats --------------------------------------------------------- */
 extern "C" int dprk_stats(dprk_ctx *c, dprk_stats_t *out) {
   if (!c || !out) { set_err("null arg"); return 1; }

@@ -73,3 +73,28 @@ def exchange_with_torch_dist(dist, keys, vals, nranks, rank):
         out_k.append(recv_bufs[p][0].numpy())
         out_v.append(recv_bufs[p][1].numpy())
     return np.concatenate(out_k), np.concatenate(out_v)
+
+
+def greedy_assign(bucket_totals, nranks: int):
+    """Deterministic greedy balancing: buckets sorted by global size
+    (descending, ties by index) each go to the currently lightest rank
+    (ties by rank id).  Identical on every rank given the all-gathered
+    count matrix — the SURVEY.md §8(e) hot-bucket mitigation: a hot
+    bucket stays whole, the rest pack around it."""
+    totals = np.asarray(bucket_totals, dtype=np.uint64)
+    order = np.lexsort((np.arange(len(totals)), -totals.astype(np.int64)))
+    load = [0] * nranks
+    assign = np.zeros(len(totals), dtype=np.int32)
+    for b in order:
+        r = min(range(nranks), key=lambda x: (load[x], x))
+        assign[b] = r
+        load[r] += int(totals[b])
+    return assign, load
+
+
+def recv_matrix_for(count_matrix, bucket_rank, me: int):
+    """count_matrix[s][b] = rank s's local size of bucket b.  Returns the
+    [nranks, nbuckets] receive matrix for rank `me` (zeros elsewhere)."""
+    m = np.asarray(count_matrix, dtype=np.uint64)
+    mask = (np.asarray(bucket_rank) == me)
+    return m * mask[None, :].astype(np.uint64)

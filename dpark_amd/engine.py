@@ -59,7 +59,8 @@ SYMBOLS = [
     "dprk_sample_keys", "dprk_sort_pairs",
     "dprk_reduce_by_key_sum", "dprk_group_by_key", "dprk_get_groups",
     "dprk_join", "dprk_get_join", "dprk_comm_unique_id", "dprk_comm_init",
-    "dprk_exchange", "dprk_stats", "dprk_stats_reset",
+    "dprk_exchange", "dprk_exchange_buckets", "dprk_stats",
+    "dprk_stats_reset",
 ]
 
 
@@ -115,6 +116,9 @@ def load_library():
                                    ctypes.c_char_p]
     lib.dprk_exchange.argtypes = [p, ctypes.POINTER(u64), ctypes.POINTER(u64),
                                   ctypes.c_int]
+    lib.dprk_exchange_buckets.argtypes = [
+        p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p,
+        ctypes.c_int]
     lib.dprk_stats.argtypes = [p, ctypes.POINTER(DprkStats)]
     lib.dprk_stats_reset.argtypes = [p]
     _lib = lib
@@ -293,6 +297,19 @@ class Engine:
         _check(self._lib,
                self._lib.dprk_comm_init(self._ctx, nranks, rank, uid),
                "dprk_comm_init")
+
+    def exchange_buckets(self, bucket_counts, bucket_rank, recv_matrix,
+                         nranks):
+        bc = np.ascontiguousarray(bucket_counts, dtype=np.uint64)
+        br = np.ascontiguousarray(bucket_rank, dtype=np.int32)
+        rm = np.ascontiguousarray(recv_matrix, dtype=np.uint64)
+        assert rm.size == nranks * len(bc)
+        _check(self._lib,
+               self._lib.dprk_exchange_buckets(
+                   self._ctx, bc.ctypes.data_as(ctypes.c_void_p),
+                   br.ctypes.data_as(ctypes.c_void_p), len(bc),
+                   rm.ctypes.data_as(ctypes.c_void_p), nranks),
+               "dprk_exchange_buckets")
 
     def exchange(self, send_counts, recv_counts):
         nr = len(send_counts)

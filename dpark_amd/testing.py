@@ -59,6 +59,14 @@ class SimEngine:
         self.vals = self.vals[order]
         return np.bincount(r, minlength=nbuckets).astype(np.uint64)
 
+    def exchange_buckets(self, bucket_counts, bucket_rank, recv_matrix,
+                         nranks):
+        assert sum(int(x) for x in bucket_counts) == len(self.keys)
+        rm = np.asarray(recv_matrix, dtype=np.uint64)
+        assert rm.shape[0] == nranks
+        assert int(rm.sum()) <= self._cap, "exchange_buckets: recv > cap"
+        # keep local shard (control-plane sim)
+
     def exchange(self, send_counts, recv_counts):
         # single-process sim cannot move data between ranks; the CONTROL
         # bookkeeping (what the driver's launch exercises) is checked:

@@ -124,6 +124,14 @@ int dprk_comm_init(dprk_ctx *ctx, int nranks, int rank, const char uid[128]);
  * received one. */
 int dprk_exchange(dprk_ctx *ctx, const uint64_t *send_counts,
                   const uint64_t *recv_counts, int nranks);
+/* Bucket-granular all-to-allv under an arbitrary bucket->rank map (after
+ * dprk_partition(ctx, nbuckets)): lets the host balance skewed buckets
+ * across ranks (SURVEY.md §8(e) hot buckets) — each bucket stays whole,
+ * so reduceByKey needs no second combine.  recv_matrix[s*nbuckets+b] =
+ * elements source rank s sends this rank from bucket b. */
+int dprk_exchange_buckets(dprk_ctx *ctx, const uint64_t *bucket_counts,
+                          const int *bucket_rank, int nbuckets,
+                          const uint64_t *recv_matrix, int nranks);
 
 /* ---- measurement ------------------------------------------------------ */
 typedef struct {

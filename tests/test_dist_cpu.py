@@ -101,3 +101,33 @@ def test_world2_exchange_reduce_matches_oracle():
         errs.append(q.get())
     assert not errs, errs
     assert all(p.exitcode == 0 for p in procs), [p.exitcode for p in procs]
+
+
+def test_greedy_assign_balances_hot_bucket():
+    """a zipf-hot bucket (just under a rank's mean
+    share) is isolated on one rank and the rest pack around it."""
+    from dpark_amd.distshuffle import greedy_assign
+    n = 1_000_000_000
+    sizes = np.full(256, (n - int(0.12 * n)) // 255, dtype=np.uint64)
+    sizes[7] = int(0.12 * n)                      # 12% hot bucket
+    assign, loads = greedy_assign(sizes, 8)
+    assert len(set(int(a) for a in assign)) == 8
+    mean = sum(loads) / 8
+    assert max(loads) / mean < 1.05               # balanced despite skew
+    # deterministic
+    a2, _ = greedy_assign(sizes, 8)
+    assert np.array_equal(assign, a2)
+    # naive rank=bucket%8 for comparison would be imbalanced:
+    naive = np.zeros(8)
+    for b, sz in enumerate(sizes):
+        naive[b % 8] += int(sz)
+    assert max(naive) / (sum(naive) / 8) > 1.8
+
+
+def test_recv_matrix_partition():
+    from dpark_amd.distshuffle import greedy_assign, recv_matrix_for
+    rng = np.random.default_rng(4)
+    m = rng.integers(0, 1000, size=(4, 64)).astype(np.uint64)
+    assign, _ = greedy_assign(m.sum(axis=0), 4)
+    got = sum(int(recv_matrix_for(m, assign, me).sum()) for me in range(4))
+    assert got == int(m.sum())                    # every element delivered

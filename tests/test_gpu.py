@@ -449,3 +449,32 @@ def test_fused_reduce_opt_in():
         e.close()
     finally:
         del os.environ["DPRK_FUSED_REDUCE"]
+
+
+def test_bucket_exchange_self():
+    """Bucket-granular RCCL exchange on the 1-GPU box: 256 buckets, all
+    assigned to rank 0 via the real grouped per-bucket ncclSend/Recv
+    self path; reduce afterwards matches the oracle."""
+    from dpark_amd.distshuffle import greedy_assign, recv_matrix_for
+    os.environ["DPRK_FORCE_SELF_EXCHANGE"] = "1"
+    try:
+        e = Engine(device=0)
+        uid = Engine.comm_unique_id()
+        e.comm_init(1, 0, uid)
+        keys, vals = gen_pairs(88, 2_000_000, 1 << 18)
+        e.reserve(len(keys), with_vals=True)
+        e.put(keys, vals)
+        send = e.partition(256)
+        m = send[None, :].astype(np.uint64)
+        assign, loads = greedy_assign(m.sum(axis=0), 1)
+        rm = recv_matrix_for(m, assign, 0)
+        e.exchange_buckets(send, assign, rm, 1)
+        assert e.count() == len(keys)
+        n = e.reduce_by_key_sum()
+        wk, wv = so.reduce_by_key_sum(keys, vals)
+        k, v = e.get()
+        assert n == len(wk) and np.array_equal(k, wk)
+        np.testing.assert_allclose(v, wv, rtol=RTOL)
+        e.close()
+    finally:
+        del os.environ["DPRK_FORCE_SELF_EXCHANGE"]
