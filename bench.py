@@ -285,7 +285,9 @@ def main():
                    "join": "M input KV-pairs/sec through join",
                    "zipf": "M KV-pairs/sec through skewed reduceByKey"}[args.op],
         "value": round(value, 2),
-        "unit": "M KV-pairs/s",
+        "unit": {"reduce": "M KV-pairs/s", "sort": "M keys/s",
+                 "join": "M input KV-pairs/s",
+                 "zipf": "M KV-pairs/s"}[args.op],
         "n_gpus": nranks,
         "steps": args.steps,
         "warmup": args.warmup,
