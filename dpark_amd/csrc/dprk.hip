@@ -1517,6 +1517,18 @@ static int stream_grid(uint64_t work_items) {
 
 static int free_all(dprk_ctx *c);
 
+/* workgroups for tile-mapped kernels: min(ntiles, cap); cap tunable via
+ * DPRK_GRID_CAP (default 2048 — measured flat 512..4096, see DESIGN) */
+static int tile_grid(uint64_t ntiles) {
+  static int cap = 0;
+  if (!cap) {
+    const char *e = getenv("DPRK_GRID_CAP");
+    cap = e ? atoi(e) : 2048;
+    if (cap < 64 || cap > 65536) cap = 2048;
+  }
+  return (int)min(ntiles, (uint64_t)cap);
+}
+
 extern "C" int dprk_init(dprk_ctx **out, int device_id) {
   *out = nullptr;
   HIP_CHECK(hipSetDevice(device_id));
@@ -1805,7 +1817,7 @@ static int radix_pass(dprk_ctx *c, int mode, int shift, uint32_t mask,
   uint64_t seg_tiles = ceil_div(ntiles, RADIX); /* nseg <= 256 */
   uint64_t nseg = ceil_div(ntiles, seg_tiles);
   int src = c->cur, dst = c->cur ^ 1;
-  int thgrid = (int)min(ntiles, (uint64_t)2048);
+  int thgrid = tile_grid(ntiles);
   int hist_ran = !(mode == 0 && c->hist_shift_valid == shift);
   if (hist_ran) {
     TIMED(c, K_TILEHIST,
@@ -2015,7 +2027,7 @@ static int fused_reduce_last_pass(dprk_ctx *c, int shift, uint64_t *n_out) {
   uint64_t seg_tiles = ceil_div(ntiles, (uint64_t)RADIX);
   uint64_t nseg = ceil_div(ntiles, seg_tiles);
   int src = c->cur, dst = c->cur ^ 1;
-  int grid = (int)min(ntiles, (uint64_t)2048);
+  int grid = tile_grid(ntiles);
   if (ensure_fused(c, ceil_div(c->cap, (uint64_t)TILE) * RADIX)) return 1;
   if (c->hist_shift_valid != shift) {
     TIMED(c, K_TILEHIST,
@@ -2069,7 +2081,7 @@ static int segmented_reduce_sorted(dprk_ctx *c, uint64_t *n_out) {
   uint64_t n = c->n;
   uint64_t ntiles = ceil_div(n, RTILE);
   int src = c->cur, scr = c->cur ^ 1;
-  int grid = (int)min(ntiles, (uint64_t)2048);
+  int grid = tile_grid(ntiles);
   TIMED(c, K_REDUCE,
         hipLaunchKernelGGL(k_reduce_tiles, dim3(grid), dim3(NTHREADS), 0,
                            c->stream, c->keys[src], c->vals[src], n,
@@ -2167,7 +2179,7 @@ static int extract_runs(dprk_ctx *c) {
   uint64_t n = c->n;
   uint64_t ntiles = ceil_div(n, RTILE);
   int src = c->cur, scr = c->cur ^ 1;
-  int grid = (int)min(ntiles, (uint64_t)2048);
+  int grid = tile_grid(ntiles);
   /* idx scratch reuses the alt VALS buffer as u64 (same 8B stride);
    * keys scratch uses the alt keys buffer */
   if (!c->with_vals) { set_err("groups need vals buffers reserved"); return 1; }
