@@ -835,7 +835,6 @@ __global__ __launch_bounds__(NTHREADS_SC, 4) void k_scatter_reduce(
      * head(p) = p==0 || key(p)!=key(p-1); frag end = p==tile_n-1 ||
      * key(p+1)!=key(p).  Values register-gathered once via pmap. */
     const uint32_t c0 = threadIdx.x * EPT;
-    const uint32_t c1 = min(c0 + (uint32_t)EPT, tile_n);
     double cv[EPT];
 #pragma unroll
     for (int j = 0; j < EPT; j++) {
@@ -1475,7 +1474,7 @@ __global__ __launch_bounds__(NTHREADS) void k_join_count(
  * output ranges are adjacent => coalesced-ish stores. */
 __global__ __launch_bounds__(NTHREADS) void k_join_emit(
     const int64_t *__restrict__ uka, const uint64_t *__restrict__ starta,
-    const uint64_t *__restrict__ cnta, uint64_t ga,
+    uint64_t ga,
     const uint64_t *__restrict__ startb, const uint64_t *__restrict__ cntb,
     const uint64_t *__restrict__ prod_off, const uint64_t *__restrict__ prod,
     const uint64_t *__restrict__ bidx, const double *__restrict__ va,
@@ -1519,6 +1518,9 @@ static int free_all(dprk_ctx *c);
 
 /* workgroups for tile-mapped kernels: min(ntiles, cap); cap tunable via
  * DPRK_GRID_CAP (default 2048 — measured flat 512..4096, see DESIGN) */
+static int run_scan_u64(dprk_ctx *c, uint64_t *data, uint64_t m,
+                        uint64_t *total_out);
+
 static int tile_grid(uint64_t ntiles) {
   static int cap = 0;
   if (!cap) {
@@ -2278,25 +2280,8 @@ extern "C" int dprk_join(dprk_ctx *a, dprk_ctx *b, uint64_t *n_out) {
   /* exclusive scan of products -> output offsets (run_boff := scan copy) */
   HIP_CHECK(hipMemcpyAsync(a->run_boff, a->run_prod, ga * 8,
                            hipMemcpyDeviceToDevice, a->stream));
-  uint64_t nchunks = ceil_div(ga, SCAN_CHUNK);
-  if (a->scan_part_cap < nchunks) {
-    if (a->scan_part) HIP_CHECK(hipFree(a->scan_part));
-    HIP_CHECK(hipMalloc(&a->scan_part, nchunks * 8));
-    a->scan_part_cap = nchunks;
-  }
-  TIMED(a, K_SCAN,
-        hipLaunchKernelGGL(k_scanA, dim3((uint32_t)nchunks), dim3(NTHREADS),
-                           0, a->stream, a->run_boff, ga, a->scan_part));
-  TIMED(a, K_SCAN,
-        hipLaunchKernelGGL(k_scanB, dim3(1), dim3(NTHREADS), 0, a->stream,
-                           a->scan_part, nchunks, a->d_scalar));
-  TIMED(a, K_SCAN,
-        hipLaunchKernelGGL(k_scanC, dim3((uint32_t)nchunks), dim3(NTHREADS),
-                           0, a->stream, a->run_boff, ga, a->scan_part));
   uint64_t total = 0;
-  HIP_CHECK(hipMemcpyAsync(&total, a->d_scalar, 8, hipMemcpyDeviceToHost,
-                           a->stream));
-  HIP_CHECK(hipStreamSynchronize(a->stream));
+  if (run_scan_u64(a, a->run_boff, ga, &total)) return 1;
   /* output buffers, capacity-cached across calls */
   if (total > a->join_cap || !a->join_k) {
     if (a->join_k) HIP_CHECK(hipFree(a->join_k));
@@ -2311,7 +2296,7 @@ extern "C" int dprk_join(dprk_ctx *a, dprk_ctx *b, uint64_t *n_out) {
   TIMED(a, K_JOIN,
         hipLaunchKernelGGL(k_join_emit, dim3(2048), dim3(NTHREADS), 0,
                            a->stream, a->grp_keys, a->grp_starts,
-                           a->grp_counts, ga, b->grp_starts, b->grp_counts,
+                           ga, b->grp_starts, b->grp_counts,
                            a->run_boff, a->run_prod, d_bidx,
                            a->vals[a->cur], b->vals[b->cur], a->join_k,
                            a->join_a, a->join_b));
