@@ -13,8 +13,8 @@ from __future__ import annotations
 
 import re
 
-from .rdd import (LocalScheduler, ParallelCollection, ShuffleStore,
-                  TextFileRDD, UnionRDD)
+from .rdd import (LocalScheduler, MultiProcessScheduler, ParallelCollection,
+                  ShuffleStore, TextFileRDD, UnionRDD)
 
 
 class Broadcast:
@@ -40,12 +40,14 @@ class Accumulator:
 
 class DparkContext:
     def __init__(self, master="local", parallel=None):
-        m = re.match(r"local\[(\d+)\]", master or "local")
-        workers = int(m.group(1)) if m else 4
+        m = re.match(r"(local|process)\[(\d+)\]", master or "local")
+        mode = m.group(1) if m else (master or "local")
+        workers = int(m.group(2)) if m else 4
         self.master = master
         self.defaultParallelism = parallel or workers
         self.shuffle_store = ShuffleStore()
-        self.scheduler = LocalScheduler(self, workers=workers)
+        cls = MultiProcessScheduler if mode == "process" else LocalScheduler
+        self.scheduler = cls(self, workers=workers)
         self._stopped = False
 
     # -- RDD constructors -------------------------------------------------

@@ -142,6 +142,11 @@ class ShuffleStore:
 # RDD  [dpark/rdd.py [U]]
 # ---------------------------------------------------------------------------
 
+# set in worker processes: a {shuffleId: {mapId: {bucket: items}}} slice
+# shipped with each task (dpark ships file URIs; single node ships data)
+_CHILD_STORE = None
+
+
 class RDD:
     _next_id = itertools.count(1)
 
@@ -150,6 +155,14 @@ class RDD:
         self.id = next(self._next_id)
         self.dependencies = []
         self.partitioner = None
+
+    def __getstate__(self):
+        # process mode (dpark -m process [U]): the task closure is pickled
+        # to a forked worker; the context (scheduler, locks) stays in the
+        # driver — dpark/serialize.py :: dump_closure analogue
+        d = dict(self.__dict__)
+        d["ctx"] = None
+        return d
 
     # -- to be provided by subclasses ------------------------------------
     def splits(self):
@@ -447,11 +460,14 @@ class ShuffledRDD(RDD):
         return self._splits
 
     def compute(self, split):
-        self.ctx.scheduler.ensure_shuffle(self.dep)
+        if self.ctx is not None:
+            self.ctx.scheduler.ensure_shuffle(self.dep)
+            store = self.ctx.shuffle_store
+        else:
+            store = _CHILD_STORE   # worker process: parent pre-materialized
         merge_combiners = self.aggregator.mergeCombiners
         d = {}
-        for bucket in self.ctx.shuffle_store.fetch(self.dep.shuffleId,
-                                                   split.index):
+        for bucket in store.fetch(self.dep.shuffleId, split.index):
             for k, c in bucket:
                 if k in d:
                     d[k] = merge_combiners(d[k], c)
@@ -488,9 +504,12 @@ class CoGroupedRDD(RDD):
         d = {}
         for i, dep in enumerate(self.deps):
             if isinstance(dep, ShuffleDependency):
-                self.ctx.scheduler.ensure_shuffle(dep)
-                for bucket in self.ctx.shuffle_store.fetch(dep.shuffleId,
-                                                           split.index):
+                if self.ctx is not None:
+                    self.ctx.scheduler.ensure_shuffle(dep)
+                    store = self.ctx.shuffle_store
+                else:
+                    store = _CHILD_STORE
+                for bucket in store.fetch(dep.shuffleId, split.index):
                     for k, vs in bucket:
                         if k not in d:
                             d[k] = tuple([] for _ in range(nrdds))
@@ -507,6 +526,43 @@ class CoGroupedRDD(RDD):
 # local scheduler  [dpark/schedule.py :: LocalScheduler /
 # MultiProcessScheduler, dpark/task.py :: ShuffleMapTask / ResultTask [U]]
 # ---------------------------------------------------------------------------
+
+def _collect_shuffle_slices(rdd, split_index, store, seen=None):
+    # The bucket data a worker task needs: for every ShuffleDependency
+    # reachable through NARROW deps of rdd, its maps' buckets for this
+    # reduce split (the single-node analogue of shipping map-output URIs
+    # to the fetcher [U]).
+    seen = seen if seen is not None else set()
+    if rdd.id in seen:
+        return {}
+    seen.add(rdd.id)
+    out = {}
+    for dep in rdd.dependencies:
+        if isinstance(dep, ShuffleDependency):
+            maps = store._outputs.get(dep.shuffleId, {})
+            out[dep.shuffleId] = {
+                m: {split_index: maps[m].get(split_index, [])}
+                for m in maps}
+        else:
+            out.update(_collect_shuffle_slices(dep.rdd, split_index, store,
+                                               seen))
+    return out
+
+
+def _process_worker(payload):
+    # Runs in a forked worker: rebuild the task closure, install the
+    # shipped shuffle slice, compute one partition.
+    import cloudpickle
+    global _CHILD_STORE
+    func, rdd, split, slices = cloudpickle.loads(payload)
+    st = ShuffleStore()
+    for sid, maps in slices.items():
+        for m, buckets in maps.items():
+            st.put(sid, m, buckets)
+        st.mark_complete(sid)
+    _CHILD_STORE = st
+    return func(rdd.iterator(split))
+
 
 class LocalScheduler:
     """Stage-cutting scheduler: cuts at ShuffleDependency, runs one
@@ -578,8 +634,34 @@ class LocalScheduler:
         splits = rdd.splits()
         with self._shuffle_lock:
             self._materialize_ancestors(rdd)
+        return self._run_result_tasks(rdd, func, splits)
+
+    def _run_result_tasks(self, rdd, func, splits):
         # materialize shuffle deps bottom-up first (single-thread walk
         # avoids re-entrant pools); compute() calls ensure_shuffle lazily
         # as well, so this is an optimization not a correctness need.
         with ThreadPoolExecutor(self.workers) as pool:
             return list(pool.map(lambda s: func(rdd.iterator(s)), splits))
+
+
+class MultiProcessScheduler(LocalScheduler):
+    """dpark/schedule.py :: MultiProcessScheduler [U] (-m process):
+    result tasks fork to worker PROCESSES (closures via cloudpickle, the
+    dump_closure analogue), escaping the GIL for CPU-bound python code.
+    Shuffle map stages run in the driver's thread pool (cheap dict work);
+    each shipped result task carries the shuffle buckets it will fetch --
+    the single-node stand-in for dpark's HTTP fetch of map outputs [U].
+    Accumulator updates do not propagate from workers (documented limit;
+    dpark ships them back with task results [U])."""
+
+    def _run_result_tasks(self, rdd, func, splits):
+        import multiprocessing
+        import cloudpickle
+        store = self.ctx.shuffle_store
+        payloads = [
+            cloudpickle.dumps(
+                (func, rdd, s, _collect_shuffle_slices(rdd, s.index, store)))
+            for s in splits]
+        procs = min(self.workers, max(1, len(payloads)))
+        with multiprocessing.get_context("spawn").Pool(procs) as pool:
+            return pool.map(_process_worker, payloads)

@@ -157,3 +157,25 @@ def test_vector_pairs_cpu_path(ctx):
     assert set(gb) == set(og)
     for k in og:
         np.testing.assert_allclose(sorted(gb[k]), sorted(og[k]), rtol=1e-9)
+
+
+def test_process_mode_scheduler():
+    """dpark -m process analogue: result tasks in forked workers
+    (cloudpickle closures), results identical to thread mode."""
+    ctxp = DparkContext("process[2]")
+    pairs = [(i % 17, float(i)) for i in range(2000)]
+    got = dict(ctxp.parallelize(pairs, 4)
+               .reduceByKey(operator.add, 3).collect())
+    want = {k: v for k, v in
+            zip(*so.reduce_by_key_sum(np.array([k for k, _ in pairs]),
+                                      np.array([v for _, v in pairs])))}
+    assert set(got) == set(int(k) for k in want)
+    for k, v in want.items():
+        np.testing.assert_allclose(got[int(k)], v, rtol=1e-9)
+    # chained transforms over the shuffle boundary in workers
+    out = sorted(ctxp.parallelize(pairs, 3)
+                 .reduceByKey(operator.add, 2)
+                 .map(lambda kv: (kv[0], kv[1] * 2))
+                 .collect())
+    assert len(out) == 17
+    ctxp.stop()
