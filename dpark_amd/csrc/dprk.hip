@@ -7,13 +7,18 @@
  * replaces is documented in include/dprk.h.
  *
  * Pipeline (DESIGN.md has the full data-layout and roofline derivation):
- *   k_gen            device PRNG (splitmix64; spec mirrored in oracle/synth.py)
+ *   k_gen / k_gen_hist / k_gen_alias   device PRNG (splitmix64 / alias
+ *       table; numpy mirrors in oracle/synth.py are byte-identical);
+ *       k_gen_hist also emits the pass-1 tile histogram
  *   k_minmax_or      OR/AND reduce of biased keys -> digit-skip detection
  *   k_tile_hist      per-tile 256-bin digit histogram      (8 B/pair read)
  *   k_seg_sums/k_scan_mid/k_tile_off   3-level scan of tile histograms
  *   k_scatter        rank + LDS exchange + coalesced write (32 B/pair r+w)
- *   k_reduce_tiles / k_stitch / k_compact_runs   segmented float64 reduce
- *   k_head_* / k_join_*                group/join machinery
+ *   k_sample / mode-2 digits           device RangePartitioner (sortByKey)
+ *   k_reduce_tiles / k_stitchA-C / k_compact_runs  segmented f64 reduce
+ *   k_scatter_reduce / k_fused_gather  fused last-pass reduce (opt-in)
+ *   k_head_* / k_join_* / k_scanA-C    group/join machinery
+ *   dprk_exchange / dprk_exchange_buckets   RCCL all-to-allv over xGMI
  * Sorting int64 ascending == unsigned radix on u = key XOR 0x8000...0.
  */
 
