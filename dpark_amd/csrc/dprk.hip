@@ -123,6 +123,16 @@ struct dprk_ctx {
   int orand_valid = 0;
   uint64_t known_or = 0, known_and = 0;
   int hist_shift_valid = -1; /* tile_hist currently holds this pass shift */
+  /* lookback-scatter scratch (DPRK_LOOKBACK) */
+  unsigned long long *d_lb = nullptr;   /* [ntiles*RADIX] granules */
+  unsigned long long *d_ticket = nullptr;
+  unsigned int *d_fail = nullptr;
+  uint32_t *d_ghist = nullptr;          /* [8][RADIX] global pass hists */
+  uint64_t ghist_shifts = 0;            /* packed shifts the hists cover */
+  int ghist_nact = 0;
+  int ghist_valid = 0;
+  uint32_t *d_dbase = nullptr;          /* [RADIX] current pass bases */
+  uint64_t lb_cap_tiles = 0;
 
 
   /* radix scratch */
@@ -274,6 +284,16 @@ __device__ __forceinline__ uint32_t d_digit(int64_t key, int mode, int shift,
   return (uint32_t)(u >> shift) & mask;
 }
 
+/* wave-level inclusive scan over 64 lanes (no barriers) */
+__device__ __forceinline__ uint32_t d_wave_incl_scan(uint32_t v, int lane) {
+#pragma unroll
+  for (int d = 1; d < 64; d <<= 1) {
+    uint32_t up = (uint32_t)__shfl_up((int)v, d, 64);
+    if (lane >= d) v += up;
+  }
+  return v;
+}
+
 /* strided sample of the current keys (sortByKey splitter sampling —
  * dpark samples partitions to the driver [U]) */
 __global__ __launch_bounds__(NTHREADS) void k_sample(
@@ -325,9 +345,13 @@ __global__ __launch_bounds__(NTHREADS) void k_gen(
 __global__ __launch_bounds__(NTHREADS) void k_gen_hist(
     int64_t *__restrict__ keys, double *__restrict__ vals, uint64_t n,
     uint64_t seed_hi, uint64_t keymask, uint32_t *__restrict__ tile_hist,
-    uint64_t ntiles, int shift) {
+    uint64_t ntiles, int shift, uint64_t active_shifts, int nact,
+    uint32_t *__restrict__ g_hist /* [nact][RADIX]; lookback path */) {
   __shared__ uint32_t h[NTHREADS / 64][RADIX];
+  __shared__ uint32_t gh[NTHREADS / 64][8][RADIX]; /* 32 KiB */
   const int wave = threadIdx.x / 64;
+  for (int i = threadIdx.x; i < (NTHREADS / 64) * 8 * RADIX; i += NTHREADS)
+    ((uint32_t *)gh)[i] = 0;
   for (uint64_t t = blockIdx.x; t < ntiles; t += gridDim.x) {
     for (int j = threadIdx.x; j < (NTHREADS / 64) * RADIX; j += NTHREADS)
       ((uint32_t *)h)[j] = 0;
@@ -343,6 +367,11 @@ __global__ __launch_bounds__(NTHREADS) void k_gen_hist(
       *reinterpret_cast<longlong2 *>(&keys[i]) = kk;
       atomicAdd(&h[wave][d_digit(kk.x, 0, shift, 0xFF)], 1u);
       atomicAdd(&h[wave][d_digit(kk.y, 0, shift, 0xFF)], 1u);
+      for (int p = 0; p < nact; p++) {
+        int sh = (int)((active_shifts >> (8 * p)) & 0xFF);
+        atomicAdd(&gh[wave][p][d_digit(kk.x, 0, sh, 0xFF)], 1u);
+        atomicAdd(&gh[wave][p][d_digit(kk.y, 0, sh, 0xFF)], 1u);
+      }
       if (vals) {
         double2 vv;
         uint64_t v0 = d_splitmix64(x0 ^ 0xDEADBEEFCAFEBABEull);
@@ -358,6 +387,10 @@ __global__ __launch_bounds__(NTHREADS) void k_gen_hist(
       int64_t k = (int64_t)(d_splitmix64(x) & keymask);
       keys[i] = k;
       atomicAdd(&h[wave][d_digit(k, 0, shift, 0xFF)], 1u);
+      for (int p = 0; p < nact; p++) {
+        int sh = (int)((active_shifts >> (8 * p)) & 0xFF);
+        atomicAdd(&gh[wave][p][d_digit(k, 0, sh, 0xFF)], 1u);
+      }
       if (vals) {
         uint64_t vb = d_splitmix64(x ^ 0xDEADBEEFCAFEBABEull);
         vals[i] = (double)(vb >> 11) * (1.0 / 9007199254740992.0);
@@ -371,6 +404,13 @@ __global__ __launch_bounds__(NTHREADS) void k_gen_hist(
     }
     __syncthreads();
   }
+  if (g_hist)
+    for (int p = 0; p < nact; p++)
+      for (int d = threadIdx.x; d < RADIX; d += NTHREADS) {
+        uint32_t s = 0;
+        for (int w = 0; w < NTHREADS / 64; w++) s += gh[w][p][d];
+        if (s) atomicAdd(&g_hist[p * RADIX + d], s);
+      }
 }
 
 /* Zipf (or any discrete distribution) sampling via a caller-provided
@@ -397,6 +437,46 @@ __global__ __launch_bounds__(NTHREADS) void k_gen_alias(
       vals[i] = (double)(vb >> 11) * (1.0 / 9007199254740992.0); /* 2^-53 */
     }
   }
+}
+
+/* global per-pass digit histograms for every ACTIVE pass in ONE read of
+ * the keys (lookback path: replaces per-pass k_tile_hist + the 3-level
+ * scans).  active_shifts packs the byte shifts of up to 8 active passes;
+ * per-wave LDS replicas, one global atomic merge per workgroup. */
+__global__ __launch_bounds__(NTHREADS) void k_hist_all(
+    const int64_t *__restrict__ keys, uint64_t n, uint64_t active_shifts,
+    int nact, uint32_t *__restrict__ g_hist /* [nact][RADIX] */) {
+  __shared__ uint32_t h[NTHREADS / 64][8][RADIX];   /* 32 KiB */
+  const int wave = threadIdx.x / 64;
+  for (int i = threadIdx.x; i < (NTHREADS / 64) * 8 * RADIX; i += NTHREADS)
+    ((uint32_t *)h)[i] = 0;
+  __syncthreads();
+  const uint64_t stride = 2 * (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = 2 * ((uint64_t)blockIdx.x * blockDim.x + threadIdx.x);
+       i + 1 < n; i += stride) {
+    longlong2 kk = *reinterpret_cast<const longlong2 *>(&keys[i]);
+    uint64_t u0 = (uint64_t)kk.x ^ SIGNFLIP;
+    uint64_t u1 = (uint64_t)kk.y ^ SIGNFLIP;
+    for (int p = 0; p < nact; p++) {
+      int sh = (int)((active_shifts >> (8 * p)) & 0xFF);
+      atomicAdd(&h[wave][p][(u0 >> sh) & 0xFF], 1u);
+      atomicAdd(&h[wave][p][(u1 >> sh) & 0xFF], 1u);
+    }
+  }
+  if ((n & 1) && blockIdx.x == 0 && threadIdx.x == 0) {
+    uint64_t u = (uint64_t)keys[n - 1] ^ SIGNFLIP;
+    for (int p = 0; p < nact; p++) {
+      int sh = (int)((active_shifts >> (8 * p)) & 0xFF);
+      atomicAdd(&h[wave][p][(u >> sh) & 0xFF], 1u);
+    }
+  }
+  __syncthreads();
+  for (int p = 0; p < nact; p++)
+    for (int d = threadIdx.x; d < RADIX; d += NTHREADS) {
+      uint32_t s = 0;
+      for (int w = 0; w < NTHREADS / 64; w++) s += h[w][p][d];
+      if (s) atomicAdd(&g_hist[p * RADIX + d], s);
+    }
 }
 
 /* OR/AND reduction of biased keys -> constant-digit (skip-pass) detection */
@@ -577,14 +657,189 @@ __global__ __launch_bounds__(NTHREADS) void k_tile_off(
  * running counters — no cross-wave traffic between two barriers.
  * Digit-sorted staging in LDS, then a sequential sweep writes
  * equal-digit runs to contiguous global addresses. */
-/* wave-level inclusive scan over 64 lanes (no barriers) */
-__device__ __forceinline__ uint32_t d_wave_incl_scan(uint32_t v, int lane) {
+/* lookback scatter: one kernel per pass does per-tile histogram (from
+ * registers), STABLE rank, LDS exchange, AND its own global offsets via
+ * onesweep-style decoupled lookback over agent-scope 8-byte granules
+ * (the R2 recipe: one relaxed agent store publishes {state,value}; the
+ * per-XCD L2s are not coherent, so plain loads/stores would be stale —
+ * cdna_hip_programming.md §6 G16).  Tiles are taken from a global
+ * ticket so assignment order == issue order (no circular waits with a
+ * partially resident grid); spins are BOUNDED and set a failure flag
+ * the host turns into a 3-kernel-path retry — hang-safe by design.
+ * Output is bit-identical to the k_tile_hist+scan+k_scatter path (the
+ * same permutation), so path selection has no numerics implications. */
+#define LB_AGG  (1ull << 62)
+#define LB_INC  (2ull << 62)
+#define LB_VAL  ((1ull << 62) - 1)
+#define LB_SPIN_LIMIT (1u << 16)
+#define LB_BATCH 1ull /* batch>1 delays AGG publication by batch-1 tile
+    times, degenerating successors' lookbacks into long AGG walks:
+    batch=4 measured 169 ms/launch vs 1.25 at batch=1 (parity held) */
+
+template <bool WITH_VALS>
+__global__ __launch_bounds__(NTHREADS_SC, 2) void k_scatter_lb(
+    const int64_t *__restrict__ keys_in, const double *__restrict__ vals_in,
+    int64_t *__restrict__ keys_out, double *__restrict__ vals_out,
+    uint64_t n, uint64_t ntiles, int shift,
+    const uint32_t *__restrict__ digit_base,
+    unsigned long long *__restrict__ lb,
+    unsigned long long *__restrict__ ticket,
+    unsigned int *__restrict__ d_fail) {
+  constexpr int NW = NTHREADS_SC / 64;
+  constexpr int WBLK = TILE / NW;
+  constexpr int EPL = WBLK / 64;
+  __shared__ int64_t lk[TILE];
+  __shared__ double lv[WITH_VALS ? TILE : 1];
+  __shared__ uint32_t excl[RADIX];
+  __shared__ uint32_t goff[RADIX];
+  __shared__ uint32_t h[NW][RADIX];
+  __shared__ uint32_t wtot[NW > 4 ? NW : 4];
+  __shared__ uint64_t t_sh;
+
+  const int wave = threadIdx.x / 64, lane = threadIdx.x % 64;
+  const uint64_t lowmask = ((uint64_t)1 << lane) - 1;
+  const uint32_t mask = 0xFF;
+
+  for (;;) {
+    /* batched tickets (LB_BATCH consecutive tiles per grab): one shared
+     * ticket word saturates near 88 atomics/us (microarch dequeue row) —
+     * per-tile grabs cost ~0.28 ms/pass at 24k tiles.  Thread 0 folds
+     * the failure flag into the ticket so every thread sees the same
+     * decision (a divergent return before a barrier would strand the
+     * workgroup). */
+    if (threadIdx.x == 0) {
+      t_sh = __hip_atomic_load(d_fail, __ATOMIC_RELAXED,
+                               __HIP_MEMORY_SCOPE_AGENT)
+                 ? ntiles
+                 : __hip_atomic_fetch_add(ticket, LB_BATCH, __ATOMIC_RELAXED,
+                                          __HIP_MEMORY_SCOPE_AGENT);
+    }
+    __syncthreads();
+    const uint64_t t0_batch = t_sh;
+    __syncthreads();
+    if (t0_batch >= ntiles) return;
+    for (uint64_t t = t0_batch; t < min(t0_batch + LB_BATCH, ntiles); t++) {
+    const uint64_t start = t * (uint64_t)TILE;
+    const uint64_t end = min(start + TILE, n);
+    const uint32_t tile_n = (uint32_t)(end - start);
+    const uint32_t wbeg = wave * WBLK;
+
+    /* phase 1: load + wave-private histogram */
+    for (int j = 0; j < RADIX / 64; j++) h[wave][lane * (RADIX / 64) + j] = 0;
+    __syncthreads();
+    int64_t rk[EPL];
+    double rv[EPL];
+    uint32_t rd[EPL];
 #pragma unroll
-  for (int d = 1; d < 64; d <<= 1) {
-    uint32_t up = (uint32_t)__shfl_up((int)v, d, 64);
-    if (lane >= d) v += up;
+    for (int i = 0; i < EPL; i++) {
+      uint32_t idx = wbeg + i * 64 + lane;
+      if (idx < tile_n) {
+        rk[i] = keys_in[start + idx];
+        if (WITH_VALS) rv[i] = vals_in[start + idx];
+        rd[i] = d_digit(rk[i], 0, shift, mask);
+        atomicAdd(&h[wave][rd[i]], 1u);
+      }
+    }
+    __syncthreads();
+
+    /* phase 2: tile counts -> publish AGG granule; local digit excl scan
+     * (shfl) + per-wave bases in place */
+    uint32_t tile_cnt = 0;
+    if (threadIdx.x < RADIX) {
+      for (int w = 0; w < NW; w++) tile_cnt += h[w][threadIdx.x];
+      __hip_atomic_store(&lb[t * RADIX + threadIdx.x], LB_AGG | tile_cnt,
+                         __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+      uint32_t inc = d_wave_incl_scan(tile_cnt, lane);
+      if (lane == 63) wtot[wave] = inc;
+      excl[threadIdx.x] = inc - tile_cnt;
+    }
+    __syncthreads();
+    if (threadIdx.x < RADIX) {
+      if (wave > 0) {
+        uint32_t add = 0;
+        for (int w = 0; w < wave; w++) add += wtot[w];
+        excl[threadIdx.x] += add;
+      }
+      uint32_t b = excl[threadIdx.x];
+      for (int w = 0; w < NW; w++) {
+        uint32_t cnt = h[w][threadIdx.x];
+        h[w][threadIdx.x] = b;
+        b += cnt;
+      }
+    }
+    __syncthreads();
+
+    /* phase 3: stable rank + LDS scatter */
+#pragma unroll
+    for (int i = 0; i < EPL; i++) {
+      uint32_t idx = wbeg + i * 64 + lane;
+      if (idx < tile_n) {
+        uint32_t dig = rd[i];
+        uint64_t same = __ballot(1);
+        for (int bit = 0; bit < RADIX_BITS; bit++) {
+          uint64_t bb = __ballot((dig >> bit) & 1);
+          same &= ((dig >> bit) & 1) ? bb : ~bb;
+        }
+        uint32_t rank_in_wave = (uint32_t)__popcll(same & lowmask);
+        uint32_t base = h[wave][dig];
+        if (rank_in_wave == 0)
+          h[wave][dig] = base + (uint32_t)__popcll(same);
+        uint32_t pos = base + rank_in_wave;
+        lk[pos] = rk[i];
+        if (WITH_VALS) lv[pos] = rv[i];
+      }
+    }
+
+    /* lookback (thread = digit): sum predecessors' granules back to the
+     * nearest INC; publish own INC; bounded spins -> d_fail */
+    if (threadIdx.x < RADIX) {
+      uint64_t excl_g = 0;
+      uint32_t spins = 0;
+      bool ok = true;
+      int64_t s = (int64_t)t - 1;
+      while (s >= 0) {
+        uint64_t e = __hip_atomic_load(&lb[(uint64_t)s * RADIX + threadIdx.x],
+                                       __ATOMIC_RELAXED,
+                                       __HIP_MEMORY_SCOPE_AGENT);
+        if (!(e >> 62)) {
+          __builtin_amdgcn_s_sleep(2);
+          if (++spins > LB_SPIN_LIMIT) {
+            __hip_atomic_store(d_fail, 1u, __ATOMIC_RELAXED,
+                               __HIP_MEMORY_SCOPE_AGENT);
+            ok = false;
+            break;
+          }
+          continue;
+        }
+        excl_g += e & LB_VAL;
+        if ((e >> 62) >= 2) break;
+        s--;
+      }
+      if (ok) {
+        __hip_atomic_store(&lb[t * RADIX + threadIdx.x],
+                           LB_INC | (excl_g + tile_cnt), __ATOMIC_RELAXED,
+                           __HIP_MEMORY_SCOPE_AGENT);
+        goff[threadIdx.x] = digit_base[threadIdx.x] + (uint32_t)excl_g;
+      }
+    }
+    /* uniform post-lookback failure check (reuse wtot[0] as a flag) */
+    if (threadIdx.x == 0)
+      wtot[0] = __hip_atomic_load(d_fail, __ATOMIC_RELAXED,
+                                  __HIP_MEMORY_SCOPE_AGENT);
+    __syncthreads();
+    if (wtot[0]) return;
+
+    /* phase 4: write out in digit-sorted tile order */
+    for (uint32_t idx = threadIdx.x; idx < tile_n; idx += NTHREADS_SC) {
+      int64_t key = lk[idx];
+      uint32_t dig = d_digit(key, 0, shift, mask);
+      uint32_t dst = goff[dig] + (idx - excl[dig]);
+      keys_out[dst] = key;
+      if (WITH_VALS) vals_out[dst] = lv[idx];
+    }
+    __syncthreads();
+    }  /* batch loop */
   }
-  return v;
 }
 
 template <bool WITH_VALS>
@@ -1623,6 +1878,13 @@ static int free_all(dprk_ctx *c) {
     if (c->vals[i]) HIP_CHECK(hipFree(c->vals[i]));
     c->keys[i] = nullptr; c->vals[i] = nullptr;
   }
+  if (c->d_lb) HIP_CHECK(hipFree(c->d_lb));
+  if (c->d_ticket) HIP_CHECK(hipFree(c->d_ticket));
+  if (c->d_fail) HIP_CHECK(hipFree(c->d_fail));
+  if (c->d_ghist) HIP_CHECK(hipFree(c->d_ghist));
+  if (c->d_dbase) HIP_CHECK(hipFree(c->d_dbase));
+  c->d_lb = nullptr; c->d_ticket = nullptr; c->d_fail = nullptr;
+  c->d_ghist = nullptr; c->d_dbase = nullptr; c->lb_cap_tiles = 0;
   if (c->tile_hist) HIP_CHECK(hipFree(c->tile_hist));
   if (c->tile_off) HIP_CHECK(hipFree(c->tile_off));
   if (c->seg_sums) HIP_CHECK(hipFree(c->seg_sums));
@@ -1735,6 +1997,7 @@ extern "C" int dprk_gen_pairs(dprk_ctx *c, uint64_t seed, uint64_t n,
   uint64_t keymask =
       (key_universe == 0) ? ~0ull : (key_universe - 1); /* power of two */
   c->hist_shift_valid = -1;
+  c->ghist_valid = 0;
   if (dist == 1) {
     TIMED(c, K_GEN,
           hipLaunchKernelGGL(k_gen_alias, dim3(stream_grid(n)), dim3(NTHREADS),
@@ -1742,13 +2005,25 @@ extern "C" int dprk_gen_pairs(dprk_ctx *c, uint64_t seed, uint64_t n,
                              c->with_vals ? c->vals[c->cur] : nullptr, n,
                              seed << 32, c->alias_table, key_universe));
   } else {
-    /* fused gen+hist: the first sort pass's digit shift for masked keys
-     * is the lowest nonzero mask byte (= 0 unless the low byte is
-     * masked out); full-range keys also start at shift 0 */
+    /* fused gen+hist: per-tile hist for the first sort pass (base path)
+     * AND global per-pass histograms (lookback path) come out of the
+     * generation pass itself — sorting gen-sourced data then reads the
+     * keys zero extra times before the first scatter.  Active passes
+     * for masked keys = the nonzero mask bytes (biased bit 63 is
+     * constant for universes < 2^63). */
     int shift0 = 0;
     while (keymask != ~0ull && ((keymask >> shift0) & 0xFF) == 0 &&
            shift0 < 56)
       shift0 += 8;
+    uint64_t shifts = 0;
+    int nact = 0;
+    for (int p = 0; p < 8; p++) {
+      uint32_t mb = (uint32_t)(keymask >> (8 * p)) & 0xFF;
+      if (keymask == ~0ull || mb)
+        shifts |= ((uint64_t)(8 * p)) << (8 * (nact++));
+    }
+    if (!c->d_ghist) HIP_CHECK(hipMalloc(&c->d_ghist, 8 * RADIX * 4));
+    HIP_CHECK(hipMemsetAsync(c->d_ghist, 0, 8 * RADIX * 4, c->stream));
     uint64_t ntiles = ceil_div(n, (uint64_t)TILE);
     TIMED(c, K_GEN,
           hipLaunchKernelGGL(k_gen_hist,
@@ -1756,8 +2031,11 @@ extern "C" int dprk_gen_pairs(dprk_ctx *c, uint64_t seed, uint64_t n,
                              dim3(NTHREADS), 0, c->stream, c->keys[c->cur],
                              c->with_vals ? c->vals[c->cur] : nullptr, n,
                              seed << 32, keymask, c->tile_hist, ntiles,
-                             shift0));
+                             shift0, shifts, nact, c->d_ghist));
     c->hist_shift_valid = shift0;
+    c->ghist_shifts = shifts;
+    c->ghist_nact = nact;
+    c->ghist_valid = 1;
   }
   c->n = n;
   c->sorted = 0;
@@ -1794,6 +2072,7 @@ extern "C" int dprk_put_pairs(dprk_ctx *c, const int64_t *keys,
   c->sorted = 0;
   c->orand_valid = 0;
   c->hist_shift_valid = -1;
+  c->ghist_valid = 0;
   return 0;
 }
 
@@ -1924,6 +2203,109 @@ extern "C" int dprk_partition_range(dprk_ctx *c, const int64_t *splitters,
   return rc ? rc : rc2;
 }
 
+static int ensure_lookback(dprk_ctx *c, uint64_t ntiles) {
+  if (ntiles <= c->lb_cap_tiles && c->d_lb) return 0;
+  if (c->d_lb) HIP_CHECK(hipFree(c->d_lb));
+  HIP_CHECK(hipMalloc(&c->d_lb, ntiles * RADIX * 8));
+  if (!c->d_ticket) HIP_CHECK(hipMalloc(&c->d_ticket, 8));
+  if (!c->d_fail) HIP_CHECK(hipMalloc(&c->d_fail, 4));
+  if (!c->d_ghist) HIP_CHECK(hipMalloc(&c->d_ghist, 8 * RADIX * 4));
+  if (!c->d_dbase) HIP_CHECK(hipMalloc(&c->d_dbase, RADIX * 4));
+  c->lb_cap_tiles = ntiles;
+  return 0;
+}
+
+/* lookback sort (DPRK_LOOKBACK): one upfront read builds every active
+ * pass's global digit histogram; each pass is then a single
+ * k_scatter_lb launch (per-tile hist + decoupled-lookback offsets).
+ * Falls back per pass to the 3-kernel path on bounded-spin failure. */
+static int sort_pairs_lookback(dprk_ctx *c, const int *passes, int np) {
+  uint64_t n = c->n;
+  uint64_t ntiles = ceil_div(n, TILE);
+  if (ensure_lookback(c, ceil_div(c->cap, (uint64_t)TILE))) return 1;
+  /* global histograms for all active passes (8 KiB readback) */
+  uint64_t shifts = 0;
+  for (int i = 0; i < np; i++)
+    shifts |= ((uint64_t)(8 * passes[i])) << (8 * i);
+  int gh_map[8];
+  if (c->ghist_valid) {
+    /* the generator already built global hists; its pass list covers a
+     * superset (all nonzero mask bytes == the active passes).  Map each
+     * active pass to its slot. */
+    for (int i = 0; i < np; i++) {
+      gh_map[i] = -1;
+      for (int j = 0; j < c->ghist_nact; j++)
+        if (((c->ghist_shifts >> (8 * j)) & 0xFF) ==
+            (uint64_t)(8 * passes[i]))
+          gh_map[i] = j;
+    }
+    bool all = true;
+    for (int i = 0; i < np; i++) all = all && gh_map[i] >= 0;
+    if (!all) c->ghist_valid = 0;
+  }
+  if (!c->ghist_valid) {
+    HIP_CHECK(hipMemsetAsync(c->d_ghist, 0, 8 * RADIX * 4, c->stream));
+    TIMED(c, K_HIST,
+          hipLaunchKernelGGL(k_hist_all, dim3(stream_grid(n / 2 + 1)),
+                             dim3(NTHREADS), 0, c->stream, c->keys[c->cur],
+                             n, shifts, np, c->d_ghist));
+    c->stats.bytes_total += n * 8;
+    for (int i = 0; i < np; i++) gh_map[i] = i;
+  }
+  uint32_t gh[8 * RADIX];
+  HIP_CHECK(hipMemcpyAsync(gh, c->d_ghist, 8 * RADIX * 4,
+                           hipMemcpyDeviceToHost, c->stream));
+  HIP_CHECK(hipStreamSynchronize(c->stream));
+  for (int i = 0; i < np; i++) {
+    /* exclusive digit bases for this pass */
+    uint32_t base[RADIX], run = 0;
+    for (int d = 0; d < RADIX; d++) {
+      base[d] = run;
+      run += gh[gh_map[i] * RADIX + d];
+    }
+    HIP_CHECK(hipMemcpyAsync(c->d_dbase, base, RADIX * 4,
+                             hipMemcpyHostToDevice, c->stream));
+    HIP_CHECK(hipMemsetAsync(c->d_lb, 0, ntiles * RADIX * 8, c->stream));
+    HIP_CHECK(hipMemsetAsync(c->d_ticket, 0, 8, c->stream));
+    HIP_CHECK(hipMemsetAsync(c->d_fail, 0, 4, c->stream));
+    int src = c->cur, dst = c->cur ^ 1;
+    int grid = tile_grid(ntiles);
+    if (c->with_vals) {
+      TIMED(c, K_SCATTER,
+            hipLaunchKernelGGL(k_scatter_lb<true>, dim3(grid),
+                               dim3(NTHREADS_SC), 0, c->stream,
+                               c->keys[src], c->vals[src], c->keys[dst],
+                               c->vals[dst], n, ntiles, 8 * passes[i],
+                               c->d_dbase, c->d_lb, c->d_ticket, c->d_fail));
+    } else {
+      TIMED(c, K_SCATTER,
+            hipLaunchKernelGGL(k_scatter_lb<false>, dim3(grid),
+                               dim3(NTHREADS_SC), 0, c->stream,
+                               c->keys[src], nullptr, c->keys[dst], nullptr,
+                               n, ntiles, 8 * passes[i], c->d_dbase,
+                               c->d_lb, c->d_ticket, c->d_fail));
+    }
+    uint32_t fail = 0;
+    HIP_CHECK(hipMemcpyAsync(&fail, c->d_fail, 4, hipMemcpyDeviceToHost,
+                             c->stream));
+    HIP_CHECK(hipStreamSynchronize(c->stream));
+    if (fail) {
+      /* bounded-spin bailout: src buffers are intact — redo this pass on
+       * the decoupled 3-kernel path (never observed; safety net) */
+      if (radix_pass(c, 0, 8 * passes[i], 0xFF, 0)) return 1;
+    } else {
+      c->stats.bytes_scatter += n * (c->with_vals ? 32 : 16);
+      c->stats.bytes_total += n * (c->with_vals ? 32 : 16);
+      c->cur = dst;
+    }
+    c->hist_shift_valid = -1;
+    c->stats.sort_passes++;
+  }
+  if (drain_timing(c)) return 1;
+  c->sorted = 1;
+  return 0;
+}
+
 extern "C" int dprk_sort_pairs(dprk_ctx *c) {
   if (!c) { set_err("null ctx"); return 1; }
   HIP_CHECK(hipSetDevice(c->device));
@@ -1949,11 +2331,19 @@ extern "C" int dprk_sort_pairs(dprk_ctx *c) {
     c->stats.bytes_total += c->n * 8;
   }
   c->stats.sort_passes = 0;
+  int passes[8], np = 0;
   for (int p = 0; p < 8; p++) {
     uint32_t ob = (uint32_t)(orand[0] >> (8 * p)) & 0xFF;
     uint32_t ab = (uint32_t)(orand[1] >> (8 * p)) & 0xFF;
-    if (ob == ab) continue; /* constant digit across all keys: skip pass */
-    if (radix_pass(c, /*mode=*/0, /*shift=*/8 * p, /*mask=*/0xFF, 0))
+    if (ob != ab) passes[np++] = p;
+  }
+  /* default path: decoupled-lookback scatter (bit-identical output to
+   * the 3-kernel path — same permutation; interleaved A/B: 5.10 vs 5.33
+   * ms/step on config 2).  DPRK_NO_LOOKBACK restores the old path. */
+  if (np && !getenv("DPRK_NO_LOOKBACK"))
+    return sort_pairs_lookback(c, passes, np);
+  for (int i = 0; i < np; i++) {
+    if (radix_pass(c, /*mode=*/0, /*shift=*/8 * passes[i], /*mask=*/0xFF, 0))
       return 1;
     c->stats.sort_passes++;
   }
@@ -2122,6 +2512,7 @@ static int segmented_reduce_sorted(dprk_ctx *c, uint64_t *n_out) {
   c->n = total;
   c->cur = src;
   c->sorted = 1;
+  c->ghist_valid = 0;   /* dataset replaced */
   if (n_out) *n_out = total;
   return drain_timing(c);
 }
@@ -2403,6 +2794,7 @@ extern "C" int dprk_exchange(dprk_ctx *c, const uint64_t *send_counts,
   c->sorted = 0;
   c->orand_valid = 0;
   c->hist_shift_valid = -1;
+  c->ghist_valid = 0;
   return drain_timing(c);
 }
 
@@ -2482,6 +2874,7 @@ extern "C" int dprk_exchange_buckets(dprk_ctx *c,
   c->sorted = 0;
   c->orand_valid = 0;
   c->hist_shift_valid = -1;
+  c->ghist_valid = 0;
   return drain_timing(c);
 }
 
