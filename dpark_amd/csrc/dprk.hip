@@ -677,13 +677,12 @@ __global__ __launch_bounds__(NTHREADS) void k_tile_off(
     batch=4 measured 169 ms/launch vs 1.25 at batch=1 (parity held) */
 
 template <bool WITH_VALS>
-__global__ __launch_bounds__(NTHREADS_SC, 2) void k_scatter_lb(
+__global__ __launch_bounds__(NTHREADS_SC, 4) void k_scatter_lb(
     const int64_t *__restrict__ keys_in, const double *__restrict__ vals_in,
     int64_t *__restrict__ keys_out, double *__restrict__ vals_out,
     uint64_t n, uint64_t ntiles, int shift,
     const uint32_t *__restrict__ digit_base,
     unsigned long long *__restrict__ lb,
-    unsigned long long *__restrict__ ticket,
     unsigned int *__restrict__ d_fail) {
   constexpr int NW = NTHREADS_SC / 64;
   constexpr int WBLK = TILE / NW;
@@ -694,31 +693,25 @@ __global__ __launch_bounds__(NTHREADS_SC, 2) void k_scatter_lb(
   __shared__ uint32_t goff[RADIX];
   __shared__ uint32_t h[NW][RADIX];
   __shared__ uint32_t wtot[NW > 4 ? NW : 4];
-  __shared__ uint64_t t_sh;
 
   const int wave = threadIdx.x / 64, lane = threadIdx.x % 64;
   const uint64_t lowmask = ((uint64_t)1 << lane) - 1;
   const uint32_t mask = 0xFF;
 
-  for (;;) {
-    /* batched tickets (LB_BATCH consecutive tiles per grab): one shared
-     * ticket word saturates near 88 atomics/us (microarch dequeue row) —
-     * per-tile grabs cost ~0.28 ms/pass at 24k tiles.  Thread 0 folds
-     * the failure flag into the ticket so every thread sees the same
-     * decision (a divergent return before a barrier would strand the
-     * workgroup). */
-    if (threadIdx.x == 0) {
-      t_sh = __hip_atomic_load(d_fail, __ATOMIC_RELAXED,
-                               __HIP_MEMORY_SCOPE_AGENT)
-                 ? ntiles
-                 : __hip_atomic_fetch_add(ticket, LB_BATCH, __ATOMIC_RELAXED,
-                                          __HIP_MEMORY_SCOPE_AGENT);
-    }
+  /* STATIC issue-ordered assignment: block b owns tiles b, b+R, b+2R...
+   * with grid R <= true residency (host derives R from the occupancy
+   * query, clamped), so every predecessor tile is finished or being
+   * processed by a RESIDENT block — progress is guaranteed without a
+   * ticket (a single ticket word saturates near 88 atomics/us: ~2.8 ms
+   * per pass at 244k tiles; a batched ticket delays AGG publication,
+   * measured 130x worse — both in git history). */
+  for (uint64_t t = blockIdx.x; t < ntiles; t += gridDim.x) {
+    /* uniform failure bailout (reuse wtot[0]) */
+    if (threadIdx.x == 0)
+      wtot[0] = __hip_atomic_load(d_fail, __ATOMIC_RELAXED,
+                                  __HIP_MEMORY_SCOPE_AGENT);
     __syncthreads();
-    const uint64_t t0_batch = t_sh;
-    __syncthreads();
-    if (t0_batch >= ntiles) return;
-    for (uint64_t t = t0_batch; t < min(t0_batch + LB_BATCH, ntiles); t++) {
+    if (wtot[0]) return;
     const uint64_t start = t * (uint64_t)TILE;
     const uint64_t end = min(start + TILE, n);
     const uint32_t tile_n = (uint32_t)(end - start);
@@ -743,7 +736,11 @@ __global__ __launch_bounds__(NTHREADS_SC, 2) void k_scatter_lb(
     __syncthreads();
 
     /* phase 2: tile counts -> publish AGG granule; local digit excl scan
-     * (shfl) + per-wave bases in place */
+     * (shfl) + per-wave bases in place.  The LOOKBACK runs after phase 3
+     * (LDS stage) — measured better than the eager onesweep ordering on
+     * this chip (walking right after publish spins on predecessors that
+     * have not reached their publish point yet: +0.24/+8 ms per pass at
+     * 1e8/1e9; the extra phase-3 time lets the generation's AGGs land). */
     uint32_t tile_cnt = 0;
     if (threadIdx.x < RADIX) {
       for (int w = 0; w < NW; w++) tile_cnt += h[w][threadIdx.x];
@@ -838,7 +835,6 @@ __global__ __launch_bounds__(NTHREADS_SC, 2) void k_scatter_lb(
       if (WITH_VALS) vals_out[dst] = lv[idx];
     }
     __syncthreads();
-    }  /* batch loop */
   }
 }
 
@@ -2266,24 +2262,38 @@ static int sort_pairs_lookback(dprk_ctx *c, const int *passes, int np) {
     HIP_CHECK(hipMemcpyAsync(c->d_dbase, base, RADIX * 4,
                              hipMemcpyHostToDevice, c->stream));
     HIP_CHECK(hipMemsetAsync(c->d_lb, 0, ntiles * RADIX * 8, c->stream));
-    HIP_CHECK(hipMemsetAsync(c->d_ticket, 0, 8, c->stream));
     HIP_CHECK(hipMemsetAsync(c->d_fail, 0, 4, c->stream));
     int src = c->cur, dst = c->cur ^ 1;
-    int grid = tile_grid(ntiles);
+    /* grid = guaranteed-resident block count (static issue-ordered
+     * assignment needs every block resident; the occupancy API can
+     * over-report by one for SGPR-heavy kernels — clamp to 2/CU, which
+     * the 74 KiB LDS independently bounds) */
+    static int occ_v = 0, occ_k = 0;
+    if (!occ_v) {
+      HIP_CHECK(hipOccupancyMaxActiveBlocksPerMultiprocessor(
+          &occ_v, (const void *)k_scatter_lb<true>, NTHREADS_SC, 0));
+      HIP_CHECK(hipOccupancyMaxActiveBlocksPerMultiprocessor(
+          &occ_k, (const void *)k_scatter_lb<false>, NTHREADS_SC, 0));
+      hipDeviceProp_t props;
+      HIP_CHECK(hipGetDeviceProperties(&props, c->device));
+      occ_v = min(2, max(1, occ_v)) * props.multiProcessorCount;
+      occ_k = min(2, max(1, occ_k)) * props.multiProcessorCount;
+    }
+    int grid = (int)min(ntiles, (uint64_t)(c->with_vals ? occ_v : occ_k));
     if (c->with_vals) {
       TIMED(c, K_SCATTER,
             hipLaunchKernelGGL(k_scatter_lb<true>, dim3(grid),
                                dim3(NTHREADS_SC), 0, c->stream,
                                c->keys[src], c->vals[src], c->keys[dst],
                                c->vals[dst], n, ntiles, 8 * passes[i],
-                               c->d_dbase, c->d_lb, c->d_ticket, c->d_fail));
+                               c->d_dbase, c->d_lb, c->d_fail));
     } else {
       TIMED(c, K_SCATTER,
             hipLaunchKernelGGL(k_scatter_lb<false>, dim3(grid),
                                dim3(NTHREADS_SC), 0, c->stream,
                                c->keys[src], nullptr, c->keys[dst], nullptr,
                                n, ntiles, 8 * passes[i], c->d_dbase,
-                               c->d_lb, c->d_ticket, c->d_fail));
+                               c->d_lb, c->d_fail));
     }
     uint32_t fail = 0;
     HIP_CHECK(hipMemcpyAsync(&fail, c->d_fail, 4, hipMemcpyDeviceToHost,
@@ -2340,7 +2350,13 @@ extern "C" int dprk_sort_pairs(dprk_ctx *c) {
   /* default path: decoupled-lookback scatter (bit-identical output to
    * the 3-kernel path — same permutation; interleaved A/B: 5.10 vs 5.33
    * ms/step on config 2).  DPRK_NO_LOOKBACK restores the old path. */
-  if (np && !getenv("DPRK_NO_LOOKBACK"))
+  /* the granule array must stay L3-resident (256 MiB): past it every
+   * lookback hop is an HBM-random load and the walk cost explodes
+   * (measured: 1B keys, 500 MB granules -> +4.3 ms/pass).  The gate is
+   * a pure function of n => path choice stays input-deterministic. */
+  uint64_t lb_tiles = ceil_div(c->n, (uint64_t)TILE);
+  if (np && !getenv("DPRK_NO_LOOKBACK") &&
+      lb_tiles * RADIX * 8 <= (200ull << 20))
     return sort_pairs_lookback(c, passes, np);
   for (int i = 0; i < np; i++) {
     if (radix_pass(c, /*mode=*/0, /*shift=*/8 * passes[i], /*mask=*/0xFF, 0))
