@@ -2018,6 +2018,11 @@ extern "C" int dprk_gen_pairs(dprk_ctx *c, uint64_t seed, uint64_t n,
       if (keymask == ~0ull || mb)
         shifts |= ((uint64_t)(8 * p)) << (8 * (nact++));
     }
+    /* the global hists feed only the lookback path; when n exceeds its
+     * L3-residency gate (see dprk_sort_pairs) skip the extra LDS-atomic
+     * work here */
+    if (ceil_div(n, (uint64_t)TILE) * RADIX * 8 > (200ull << 20))
+      nact = 0;
     if (!c->d_ghist) HIP_CHECK(hipMalloc(&c->d_ghist, 8 * RADIX * 4));
     HIP_CHECK(hipMemsetAsync(c->d_ghist, 0, 8 * RADIX * 4, c->stream));
     uint64_t ntiles = ceil_div(n, (uint64_t)TILE);
