@@ -125,7 +125,6 @@ struct dprk_ctx {
   int hist_shift_valid = -1; /* tile_hist currently holds this pass shift */
   /* lookback-scatter scratch (DPRK_LOOKBACK) */
   unsigned long long *d_lb = nullptr;   /* [ntiles*RADIX] granules */
-  unsigned long long *d_ticket = nullptr;
   unsigned int *d_fail = nullptr;
   uint32_t *d_ghist = nullptr;          /* [8][RADIX] global pass hists */
   uint64_t ghist_shifts = 0;            /* packed shifts the hists cover */
@@ -672,9 +671,6 @@ __global__ __launch_bounds__(NTHREADS) void k_tile_off(
 #define LB_INC  (2ull << 62)
 #define LB_VAL  ((1ull << 62) - 1)
 #define LB_SPIN_LIMIT (1u << 16)
-#define LB_BATCH 1ull /* batch>1 delays AGG publication by batch-1 tile
-    times, degenerating successors' lookbacks into long AGG walks:
-    batch=4 measured 169 ms/launch vs 1.25 at batch=1 (parity held) */
 
 template <bool WITH_VALS>
 __global__ __launch_bounds__(NTHREADS_SC, 4) void k_scatter_lb(
@@ -1875,11 +1871,10 @@ static int free_all(dprk_ctx *c) {
     c->keys[i] = nullptr; c->vals[i] = nullptr;
   }
   if (c->d_lb) HIP_CHECK(hipFree(c->d_lb));
-  if (c->d_ticket) HIP_CHECK(hipFree(c->d_ticket));
   if (c->d_fail) HIP_CHECK(hipFree(c->d_fail));
   if (c->d_ghist) HIP_CHECK(hipFree(c->d_ghist));
   if (c->d_dbase) HIP_CHECK(hipFree(c->d_dbase));
-  c->d_lb = nullptr; c->d_ticket = nullptr; c->d_fail = nullptr;
+  c->d_lb = nullptr; c->d_fail = nullptr;
   c->d_ghist = nullptr; c->d_dbase = nullptr; c->lb_cap_tiles = 0;
   if (c->tile_hist) HIP_CHECK(hipFree(c->tile_hist));
   if (c->tile_off) HIP_CHECK(hipFree(c->tile_off));
@@ -2208,7 +2203,6 @@ static int ensure_lookback(dprk_ctx *c, uint64_t ntiles) {
   if (ntiles <= c->lb_cap_tiles && c->d_lb) return 0;
   if (c->d_lb) HIP_CHECK(hipFree(c->d_lb));
   HIP_CHECK(hipMalloc(&c->d_lb, ntiles * RADIX * 8));
-  if (!c->d_ticket) HIP_CHECK(hipMalloc(&c->d_ticket, 8));
   if (!c->d_fail) HIP_CHECK(hipMalloc(&c->d_fail, 4));
   if (!c->d_ghist) HIP_CHECK(hipMalloc(&c->d_ghist, 8 * RADIX * 4));
   if (!c->d_dbase) HIP_CHECK(hipMalloc(&c->d_dbase, RADIX * 4));
