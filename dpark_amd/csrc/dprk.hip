@@ -345,10 +345,13 @@ __global__ __launch_bounds__(NTHREADS) void k_gen_hist(
     int64_t *__restrict__ keys, double *__restrict__ vals, uint64_t n,
     uint64_t seed_hi, uint64_t keymask, uint32_t *__restrict__ tile_hist,
     uint64_t ntiles, int shift, uint64_t active_shifts, int nact,
-    uint32_t *__restrict__ g_hist /* [nact][RADIX]; lookback path */) {
+    uint32_t *__restrict__ g_hist /* [nact][RADIX]; lookback path */,
+    uint64_t *__restrict__ orand /* [2] = {OR, AND} of biased keys */) {
   __shared__ uint32_t h[NTHREADS / 64][RADIX];
   __shared__ uint32_t gh[NTHREADS / 64][8][RADIX]; /* 32 KiB */
+  __shared__ uint64_t s_or[NTHREADS / 64], s_and[NTHREADS / 64];
   const int wave = threadIdx.x / 64;
+  uint64_t acc_or = 0, acc_and = ~0ull;
   for (int i = threadIdx.x; i < (NTHREADS / 64) * 8 * RADIX; i += NTHREADS)
     ((uint32_t *)gh)[i] = 0;
   for (uint64_t t = blockIdx.x; t < ntiles; t += gridDim.x) {
@@ -364,6 +367,8 @@ __global__ __launch_bounds__(NTHREADS) void k_gen_hist(
       kk.x = (int64_t)(d_splitmix64(x0) & keymask);
       kk.y = (int64_t)(d_splitmix64(x1) & keymask);
       *reinterpret_cast<longlong2 *>(&keys[i]) = kk;
+      acc_or |= ((uint64_t)kk.x ^ SIGNFLIP) | ((uint64_t)kk.y ^ SIGNFLIP);
+      acc_and &= ((uint64_t)kk.x ^ SIGNFLIP) & ((uint64_t)kk.y ^ SIGNFLIP);
       atomicAdd(&h[wave][d_digit(kk.x, 0, shift, 0xFF)], 1u);
       atomicAdd(&h[wave][d_digit(kk.y, 0, shift, 0xFF)], 1u);
       for (int p = 0; p < nact; p++) {
@@ -385,6 +390,8 @@ __global__ __launch_bounds__(NTHREADS) void k_gen_hist(
       uint64_t x = seed_hi + i;
       int64_t k = (int64_t)(d_splitmix64(x) & keymask);
       keys[i] = k;
+      acc_or |= (uint64_t)k ^ SIGNFLIP;
+      acc_and &= (uint64_t)k ^ SIGNFLIP;
       atomicAdd(&h[wave][d_digit(k, 0, shift, 0xFF)], 1u);
       for (int p = 0; p < nact; p++) {
         int sh = (int)((active_shifts >> (8 * p)) & 0xFF);
@@ -410,6 +417,21 @@ __global__ __launch_bounds__(NTHREADS) void k_gen_hist(
         for (int w = 0; w < NTHREADS / 64; w++) s += gh[w][p][d];
         if (s) atomicAdd(&g_hist[p * RADIX + d], s);
       }
+  /* exact biased-key OR/AND -> digit-skip without a k_minmax_or read */
+  for (int off = 32; off; off >>= 1) {
+    acc_or |= (uint64_t)__shfl_down((long long)acc_or, off, 64);
+    acc_and &= (uint64_t)__shfl_down((long long)acc_and, off, 64);
+  }
+  if (threadIdx.x % 64 == 0) { s_or[wave] = acc_or; s_and[wave] = acc_and; }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    for (int w = 1; w < NTHREADS / 64; w++) {
+      acc_or |= s_or[w];
+      acc_and &= s_and[w];
+    }
+    atomicOr((unsigned long long *)&orand[0], (unsigned long long)acc_or);
+    atomicAnd((unsigned long long *)&orand[1], (unsigned long long)acc_and);
+  }
 }
 
 /* Zipf (or any discrete distribution) sampling via a caller-provided
@@ -2020,6 +2042,9 @@ extern "C" int dprk_gen_pairs(dprk_ctx *c, uint64_t seed, uint64_t n,
       nact = 0;
     if (!c->d_ghist) HIP_CHECK(hipMalloc(&c->d_ghist, 8 * RADIX * 4));
     HIP_CHECK(hipMemsetAsync(c->d_ghist, 0, 8 * RADIX * 4, c->stream));
+    uint64_t oa_init[2] = {0, ~0ull};
+    HIP_CHECK(hipMemcpyAsync(c->d_orand, oa_init, 16, hipMemcpyHostToDevice,
+                             c->stream));
     uint64_t ntiles = ceil_div(n, (uint64_t)TILE);
     TIMED(c, K_GEN,
           hipLaunchKernelGGL(k_gen_hist,
@@ -2027,21 +2052,29 @@ extern "C" int dprk_gen_pairs(dprk_ctx *c, uint64_t seed, uint64_t n,
                              dim3(NTHREADS), 0, c->stream, c->keys[c->cur],
                              c->with_vals ? c->vals[c->cur] : nullptr, n,
                              seed << 32, keymask, c->tile_hist, ntiles,
-                             shift0, shifts, nact, c->d_ghist));
+                             shift0, shifts, nact, c->d_ghist, c->d_orand));
+    uint64_t oa[2];
+    HIP_CHECK(hipMemcpyAsync(oa, c->d_orand, 16, hipMemcpyDeviceToHost,
+                             c->stream));
+    HIP_CHECK(hipStreamSynchronize(c->stream));
     c->hist_shift_valid = shift0;
     c->ghist_shifts = shifts;
     c->ghist_nact = nact;
     c->ghist_valid = 1;
+    /* EXACT biased-key bounds from generation: digit-skip needs no
+     * separate k_minmax_or read (matters for full-universe keys, where
+     * the conservative mask bound is useless) */
+    c->orand_valid = 1;
+    c->known_or = oa[0];
+    c->known_and = oa[1];
   }
   c->n = n;
   c->sorted = 0;
   /* keys = splitmix64 & keymask, so biased keys lie in
    * [SIGNFLIP, SIGNFLIP | keymask]: conservative OR/AND bound — may only
    * RUN an avoidable pass, never skip a needed one */
-  if (dist == 0 && keymask != ~0ull) {
-    c->orand_valid = 1;
-    c->known_or = SIGNFLIP | keymask;
-    c->known_and = SIGNFLIP;
+  if (dist == 0) {
+    /* exact values were read back from k_gen_hist above */
   } else if (dist == 1 && (key_universe & (key_universe - 1)) == 0) {
     c->orand_valid = 1;   /* alias keys are < universe (power of two) */
     c->known_or = SIGNFLIP | (key_universe - 1);
