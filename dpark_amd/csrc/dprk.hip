@@ -13,7 +13,9 @@
  *   k_minmax_or      OR/AND reduce of biased keys -> digit-skip detection
  *   k_tile_hist      per-tile 256-bin digit histogram      (8 B/pair read)
  *   k_seg_sums/k_scan_mid/k_tile_off   3-level scan of tile histograms
- *   k_scatter        rank + LDS exchange + coalesced write (32 B/pair r+w)
+ *   k_scatter / k_scatter_lb  rank + LDS exchange + coalesced write
+ *       (32 B/pair r+w); _lb (the default sort path) adds in-kernel
+ *       decoupled-lookback offsets, dropping tile_hist + the scans
  *   k_sample / mode-2 digits           device RangePartitioner (sortByKey)
  *   k_reduce_tiles / k_stitchA-C / k_compact_runs  segmented f64 reduce
  *   k_scatter_reduce / k_fused_gather  fused last-pass reduce (opt-in)
