@@ -88,12 +88,25 @@ def main():
         run_config1()
         return
 
+    if args.gpus > 1 and "WORLD_SIZE" not in os.environ:
+        # Self-bootstrap: plain `python bench.py
+        # --gpus N` relaunches itself under torch.distributed.run, one
+        # rank per GPU, so the driver's literal command measures N GPUs
+        # instead of silently clamping to 1.
+        port = 29400 + (os.getpid() % 1000)
+        cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+               f"--nproc-per-node={args.gpus}", "--master-addr=127.0.0.1",
+               f"--master-port={port}",
+               os.path.abspath(__file__)] + sys.argv[1:]
+        log(f"bench: bootstrapping {args.gpus} ranks via torch.distributed.run")
+        os.execv(sys.executable, cmd)
+
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
     nranks = max(world, 1)
     if args.gpus != nranks and world == 1:
-        nranks = 1  # plain `python bench.py --gpus N` without torchrun: N=1
+        nranks = 1  # torchrun-launched with WORLD_SIZE=1: honest N=1
 
     dist = None
     torch = None

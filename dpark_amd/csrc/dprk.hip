@@ -2805,6 +2805,10 @@ extern "C" int dprk_exchange(dprk_ctx *c, const uint64_t *send_counts,
   if (nranks == 1 && !getenv("DPRK_FORCE_SELF_EXCHANGE")) return 0;
   if (!c->comm) { set_err("exchange: comm not initialized"); return 1; }
   if (nranks != c->nranks) { set_err("exchange: nranks mismatch"); return 1; }
+  if (nranks < 1 || nranks > 256) {
+    set_err("exchange: nranks %d out of range [1,256]", nranks);
+    return 1;
+  }
   HIP_CHECK(hipSetDevice(c->device));
   uint64_t send_off[256], recv_off[256], total_recv = 0, total_send = 0;
   for (int r = 0; r < nranks; r++) {

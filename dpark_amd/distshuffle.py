@@ -16,9 +16,7 @@ def rank_of_keys(keys, nranks: int):
     splitmix64(key) (d_digit mode 1 in dpark_amd/csrc/dprk.hip).  The GPU
     HashPartitioner — layout intentionally differs from reference
     portable_hash."""
-    from oracle.synth import splitmix64  # test-infra import is fine here:
-    # the splitmix64 SPEC is normative and shared; the engine never calls
-    # this function (it has its own device implementation).
+    from dpark_amd.hashspec import splitmix64
     assert nranks >= 1 and (nranks & (nranks - 1)) == 0
     if nranks == 1:
         return np.zeros(len(keys), dtype=np.int64)

@@ -37,6 +37,27 @@ def test_torchrun_launch_line(op, n):
     assert d["config"]["parallelism"] == f"dp{n}"
 
 
+@pytest.mark.timeout(300)
+def test_plain_gpus_flag_self_bootstraps():
+    """`python bench.py --gpus N` WITHOUT torchrun must
+    bootstrap N ranks itself (exec torch.distributed.run), not silently
+    measure N=1."""
+    pytest.importorskip("torch")
+    env = dict(os.environ)
+    env["DPRK_BENCH_SIM"] = "1"
+    env.pop("WORLD_SIZE", None)
+    cmd = [sys.executable, "bench.py", "--gpus", "2", "--steps", "2",
+           "--warmup", "1", "--op", "reduce", "--n-pairs", "200000",
+           "--skip-cpu-baseline"]
+    out = subprocess.run(cmd, cwd=ROOT, env=env, capture_output=True,
+                         text=True, timeout=280)
+    assert out.returncode == 0, out.stderr[-3000:]
+    d = json.loads([ln for ln in out.stdout.splitlines()
+                    if ln.startswith("{")][-1])
+    assert d["n_gpus"] == 2
+    assert d["config"]["parallelism"] == "dp2"
+
+
 @pytest.mark.timeout(120)
 def test_single_rank_sim():
     env = dict(os.environ)

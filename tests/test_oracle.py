@@ -91,6 +91,18 @@ def test_synth_spec_stability():
     assert np.array_equal(k, k2) and np.array_equal(v, v2)
 
 
+def test_product_hash_mirror_agrees_with_oracle_spec():
+    """dpark_amd.hashspec.splitmix64 (product copy, used by the multi-rank
+    plan) must match oracle.synth.splitmix64 (independent test copy) —
+    guards drift between the two deliberate copies of the normative spec."""
+    from dpark_amd.hashspec import splitmix64 as product_mix
+    x = np.concatenate([
+        np.arange(1000, dtype=np.uint64),
+        np.array([0, 1, 2**63, 2**64 - 1, 0xDEADBEEF], dtype=np.uint64),
+    ])
+    assert np.array_equal(product_mix(x), splitmix64(x))
+
+
 def test_group_by_key_multiset():
     pairs = [(1, 10.0), (2, 20.0), (1, 30.0), (1, 10.0)]
     g = dpo.group_by_key(pairs)
