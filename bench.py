@@ -169,6 +169,7 @@ def main():
         # bucket (hence every key) lands wholly on one rank.
         if nranks > 1:
             from dpark_amd.distshuffle import greedy_assign, recv_matrix_for
+            e.sync_packed(dist)  # all ranks must ship the same key width
             nbuckets = 256
             send = e.partition(nbuckets)
             counts = torch.tensor(send.astype("int64"))
@@ -197,6 +198,7 @@ def main():
             if nranks > 1:
                 # distributed sortByKey: sampled range splitters (dpark
                 # RangePartitioner semantics) -> all-to-allv -> local sort
+                eng.sync_packed(dist)
                 sample = eng.sample_keys(4096)
                 obj = [None] * nranks
                 dist.all_gather_object(obj, sample)

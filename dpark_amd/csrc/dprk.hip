@@ -119,6 +119,8 @@ struct dprk_ctx {
   uint64_t cap = 0;
   int with_vals = 0;
   int sorted = 0;
+  int packed = 0;   /* keys currently stored as packed u32 (all keys in
+                       [0, 2^32); input-deterministic — see KT traits) */
   /* conservative OR/AND of biased keys known from generation (digit-skip
    * without the k_minmax_or pass); valid only straight after gen /
    * permutation-only ops */
@@ -249,6 +251,61 @@ static int drain_timing(dprk_ctx *c) {
 }
 
 /* ------------------------------------------------------------------ */
+/* key-type traits: the PACKED-u32 fast path                           */
+/*                                                                     */
+/* When every key of a dataset lies in [0, 2^32) — detected input-     */
+/* deterministically from the biased OR/AND (gen universe <= 2^32, or  */
+/* k_minmax_or at put time) — keys travel as packed uint32 through the */
+/* whole pipeline: 24 B/pair per scatter pass instead of 32 (-25%),    */
+/* and the reduce reads 12 B/pair instead of 16.  The C-ABI stays      */
+/* int64 (put/get convert); sorting u32 unsigned == int64 ascending    */
+/* because packed keys are non-negative.  Both instantiations are the  */
+/* SAME kernels — no behavioral fork, only the key element type.       */
+/* ------------------------------------------------------------------ */
+template <typename K> struct KT;
+template <> struct KT<int64_t> {
+  using V2 = longlong2;                       /* 2 keys = 16 B          */
+  static constexpr int KSZ = 8;
+  static __device__ __forceinline__ uint64_t bias(int64_t k) {
+    return (uint64_t)k ^ SIGNFLIP;            /* signed -> unsigned order */
+  }
+  static __device__ __forceinline__ uint64_t raw(int64_t k) {
+    return (uint64_t)k;
+  }
+  static __device__ __forceinline__ int64_t from_raw(uint64_t u) {
+    return (int64_t)u;
+  }
+};
+/* longlong2's members are `long long` while int64_t is `long` on LP64 —
+ * an identical specialization keeps vector-element digit calls typed */
+template <> struct KT<long long> {
+  using V2 = longlong2;
+  static constexpr int KSZ = 8;
+  static __device__ __forceinline__ uint64_t bias(long long k) {
+    return (uint64_t)k ^ SIGNFLIP;
+  }
+  static __device__ __forceinline__ uint64_t raw(long long k) {
+    return (uint64_t)k;
+  }
+  static __device__ __forceinline__ long long from_raw(uint64_t u) {
+    return (long long)u;
+  }
+};
+template <> struct KT<uint32_t> {
+  using V2 = uint2;                           /* 2 keys = 8 B           */
+  static constexpr int KSZ = 4;
+  static __device__ __forceinline__ uint64_t bias(uint32_t k) {
+    return (uint64_t)k;                       /* non-negative: no flip  */
+  }
+  static __device__ __forceinline__ uint64_t raw(uint32_t k) {
+    return (uint64_t)k;                       /* == the int64 key value */
+  }
+  static __device__ __forceinline__ uint32_t from_raw(uint64_t u) {
+    return (uint32_t)u;
+  }
+};
+
+/* ------------------------------------------------------------------ */
 /* device utilities                                                    */
 /* ------------------------------------------------------------------ */
 __device__ __forceinline__ uint64_t d_splitmix64(uint64_t x) {
@@ -267,21 +324,24 @@ __device__ __forceinline__ uint64_t d_splitmix64(uint64_t x) {
  * #{j : splitters[j] < key} + (key present ? its first index) — we use
  * rank = #{j : key >= splitters[j]}, which equals bisect_right; any
  * consistent monotone rule yields a valid total order across ranks). */
+template <typename K>
 __device__ __forceinline__ uint32_t d_digit_s(
-    int64_t key, int mode, int shift, uint32_t mask,
+    K key, int mode, int shift, uint32_t mask,
     const int64_t *__restrict__ splitters, int nsplit) {
   if (mode == 2) {
+    int64_t k = (int64_t)KT<K>::raw(key);
     uint32_t r = 0;
-    for (int j = 0; j < nsplit; j++) r += (key >= splitters[j]) ? 1u : 0u;
+    for (int j = 0; j < nsplit; j++) r += (k >= splitters[j]) ? 1u : 0u;
     return r;
   }
-  uint64_t u = mode ? d_splitmix64((uint64_t)key) : ((uint64_t)key ^ SIGNFLIP);
+  uint64_t u = mode ? d_splitmix64(KT<K>::raw(key)) : KT<K>::bias(key);
   return (uint32_t)(u >> shift) & mask;
 }
 
-__device__ __forceinline__ uint32_t d_digit(int64_t key, int mode, int shift,
+template <typename K>
+__device__ __forceinline__ uint32_t d_digit(K key, int mode, int shift,
                                             uint32_t mask) {
-  uint64_t u = mode ? d_splitmix64((uint64_t)key) : ((uint64_t)key ^ SIGNFLIP);
+  uint64_t u = mode ? d_splitmix64(KT<K>::raw(key)) : KT<K>::bias(key);
   return (uint32_t)(u >> shift) & mask;
 }
 
@@ -297,54 +357,41 @@ __device__ __forceinline__ uint32_t d_wave_incl_scan(uint32_t v, int lane) {
 
 /* strided sample of the current keys (sortByKey splitter sampling —
  * dpark samples partitions to the driver [U]) */
+template <typename K>
 __global__ __launch_bounds__(NTHREADS) void k_sample(
-    const int64_t *__restrict__ keys, uint64_t n, uint64_t stride_elems,
+    const K *__restrict__ keys, uint64_t n, uint64_t stride_elems,
     int64_t *__restrict__ out, uint64_t n_out) {
   uint64_t gstride = (uint64_t)gridDim.x * blockDim.x;
   for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
        i < n_out; i += gstride)
-    out[i] = keys[min(i * stride_elems, n - 1)];
+    out[i] = (int64_t)KT<K>::raw(keys[min(i * stride_elems, n - 1)]);
+}
+
+/* pack (int64 -> u32) / unpack (u32 -> int64); caller guarantees fit */
+__global__ __launch_bounds__(NTHREADS) void k_pack_keys(
+    const int64_t *__restrict__ in, uint32_t *__restrict__ out, uint64_t n) {
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    out[i] = (uint32_t)(uint64_t)in[i];
+}
+__global__ __launch_bounds__(NTHREADS) void k_unpack_keys(
+    const uint32_t *__restrict__ in, int64_t *__restrict__ out, uint64_t n) {
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    out[i] = (int64_t)(uint64_t)in[i];
 }
 
 /* ------------------------------------------------------------------ */
 /* kernels: generation                                                 */
 /* ------------------------------------------------------------------ */
-__global__ __launch_bounds__(NTHREADS) void k_gen(
-    int64_t *__restrict__ keys, double *__restrict__ vals, uint64_t n,
-    uint64_t seed_hi, uint64_t keymask) {
-  uint64_t stride = 2 * (uint64_t)gridDim.x * blockDim.x;
-  for (uint64_t i = 2 * ((uint64_t)blockIdx.x * blockDim.x + threadIdx.x);
-       i < n; i += stride) {
-    uint64_t x0 = seed_hi + i, x1 = seed_hi + i + 1;
-    int64_t k0 = (int64_t)(d_splitmix64(x0) & keymask);
-    if (i + 1 < n) { /* 16-B vectorized stores */
-      longlong2 kk;
-      kk.x = k0;
-      kk.y = (int64_t)(d_splitmix64(x1) & keymask);
-      *reinterpret_cast<longlong2 *>(&keys[i]) = kk;
-    } else {
-      keys[i] = k0;
-    }
-    if (vals) {
-      uint64_t v0 = d_splitmix64(x0 ^ 0xDEADBEEFCAFEBABEull);
-      if (i + 1 < n) {
-        uint64_t v1 = d_splitmix64(x1 ^ 0xDEADBEEFCAFEBABEull);
-        double2 vv;
-        vv.x = (double)(v0 >> 11) * (1.0 / 9007199254740992.0);
-        vv.y = (double)(v1 >> 11) * (1.0 / 9007199254740992.0);
-        *reinterpret_cast<double2 *>(&vals[i]) = vv;
-      } else {
-        vals[i] = (double)(v0 >> 11) * (1.0 / 9007199254740992.0);
-      }
-    }
-  }
-}
-
 /* tile-structured generator that also produces the per-tile digit-0
  * histogram (tile-major) — the first radix pass then skips its
  * k_tile_hist read (~4% of the config-2 step).  One WG per tile. */
+template <typename K>
 __global__ __launch_bounds__(NTHREADS) void k_gen_hist(
-    int64_t *__restrict__ keys, double *__restrict__ vals, uint64_t n,
+    K *__restrict__ keys, double *__restrict__ vals, uint64_t n,
     uint64_t seed_hi, uint64_t keymask, uint32_t *__restrict__ tile_hist,
     uint64_t ntiles, int shift, uint64_t active_shifts, int nact,
     uint32_t *__restrict__ g_hist /* [nact][RADIX]; lookback path */,
@@ -365,12 +412,12 @@ __global__ __launch_bounds__(NTHREADS) void k_gen_hist(
     for (uint64_t i = start + 2 * threadIdx.x; i + 1 < end;
          i += 2 * NTHREADS) {
       uint64_t x0 = seed_hi + i, x1 = seed_hi + i + 1;
-      longlong2 kk;
-      kk.x = (int64_t)(d_splitmix64(x0) & keymask);
-      kk.y = (int64_t)(d_splitmix64(x1) & keymask);
-      *reinterpret_cast<longlong2 *>(&keys[i]) = kk;
-      acc_or |= ((uint64_t)kk.x ^ SIGNFLIP) | ((uint64_t)kk.y ^ SIGNFLIP);
-      acc_and &= ((uint64_t)kk.x ^ SIGNFLIP) & ((uint64_t)kk.y ^ SIGNFLIP);
+      typename KT<K>::V2 kk;
+      kk.x = KT<K>::from_raw(d_splitmix64(x0) & keymask);
+      kk.y = KT<K>::from_raw(d_splitmix64(x1) & keymask);
+      *reinterpret_cast<typename KT<K>::V2 *>(&keys[i]) = kk;
+      acc_or |= KT<K>::bias(kk.x) | KT<K>::bias(kk.y);
+      acc_and &= KT<K>::bias(kk.x) & KT<K>::bias(kk.y);
       atomicAdd(&h[wave][d_digit(kk.x, 0, shift, 0xFF)], 1u);
       atomicAdd(&h[wave][d_digit(kk.y, 0, shift, 0xFF)], 1u);
       for (int p = 0; p < nact; p++) {
@@ -390,10 +437,10 @@ __global__ __launch_bounds__(NTHREADS) void k_gen_hist(
     if ((end - start) & 1 && threadIdx.x == 0) {
       uint64_t i = end - 1;
       uint64_t x = seed_hi + i;
-      int64_t k = (int64_t)(d_splitmix64(x) & keymask);
+      K k = KT<K>::from_raw(d_splitmix64(x) & keymask);
       keys[i] = k;
-      acc_or |= (uint64_t)k ^ SIGNFLIP;
-      acc_and &= (uint64_t)k ^ SIGNFLIP;
+      acc_or |= KT<K>::bias(k);
+      acc_and &= KT<K>::bias(k);
       atomicAdd(&h[wave][d_digit(k, 0, shift, 0xFF)], 1u);
       for (int p = 0; p < nact; p++) {
         int sh = (int)((active_shifts >> (8 * p)) & 0xFF);
@@ -442,8 +489,9 @@ __global__ __launch_bounds__(NTHREADS) void k_gen_hist(
  * else alias.  Bit-exact mirror: oracle/synth.py gen_zipf (the table is
  * BUILT host-side in numpy and shipped via dprk_set_alias_table, so no
  * cross-libm drift can break parity). */
+template <typename K>
 __global__ __launch_bounds__(NTHREADS) void k_gen_alias(
-    int64_t *__restrict__ keys, double *__restrict__ vals, uint64_t n,
+    K *__restrict__ keys, double *__restrict__ vals, uint64_t n,
     uint64_t seed_hi, const uint64_t *__restrict__ table, uint64_t universe) {
   uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
   for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
@@ -454,7 +502,7 @@ __global__ __launch_bounds__(NTHREADS) void k_gen_alias(
     uint64_t e = table[slot];
     uint64_t u2 = d_splitmix64(x ^ 0x14057B7EF767814Full);
     uint64_t k = ((u2 >> 32) < (e >> 32)) ? slot : (e & 0xFFFFFFFFull);
-    keys[i] = (int64_t)k;
+    keys[i] = KT<K>::from_raw(k);
     if (vals) {
       uint64_t vb = d_splitmix64(x ^ 0xDEADBEEFCAFEBABEull);
       vals[i] = (double)(vb >> 11) * (1.0 / 9007199254740992.0); /* 2^-53 */
@@ -466,8 +514,9 @@ __global__ __launch_bounds__(NTHREADS) void k_gen_alias(
  * the keys (lookback path: replaces per-pass k_tile_hist + the 3-level
  * scans).  active_shifts packs the byte shifts of up to 8 active passes;
  * per-wave LDS replicas, one global atomic merge per workgroup. */
+template <typename K>
 __global__ __launch_bounds__(NTHREADS) void k_hist_all(
-    const int64_t *__restrict__ keys, uint64_t n, uint64_t active_shifts,
+    const K *__restrict__ keys, uint64_t n, uint64_t active_shifts,
     int nact, uint32_t *__restrict__ g_hist /* [nact][RADIX] */) {
   __shared__ uint32_t h[NTHREADS / 64][8][RADIX];   /* 32 KiB */
   const int wave = threadIdx.x / 64;
@@ -477,9 +526,10 @@ __global__ __launch_bounds__(NTHREADS) void k_hist_all(
   const uint64_t stride = 2 * (uint64_t)gridDim.x * blockDim.x;
   for (uint64_t i = 2 * ((uint64_t)blockIdx.x * blockDim.x + threadIdx.x);
        i + 1 < n; i += stride) {
-    longlong2 kk = *reinterpret_cast<const longlong2 *>(&keys[i]);
-    uint64_t u0 = (uint64_t)kk.x ^ SIGNFLIP;
-    uint64_t u1 = (uint64_t)kk.y ^ SIGNFLIP;
+    typename KT<K>::V2 kk =
+        *reinterpret_cast<const typename KT<K>::V2 *>(&keys[i]);
+    uint64_t u0 = KT<K>::bias(kk.x);
+    uint64_t u1 = KT<K>::bias(kk.y);
     for (int p = 0; p < nact; p++) {
       int sh = (int)((active_shifts >> (8 * p)) & 0xFF);
       atomicAdd(&h[wave][p][(u0 >> sh) & 0xFF], 1u);
@@ -487,7 +537,7 @@ __global__ __launch_bounds__(NTHREADS) void k_hist_all(
     }
   }
   if ((n & 1) && blockIdx.x == 0 && threadIdx.x == 0) {
-    uint64_t u = (uint64_t)keys[n - 1] ^ SIGNFLIP;
+    uint64_t u = KT<K>::bias(keys[n - 1]);
     for (int p = 0; p < nact; p++) {
       int sh = (int)((active_shifts >> (8 * p)) & 0xFF);
       atomicAdd(&h[wave][p][(u >> sh) & 0xFF], 1u);
@@ -503,15 +553,16 @@ __global__ __launch_bounds__(NTHREADS) void k_hist_all(
 }
 
 /* OR/AND reduction of biased keys -> constant-digit (skip-pass) detection */
+template <typename K>
 __global__ __launch_bounds__(NTHREADS) void k_minmax_or(
-    const int64_t *__restrict__ keys, uint64_t n,
+    const K *__restrict__ keys, uint64_t n,
     uint64_t *__restrict__ orand /* [2] pre-set to {0, ~0} */) {
   __shared__ uint64_t s_or[NTHREADS / 64], s_and[NTHREADS / 64];
   uint64_t o = 0, a = ~0ull;
   uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
   for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += stride) {
-    uint64_t u = (uint64_t)keys[i] ^ SIGNFLIP;
+    uint64_t u = KT<K>::bias(keys[i]);
     o |= u;
     a &= u;
   }
@@ -540,8 +591,9 @@ __global__ __launch_bounds__(NTHREADS) void k_minmax_or(
  * workgroup writes one contiguous 1 KiB block — digit-major layout made
  * every 4 B store a partial-line write that each XCD's L2 evicted as a
  * full 64 B line (measured 10x write amplification, profiles/r01). */
+template <typename K>
 __global__ __launch_bounds__(NTHREADS) void k_tile_hist(
-    const int64_t *__restrict__ keys, uint64_t n,
+    const K *__restrict__ keys, uint64_t n,
     uint32_t *__restrict__ tile_hist, uint64_t ntiles, int mode, int shift,
     uint32_t mask, const int64_t *__restrict__ splitters, int nsplit) {
   __shared__ uint32_t h[NTHREADS / 64][RADIX];
@@ -552,15 +604,15 @@ __global__ __launch_bounds__(NTHREADS) void k_tile_hist(
     __syncthreads();
     uint64_t start = t * (uint64_t)TILE;
     uint64_t end = min(start + TILE, n);
-    /* 16-B loads: two keys per lane (8-B streams run at 0.54-0.70x the
+    /* vector loads: two keys per lane (8-B streams run at 0.54-0.70x the
      * 16-B rate on gfx950).  Full tiles take the constant-trip unrolled
      * path so all 8 loads issue up front (the variable-bound loop kept
      * only one load in flight: 95% wave-parked in the SQ profile). */
     if (end - start == TILE) {
-      longlong2 kk[TILE / (2 * NTHREADS)];
+      typename KT<K>::V2 kk[TILE / (2 * NTHREADS)];
 #pragma unroll
       for (int j = 0; j < TILE / (2 * NTHREADS); j++)
-        kk[j] = *reinterpret_cast<const longlong2 *>(
+        kk[j] = *reinterpret_cast<const typename KT<K>::V2 *>(
             &keys[start + 2 * threadIdx.x + (uint64_t)j * 2 * NTHREADS]);
 #pragma unroll
       for (int j = 0; j < TILE / (2 * NTHREADS); j++) {
@@ -569,7 +621,8 @@ __global__ __launch_bounds__(NTHREADS) void k_tile_hist(
       }
     } else {
       for (uint64_t i = start + 2 * threadIdx.x; i + 1 < end; i += 2 * NTHREADS) {
-        longlong2 kk = *reinterpret_cast<const longlong2 *>(&keys[i]);
+        typename KT<K>::V2 kk =
+            *reinterpret_cast<const typename KT<K>::V2 *>(&keys[i]);
         atomicAdd(&h[wave][d_digit_s(kk.x, mode, shift, mask, splitters, nsplit)], 1u);
         atomicAdd(&h[wave][d_digit_s(kk.y, mode, shift, mask, splitters, nsplit)], 1u);
       }
@@ -696,10 +749,10 @@ __global__ __launch_bounds__(NTHREADS) void k_tile_off(
 #define LB_VAL  ((1ull << 62) - 1)
 #define LB_SPIN_LIMIT (1u << 16)
 
-template <bool WITH_VALS>
+template <typename K, bool WITH_VALS>
 __global__ __launch_bounds__(NTHREADS_SC, 4) void k_scatter_lb(
-    const int64_t *__restrict__ keys_in, const double *__restrict__ vals_in,
-    int64_t *__restrict__ keys_out, double *__restrict__ vals_out,
+    const K *__restrict__ keys_in, const double *__restrict__ vals_in,
+    K *__restrict__ keys_out, double *__restrict__ vals_out,
     uint64_t n, uint64_t ntiles, int shift,
     const uint32_t *__restrict__ digit_base,
     unsigned long long *__restrict__ lb,
@@ -707,7 +760,7 @@ __global__ __launch_bounds__(NTHREADS_SC, 4) void k_scatter_lb(
   constexpr int NW = NTHREADS_SC / 64;
   constexpr int WBLK = TILE / NW;
   constexpr int EPL = WBLK / 64;
-  __shared__ int64_t lk[TILE];
+  __shared__ K lk[TILE];
   __shared__ double lv[WITH_VALS ? TILE : 1];
   __shared__ uint32_t excl[RADIX];
   __shared__ uint32_t goff[RADIX];
@@ -740,7 +793,7 @@ __global__ __launch_bounds__(NTHREADS_SC, 4) void k_scatter_lb(
     /* phase 1: load + wave-private histogram */
     for (int j = 0; j < RADIX / 64; j++) h[wave][lane * (RADIX / 64) + j] = 0;
     __syncthreads();
-    int64_t rk[EPL];
+    K rk[EPL];
     double rv[EPL];
     uint32_t rd[EPL];
 #pragma unroll
@@ -848,7 +901,7 @@ __global__ __launch_bounds__(NTHREADS_SC, 4) void k_scatter_lb(
 
     /* phase 4: write out in digit-sorted tile order */
     for (uint32_t idx = threadIdx.x; idx < tile_n; idx += NTHREADS_SC) {
-      int64_t key = lk[idx];
+      K key = lk[idx];
       uint32_t dig = d_digit(key, 0, shift, mask);
       uint32_t dst = goff[dig] + (idx - excl[dig]);
       keys_out[dst] = key;
@@ -858,10 +911,10 @@ __global__ __launch_bounds__(NTHREADS_SC, 4) void k_scatter_lb(
   }
 }
 
-template <bool WITH_VALS>
+template <typename K, bool WITH_VALS>
 __global__ __launch_bounds__(NTHREADS_SC, 2) void k_scatter(
-    const int64_t *__restrict__ keys_in, const double *__restrict__ vals_in,
-    int64_t *__restrict__ keys_out, double *__restrict__ vals_out,
+    const K *__restrict__ keys_in, const double *__restrict__ vals_in,
+    K *__restrict__ keys_out, double *__restrict__ vals_out,
     uint64_t n, const uint32_t *__restrict__ tile_hist,
     const uint32_t *__restrict__ tile_off, uint64_t ntiles, int mode,
     int shift, uint32_t mask, const int64_t *__restrict__ splitters,
@@ -869,7 +922,7 @@ __global__ __launch_bounds__(NTHREADS_SC, 2) void k_scatter(
   constexpr int NW = NTHREADS_SC / 64;     /* 8 waves per WG              */
   constexpr int WBLK = TILE / NW;          /* 512 elements per wave       */
   constexpr int EPL = WBLK / 64;           /* 8 elements per lane         */
-  __shared__ int64_t lk[TILE];
+  __shared__ K lk[TILE];
   __shared__ double lv[WITH_VALS ? TILE : 1];
   __shared__ uint32_t excl[RADIX];     /* tile-local digit excl scan  */
   __shared__ uint32_t goff[RADIX];     /* global write base per digit */
@@ -916,7 +969,7 @@ __global__ __launch_bounds__(NTHREADS_SC, 2) void k_scatter(
     }
 
     /* phase 1: load into registers + wave-private histogram */
-    int64_t rk[EPL];
+    K rk[EPL];
     double rv[EPL];
     uint32_t rd[EPL];
 #pragma unroll
@@ -992,7 +1045,7 @@ __global__ __launch_bounds__(NTHREADS_SC, 2) void k_scatter(
 
     /* phase 4: write out in digit-sorted tile order */
     for (uint32_t idx = threadIdx.x; idx < tile_n; idx += NTHREADS_SC) {
-      int64_t key = lk[idx];
+      K key = lk[idx];
       uint32_t dig = d_digit_s(key, mode, shift, mask, splitters, nsplit);
       uint32_t dst = goff[dig] + (idx - excl[dig]);
       keys_out[dst] = key;
@@ -1016,10 +1069,10 @@ __global__ __launch_bounds__(NTHREADS_SC, 2) void k_scatter(
 /* but the 36 B/segment record traffic ate the win; git history.)      */
 /* ------------------------------------------------------------------ */
 
-template <bool DUMMY>
+template <typename K>
 __global__ __launch_bounds__(NTHREADS_SC, 4) void k_scatter_reduce(
-    const int64_t *__restrict__ keys_in, const double *__restrict__ vals_in,
-    int64_t *__restrict__ keys_scr, double *__restrict__ sums_scr,
+    const K *__restrict__ keys_in, const double *__restrict__ vals_in,
+    K *__restrict__ keys_scr, double *__restrict__ sums_scr,
     uint64_t n, const uint32_t *__restrict__ tile_hist, uint64_t ntiles,
     int shift, uint32_t *__restrict__ em_counts /* [t*RADIX+d] */,
     uint32_t *__restrict__ em_excl /* [t*RADIX+d] tile-local base */) {
@@ -1028,7 +1081,7 @@ __global__ __launch_bounds__(NTHREADS_SC, 4) void k_scatter_reduce(
   constexpr int EPL = WBLK / 64;
   constexpr int EPT = TILE / NTHREADS_SC;
 #define FPIDX(j) ((j) + ((j) >> 3))
-  __shared__ int64_t lk[TILE + TILE / 8];
+  __shared__ K lk[TILE + TILE / 8];
   __shared__ uint16_t pmap[TILE + TILE / 8];
   __shared__ uint32_t excl[RADIX];
   __shared__ uint32_t h[NW][RADIX];
@@ -1064,7 +1117,7 @@ __global__ __launch_bounds__(NTHREADS_SC, 4) void k_scatter_reduce(
       for (int w = 0; w < wave; w++) add += wtot[w];
       excl[threadIdx.x] += add;
     }
-    int64_t rk[EPL];
+    K rk[EPL];
     uint32_t rd[EPL];
 #pragma unroll
     for (int i = 0; i < EPL; i++) {
@@ -1123,12 +1176,12 @@ __global__ __launch_bounds__(NTHREADS_SC, 4) void k_scatter_reduce(
     uint32_t has_head = 0, n_ends = 0;
     if (c0 < tile_n) {
       bool have_prev = c0 > 0;
-      int64_t prev_key = have_prev ? KEYAT(c0 - 1) : 0;
+      K prev_key = have_prev ? KEYAT(c0 - 1) : (K)0;
 #pragma unroll
       for (int j = 0; j < EPT; j++) {
         uint32_t p = c0 + j;
         if (p >= tile_n) break;
-        int64_t k = KEYAT(p);
+        K k = KEYAT(p);
         bool head = !have_prev || k != prev_key;
         if (head) { open = cv[j]; has_head = 1; }
         else open += cv[j];
@@ -1171,12 +1224,12 @@ __global__ __launch_bounds__(NTHREADS_SC, 4) void k_scatter_reduce(
       if (c0 < tile_n) {
         double running = incoming;
         bool have_prev = c0 > 0;
-        int64_t prev_key = have_prev ? KEYAT(c0 - 1) : 0;
+        K prev_key = have_prev ? KEYAT(c0 - 1) : (K)0;
 #pragma unroll
         for (int j = 0; j < EPT; j++) {
           uint32_t p = c0 + j;
           if (p >= tile_n) break;
-          int64_t k = KEYAT(p);
+          K k = KEYAT(p);
           bool head = !have_prev || k != prev_key;
           if (head) running = cv[j];
           else running += cv[j];
@@ -1212,12 +1265,13 @@ __global__ __launch_bounds__(NTHREADS_SC, 4) void k_scatter_reduce(
 }
 
 /* gather tile-local fragments into the d-major fragment stream */
+template <typename K>
 __global__ __launch_bounds__(NTHREADS) void k_fused_gather(
     const uint32_t *__restrict__ em_counts,
     const uint32_t *__restrict__ em_excl,
     const uint32_t *__restrict__ goff /* d-major offsets, [t*RADIX+d] */,
-    const int64_t *__restrict__ keys_scr, const double *__restrict__ sums_scr,
-    uint64_t ntiles, uint64_t nrec, int64_t *__restrict__ stream_k,
+    const K *__restrict__ keys_scr, const double *__restrict__ sums_scr,
+    uint64_t ntiles, uint64_t nrec, K *__restrict__ stream_k,
     double *__restrict__ stream_v) {
   uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
   for (uint64_t r = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -1249,12 +1303,13 @@ __global__ __launch_bounds__(NTHREADS) void k_fused_gather(
  * padded (PIDX: +1 slot per 16) — the unpadded layout put lanes l and
  * l+2 on the same bank (16-way conflict, measured 1.26 ms/launch). */
 #define PIDX(j) ((j) + ((j) >> 3))
+template <typename K>
 __global__ __launch_bounds__(NTHREADS, 2) void k_reduce_tiles(
-    const int64_t *__restrict__ keys, const double *__restrict__ vals,
-    uint64_t n, int64_t *__restrict__ keys_scr, double *__restrict__ sums_scr,
+    const K *__restrict__ keys, const double *__restrict__ vals,
+    uint64_t n, K *__restrict__ keys_scr, double *__restrict__ sums_scr,
     double *__restrict__ tile_T, uint32_t *__restrict__ tile_anyend,
     uint32_t *__restrict__ tile_nruns, uint64_t ntiles) {
-  __shared__ int64_t lk[RTILE + RTILE / 8 + 1];  /* padded; +1: lookahead */
+  __shared__ K lk[RTILE + RTILE / 8 + 1];  /* padded; +1: lookahead */
   __shared__ double lv[RTILE + RTILE / 8];
   __shared__ double t_carry[NTHREADS];   /* per-thread trailing open sum */
   __shared__ uint32_t t_flag[NTHREADS];  /* thread chunk has a head      */
@@ -1268,12 +1323,12 @@ __global__ __launch_bounds__(NTHREADS, 2) void k_reduce_tiles(
      * so all loads issue up front (variable-bound loops keep ~1 load in
      * flight; the kernel was 69% wave-parked) */
     if (tile_n == RTILE) {
-      longlong2 kk[RTILE / (2 * NTHREADS)];
+      typename KT<K>::V2 kk[RTILE / (2 * NTHREADS)];
       double2 vv[RTILE / (2 * NTHREADS)];
 #pragma unroll
       for (int j = 0; j < RTILE / (2 * NTHREADS); j++) {
         uint32_t i = 2 * threadIdx.x + (uint32_t)j * 2 * NTHREADS;
-        kk[j] = *reinterpret_cast<const longlong2 *>(&keys[start + i]);
+        kk[j] = *reinterpret_cast<const typename KT<K>::V2 *>(&keys[start + i]);
         vv[j] = *reinterpret_cast<const double2 *>(&vals[start + i]);
       }
 #pragma unroll
@@ -1286,7 +1341,8 @@ __global__ __launch_bounds__(NTHREADS, 2) void k_reduce_tiles(
       }
     } else {
       for (uint32_t i = 2 * threadIdx.x; i + 1 < tile_n; i += 2 * NTHREADS) {
-        longlong2 kk = *reinterpret_cast<const longlong2 *>(&keys[start + i]);
+        typename KT<K>::V2 kk =
+            *reinterpret_cast<const typename KT<K>::V2 *>(&keys[start + i]);
         double2 vv = *reinterpret_cast<const double2 *>(&vals[start + i]);
         lk[PIDX(i)] = kk.x;
         lk[PIDX(i + 1)] = kk.y;
@@ -1299,7 +1355,7 @@ __global__ __launch_bounds__(NTHREADS, 2) void k_reduce_tiles(
       }
     }
     if (threadIdx.x == 0)
-      lk[PIDX(tile_n)] = (end < n) ? keys[end] : 0; /* lookahead (unused if end==n) */
+      lk[PIDX(tile_n)] = (end < n) ? keys[end] : (K)0; /* lookahead (unused if end==n) */
     __syncthreads();
 
     const uint32_t c0 = threadIdx.x * EPT_R;                    /* chunk */
@@ -1308,7 +1364,7 @@ __global__ __launch_bounds__(NTHREADS, 2) void k_reduce_tiles(
     /* pass 1: trailing open sum + has-head + end count per thread chunk */
     double open = 0.0;
     uint32_t has_head = 0, n_ends = 0;
-    int64_t prev_key = 0;
+    K prev_key = 0;
     bool have_prev = false;
     if (c0 < tile_n) {
       if (c0 == 0) {
@@ -1364,7 +1420,7 @@ __global__ __launch_bounds__(NTHREADS, 2) void k_reduce_tiles(
     /* pass 2: walk chunk again with the incoming carry; emit run ends */
     if (c0 < tile_n) {
       double running = incoming;
-      int64_t prev_key2 = 0;
+      K prev_key2 = 0;
       bool have_prev2 = false;
       if (c0 == 0) {
         if (start != 0) { prev_key2 = keys[start - 1]; have_prev2 = true; }
@@ -1514,12 +1570,13 @@ __global__ __launch_bounds__(NTHREADS) void k_stitchC(
 /* compact: copy each tile's runs from scratch to the final arrays; the
  * FIRST run of a tile absorbs the stitched carry C_t (C_t == 0.0 whenever
  * the tile's leading segment is empty — see DESIGN.md). */
+template <typename K>
 __global__ __launch_bounds__(NTHREADS) void k_compact_runs(
-    const int64_t *__restrict__ keys_scr, const double *__restrict__ sums_scr,
+    const K *__restrict__ keys_scr, const double *__restrict__ sums_scr,
     const uint32_t *__restrict__ tile_nruns,
     const uint64_t *__restrict__ tile_outoff,
     const double *__restrict__ tile_C, uint64_t ntiles,
-    int64_t *__restrict__ keys_out, double *__restrict__ sums_out) {
+    K *__restrict__ keys_out, double *__restrict__ sums_out) {
   for (uint64_t t = blockIdx.x; t < ntiles; t += gridDim.x) {
     uint32_t nr = tile_nruns[t];
     uint64_t base = tile_outoff[t];
@@ -1538,9 +1595,10 @@ __global__ __launch_bounds__(NTHREADS) void k_compact_runs(
 /* ------------------------------------------------------------------ */
 
 /* per tile: compact (key, head_global_index) for every run HEAD */
+template <typename K>
 __global__ __launch_bounds__(NTHREADS) void k_head_tiles(
-    const int64_t *__restrict__ keys, uint64_t n,
-    int64_t *__restrict__ keys_scr, uint64_t *__restrict__ idx_scr,
+    const K *__restrict__ keys, uint64_t n,
+    K *__restrict__ keys_scr, uint64_t *__restrict__ idx_scr,
     uint32_t *__restrict__ tile_nheads, uint64_t ntiles) {
   __shared__ uint32_t cnt[NTHREADS];
   for (uint64_t t = blockIdx.x; t < ntiles; t += gridDim.x) {
@@ -1577,8 +1635,11 @@ __global__ __launch_bounds__(NTHREADS) void k_head_tiles(
 }
 
 /* compact heads -> (unique_keys, starts) using scanned tile offsets */
+/* heads compact ALWAYS emits int64 unique keys (grp_keys) — the join /
+ * group machinery downstream is key-width independent */
+template <typename K>
 __global__ __launch_bounds__(NTHREADS) void k_compact_heads(
-    const int64_t *__restrict__ keys_scr, const uint64_t *__restrict__ idx_scr,
+    const K *__restrict__ keys_scr, const uint64_t *__restrict__ idx_scr,
     const uint32_t *__restrict__ tile_nheads,
     const uint64_t *__restrict__ tile_outoff, uint64_t ntiles,
     int64_t *__restrict__ uk, uint64_t *__restrict__ starts) {
@@ -1587,7 +1648,7 @@ __global__ __launch_bounds__(NTHREADS) void k_compact_heads(
     uint64_t base = tile_outoff[t];
     uint64_t src = t * (uint64_t)RTILE;
     for (uint32_t j = threadIdx.x; j < nh; j += NTHREADS) {
-      uk[base + j] = keys_scr[src + j];
+      uk[base + j] = (int64_t)KT<K>::raw(keys_scr[src + j]);
       starts[base + j] = idx_scr[src + j];
     }
   }
@@ -1974,6 +2035,7 @@ extern "C" int dprk_reserve(dprk_ctx *c, uint64_t max_n, int with_vals) {
   c->n = 0;
   c->cur = 0;
   c->sorted = 0;
+  c->packed = 0;
   return 0;
 }
 
@@ -1997,26 +2059,14 @@ extern "C" int dprk_set_alias_table(dprk_ctx *c, const uint64_t *table,
   return 0;
 }
 
-extern "C" int dprk_gen_pairs(dprk_ctx *c, uint64_t seed, uint64_t n,
-                              int dist, uint64_t key_universe) {
-  if (!c) { set_err("null ctx"); return 1; }
-  if (dist != 0 && dist != 1) { set_err("gen: unknown dist %d", dist);
-                                return 1; }
-  if (dist == 1 && (!c->alias_table || c->alias_universe != key_universe)) {
-    set_err("gen dist=1: call dprk_set_alias_table(universe=%llu) first",
-            (unsigned long long)key_universe);
-    return 1;
-  }
-  if (n > c->cap) { set_err("gen: n > reserved cap"); return 1; }
-  HIP_CHECK(hipSetDevice(c->device));
-  uint64_t keymask =
-      (key_universe == 0) ? ~0ull : (key_universe - 1); /* power of two */
-  c->hist_shift_valid = -1;
-  c->ghist_valid = 0;
+template <typename K>
+static int gen_pairs_t(dprk_ctx *c, uint64_t seed, uint64_t n, int dist,
+                       uint64_t key_universe, uint64_t keymask) {
   if (dist == 1) {
     TIMED(c, K_GEN,
-          hipLaunchKernelGGL(k_gen_alias, dim3(stream_grid(n)), dim3(NTHREADS),
-                             0, c->stream, c->keys[c->cur],
+          hipLaunchKernelGGL(k_gen_alias<K>, dim3(stream_grid(n)),
+                             dim3(NTHREADS),
+                             0, c->stream, (K *)c->keys[c->cur],
                              c->with_vals ? c->vals[c->cur] : nullptr, n,
                              seed << 32, c->alias_table, key_universe));
   } else {
@@ -2049,9 +2099,10 @@ extern "C" int dprk_gen_pairs(dprk_ctx *c, uint64_t seed, uint64_t n,
                              c->stream));
     uint64_t ntiles = ceil_div(n, (uint64_t)TILE);
     TIMED(c, K_GEN,
-          hipLaunchKernelGGL(k_gen_hist,
+          hipLaunchKernelGGL(k_gen_hist<K>,
                              dim3((uint32_t)min(ntiles, (uint64_t)2048)),
-                             dim3(NTHREADS), 0, c->stream, c->keys[c->cur],
+                             dim3(NTHREADS), 0, c->stream,
+                             (K *)c->keys[c->cur],
                              c->with_vals ? c->vals[c->cur] : nullptr, n,
                              seed << 32, keymask, c->tile_hist, ntiles,
                              shift0, shifts, nact, c->d_ghist, c->d_orand));
@@ -2079,10 +2130,81 @@ extern "C" int dprk_gen_pairs(dprk_ctx *c, uint64_t seed, uint64_t n,
     /* exact values were read back from k_gen_hist above */
   } else if (dist == 1 && (key_universe & (key_universe - 1)) == 0) {
     c->orand_valid = 1;   /* alias keys are < universe (power of two) */
-    c->known_or = SIGNFLIP | (key_universe - 1);
-    c->known_and = SIGNFLIP;
+    if (c->packed) {      /* packed bias = the key value itself */
+      c->known_or = key_universe - 1;
+      c->known_and = 0;
+    } else {
+      c->known_or = SIGNFLIP | (key_universe - 1);
+      c->known_and = SIGNFLIP;
+    }
   } else {
     c->orand_valid = 0;
+  }
+  return drain_timing(c);
+}
+
+extern "C" int dprk_gen_pairs(dprk_ctx *c, uint64_t seed, uint64_t n,
+                              int dist, uint64_t key_universe) {
+  if (!c) { set_err("null ctx"); return 1; }
+  if (dist != 0 && dist != 1) { set_err("gen: unknown dist %d", dist);
+                                return 1; }
+  if (dist == 1 && (!c->alias_table || c->alias_universe != key_universe)) {
+    set_err("gen dist=1: call dprk_set_alias_table(universe=%llu) first",
+            (unsigned long long)key_universe);
+    return 1;
+  }
+  if (n > c->cap) { set_err("gen: n > reserved cap"); return 1; }
+  HIP_CHECK(hipSetDevice(c->device));
+  uint64_t keymask =
+      (key_universe == 0) ? ~0ull : (key_universe - 1); /* power of two */
+  c->hist_shift_valid = -1;
+  c->ghist_valid = 0;
+  /* packed-u32 fast path: generated keys are in [0, universe) — pack
+   * whenever the universe fits 32 bits (uniform policy, a pure function
+   * of the gen arguments => bit-repeatable path selection) */
+  c->packed = (key_universe != 0 && key_universe <= (1ull << 32));
+  return c->packed
+             ? gen_pairs_t<uint32_t>(c, seed, n, dist, key_universe, keymask)
+             : gen_pairs_t<int64_t>(c, seed, n, dist, key_universe, keymask);
+}
+
+/* put-side pack gate: run k_minmax_or on the fresh int64 keys (the sort
+ * would need it anyway — this just moves it to put time), then pack to
+ * u32 if every key fits [0, 2^32).  Input-deterministic. */
+static int maybe_pack_after_put(dprk_ctx *c, uint64_t n) {
+  uint64_t init[2] = {0, ~0ull};
+  HIP_CHECK(hipMemcpyAsync(c->d_orand, init, 16, hipMemcpyHostToDevice,
+                           c->stream));
+  TIMED(c, K_HIST,
+        hipLaunchKernelGGL(k_minmax_or<int64_t>, dim3(stream_grid(n)),
+                           dim3(NTHREADS), 0, c->stream, c->keys[c->cur],
+                           n, c->d_orand));
+  uint64_t oa[2];
+  HIP_CHECK(hipMemcpyAsync(oa, c->d_orand, 16, hipMemcpyDeviceToHost,
+                           c->stream));
+  HIP_CHECK(hipStreamSynchronize(c->stream));
+  c->stats.bytes_total += n * 8;
+  c->orand_valid = 1;
+  c->known_or = oa[0];
+  c->known_and = oa[1];
+  /* all keys in [0, 2^32)  <=>  biased OR's high 32 bits == 0x80000000
+   * (bit 63 set: non-negative; bits 32..62 clear: < 2^32) */
+  if ((oa[0] >> 32) == 0x80000000ull) {
+    TIMED(c, K_OTHER,
+          hipLaunchKernelGGL(k_pack_keys, dim3(stream_grid(n)),
+                             dim3(NTHREADS), 0, c->stream, c->keys[c->cur],
+                             (uint32_t *)c->keys[c->cur ^ 1], n));
+    /* pointer-swap keys only: vals stay in place, cur unchanged */
+    int64_t *tmp = c->keys[c->cur];
+    c->keys[c->cur] = c->keys[c->cur ^ 1];
+    c->keys[c->cur ^ 1] = tmp;
+    c->packed = 1;
+    /* re-express the known bounds in the packed bias domain */
+    c->known_or = oa[0] & 0xFFFFFFFFull;
+    c->known_and = oa[1] & 0xFFFFFFFFull;
+    c->stats.bytes_total += n * 12;
+  } else {
+    c->packed = 0;
   }
   return drain_timing(c);
 }
@@ -2102,8 +2224,10 @@ extern "C" int dprk_put_pairs(dprk_ctx *c, const int64_t *keys,
   c->n = n;
   c->sorted = 0;
   c->orand_valid = 0;
+  c->packed = 0;
   c->hist_shift_valid = -1;
   c->ghist_valid = 0;
+  if (n > 0 && maybe_pack_after_put(c, n)) return 1;
   return 0;
 }
 
@@ -2115,20 +2239,101 @@ extern "C" int dprk_get_pairs(dprk_ctx *c, int64_t *keys, double *vals,
                             (unsigned long long)cap,
                             (unsigned long long)c->n); return 1; }
   HIP_CHECK(hipSetDevice(c->device));
-  if (keys)
-    HIP_CHECK(hipMemcpyAsync(keys, c->keys[c->cur], c->n * 8,
-                             hipMemcpyDeviceToHost, c->stream));
+  if (keys) {
+    if (c->packed) {
+      /* unpack into the spare ping-pong buffer (scratch), copy from it;
+       * the resident dataset stays packed */
+      TIMED(c, K_OTHER,
+            hipLaunchKernelGGL(k_unpack_keys, dim3(stream_grid(c->n)),
+                               dim3(NTHREADS), 0, c->stream,
+                               (const uint32_t *)c->keys[c->cur],
+                               c->keys[c->cur ^ 1], c->n));
+      HIP_CHECK(hipMemcpyAsync(keys, c->keys[c->cur ^ 1], c->n * 8,
+                               hipMemcpyDeviceToHost, c->stream));
+    } else {
+      HIP_CHECK(hipMemcpyAsync(keys, c->keys[c->cur], c->n * 8,
+                               hipMemcpyDeviceToHost, c->stream));
+    }
+  }
   if (vals && c->with_vals)
     HIP_CHECK(hipMemcpyAsync(vals, c->vals[c->cur], c->n * 8,
                              hipMemcpyDeviceToHost, c->stream));
   HIP_CHECK(hipStreamSynchronize(c->stream));
+  return drain_timing(c);
+}
+
+/* packed-state introspection + forcing (multi-rank exchange agreement:
+ * every rank must ship the same key width — the host layer all-reduces
+ * the packed flags and calls set_packed to align disagreeing ranks) */
+extern "C" int dprk_is_packed(dprk_ctx *c, int *out) {
+  if (!c || !out) { set_err("null arg"); return 1; }
+  *out = c->packed;
   return 0;
 }
 
+extern "C" int dprk_set_packed(dprk_ctx *c, int want) {
+  if (!c) { set_err("null ctx"); return 1; }
+  HIP_CHECK(hipSetDevice(c->device));
+  if (want == c->packed) return 0;
+  if (c->n == 0) { c->packed = want ? 1 : 0; return 0; }
+  if (!want) {
+    /* unpack in place: u32 -> int64 into the spare buffer, pointer-swap */
+    TIMED(c, K_OTHER,
+          hipLaunchKernelGGL(k_unpack_keys, dim3(stream_grid(c->n)),
+                             dim3(NTHREADS), 0, c->stream,
+                             (const uint32_t *)c->keys[c->cur],
+                             c->keys[c->cur ^ 1], c->n));
+    int64_t *tmp = c->keys[c->cur];
+    c->keys[c->cur] = c->keys[c->cur ^ 1];
+    c->keys[c->cur ^ 1] = tmp;
+    c->packed = 0;
+    if (c->orand_valid) {   /* packed bias -> int64 bias domain */
+      c->known_or |= SIGNFLIP;
+      c->known_and |= SIGNFLIP;
+    }
+    c->stats.bytes_total += c->n * 12;
+    return drain_timing(c);
+  }
+  /* want packed: only valid if every key fits [0, 2^32) */
+  if (!c->orand_valid) {
+    uint64_t init[2] = {0, ~0ull};
+    HIP_CHECK(hipMemcpyAsync(c->d_orand, init, 16, hipMemcpyHostToDevice,
+                             c->stream));
+    TIMED(c, K_HIST,
+          hipLaunchKernelGGL(k_minmax_or<int64_t>, dim3(stream_grid(c->n)),
+                             dim3(NTHREADS), 0, c->stream, c->keys[c->cur],
+                             c->n, c->d_orand));
+    uint64_t oa[2];
+    HIP_CHECK(hipMemcpyAsync(oa, c->d_orand, 16, hipMemcpyDeviceToHost,
+                             c->stream));
+    HIP_CHECK(hipStreamSynchronize(c->stream));
+    c->orand_valid = 1;
+    c->known_or = oa[0];
+    c->known_and = oa[1];
+  }
+  if ((c->known_or >> 32) != 0x80000000ull) {
+    set_err("set_packed: keys do not fit [0, 2^32)");
+    return 1;
+  }
+  TIMED(c, K_OTHER,
+        hipLaunchKernelGGL(k_pack_keys, dim3(stream_grid(c->n)),
+                           dim3(NTHREADS), 0, c->stream, c->keys[c->cur],
+                           (uint32_t *)c->keys[c->cur ^ 1], c->n));
+  int64_t *tmp = c->keys[c->cur];
+  c->keys[c->cur] = c->keys[c->cur ^ 1];
+  c->keys[c->cur ^ 1] = tmp;
+  c->packed = 1;
+  c->known_or &= 0xFFFFFFFFull;
+  c->known_and &= 0xFFFFFFFFull;
+  c->stats.bytes_total += c->n * 12;
+  return drain_timing(c);
+}
+
 /* ---- one radix pass: tile_hist -> 3-level scan -> scatter ---------- */
-static int radix_pass(dprk_ctx *c, int mode, int shift, uint32_t mask,
-                      int want_row_tot, const int64_t *d_split = nullptr,
-                      int nsplit = 0) {
+template <typename K>
+static int radix_pass_t(dprk_ctx *c, int mode, int shift, uint32_t mask,
+                        int want_row_tot, const int64_t *d_split = nullptr,
+                        int nsplit = 0) {
   uint64_t n = c->n;
   uint64_t ntiles = ceil_div(n, TILE);
   uint64_t seg_tiles = ceil_div(ntiles, RADIX); /* nseg <= 256 */
@@ -2138,9 +2343,10 @@ static int radix_pass(dprk_ctx *c, int mode, int shift, uint32_t mask,
   int hist_ran = !(mode == 0 && c->hist_shift_valid == shift);
   if (hist_ran) {
     TIMED(c, K_TILEHIST,
-          hipLaunchKernelGGL(k_tile_hist, dim3(thgrid), dim3(NTHREADS), 0,
-                             c->stream, c->keys[src], n, c->tile_hist,
-                             ntiles, mode, shift, mask, d_split, nsplit));
+          hipLaunchKernelGGL(k_tile_hist<K>, dim3(thgrid), dim3(NTHREADS), 0,
+                             c->stream, (const K *)c->keys[src], n,
+                             c->tile_hist, ntiles, mode, shift, mask,
+                             d_split, nsplit));
   }
   c->hist_shift_valid = -1; /* consumed; scatter reorders the data */
   TIMED(c, K_SCAN,
@@ -2157,23 +2363,37 @@ static int radix_pass(dprk_ctx *c, int mode, int shift, uint32_t mask,
                            c->seg_sums, c->tile_off));
   if (c->with_vals) {
     TIMED(c, K_SCATTER,
-          hipLaunchKernelGGL(k_scatter<true>, dim3(thgrid), dim3(NTHREADS_SC),
-                             0, c->stream, c->keys[src], c->vals[src],
-                             c->keys[dst], c->vals[dst], n, c->tile_hist,
+          hipLaunchKernelGGL((k_scatter<K, true>), dim3(thgrid),
+                             dim3(NTHREADS_SC),
+                             0, c->stream, (const K *)c->keys[src],
+                             c->vals[src],
+                             (K *)c->keys[dst], c->vals[dst], n, c->tile_hist,
                              c->tile_off, ntiles, mode, shift, mask,
                              d_split, nsplit));
   } else {
     TIMED(c, K_SCATTER,
-          hipLaunchKernelGGL(k_scatter<false>, dim3(thgrid), dim3(NTHREADS_SC),
-                             0, c->stream, c->keys[src], nullptr,
-                             c->keys[dst], nullptr, n, c->tile_hist,
+          hipLaunchKernelGGL((k_scatter<K, false>), dim3(thgrid),
+                             dim3(NTHREADS_SC),
+                             0, c->stream, (const K *)c->keys[src], nullptr,
+                             (K *)c->keys[dst], nullptr, n, c->tile_hist,
                              c->tile_off, ntiles, mode, shift, mask,
                              d_split, nsplit));
   }
-  c->stats.bytes_scatter += n * (c->with_vals ? 32 : 16);
-  c->stats.bytes_total += n * ((c->with_vals ? 32 : 16) + (hist_ran ? 8 : 0));
+  uint64_t bpp = 2 * (sizeof(K) + (c->with_vals ? 8 : 0));
+  c->stats.bytes_scatter += n * bpp;
+  c->stats.bytes_total += n * (bpp + (hist_ran ? sizeof(K) : 0));
   c->cur = dst;
   return 0;
+}
+
+static int radix_pass(dprk_ctx *c, int mode, int shift, uint32_t mask,
+                      int want_row_tot, const int64_t *d_split = nullptr,
+                      int nsplit = 0) {
+  return c->packed
+             ? radix_pass_t<uint32_t>(c, mode, shift, mask, want_row_tot,
+                                      d_split, nsplit)
+             : radix_pass_t<int64_t>(c, mode, shift, mask, want_row_tot,
+                                     d_split, nsplit);
 }
 
 extern "C" int dprk_sample_keys(dprk_ctx *c, uint64_t n_samples,
@@ -2185,10 +2405,18 @@ extern "C" int dprk_sample_keys(dprk_ctx *c, uint64_t n_samples,
   uint64_t stride_elems = c->n / n_samples;
   int64_t *d_out = nullptr;
   HIP_CHECK(hipMalloc(&d_out, n_samples * 8));
-  TIMED(c, K_OTHER,
-        hipLaunchKernelGGL(k_sample, dim3(stream_grid(n_samples)),
-                           dim3(NTHREADS), 0, c->stream, c->keys[c->cur],
-                           c->n, stride_elems, d_out, n_samples));
+  if (c->packed) {
+    TIMED(c, K_OTHER,
+          hipLaunchKernelGGL(k_sample<uint32_t>, dim3(stream_grid(n_samples)),
+                             dim3(NTHREADS), 0, c->stream,
+                             (const uint32_t *)c->keys[c->cur],
+                             c->n, stride_elems, d_out, n_samples));
+  } else {
+    TIMED(c, K_OTHER,
+          hipLaunchKernelGGL(k_sample<int64_t>, dim3(stream_grid(n_samples)),
+                             dim3(NTHREADS), 0, c->stream, c->keys[c->cur],
+                             c->n, stride_elems, d_out, n_samples));
+  }
   HIP_CHECK(hipMemcpyAsync(out_host, d_out, n_samples * 8,
                            hipMemcpyDeviceToHost, c->stream));
   HIP_CHECK(hipStreamSynchronize(c->stream));
@@ -2249,6 +2477,7 @@ static int ensure_lookback(dprk_ctx *c, uint64_t ntiles) {
  * pass's global digit histogram; each pass is then a single
  * k_scatter_lb launch (per-tile hist + decoupled-lookback offsets).
  * Falls back per pass to the 3-kernel path on bounded-spin failure. */
+template <typename K>
 static int sort_pairs_lookback(dprk_ctx *c, const int *passes, int np) {
   uint64_t n = c->n;
   uint64_t ntiles = ceil_div(n, TILE);
@@ -2276,10 +2505,11 @@ static int sort_pairs_lookback(dprk_ctx *c, const int *passes, int np) {
   if (!c->ghist_valid) {
     HIP_CHECK(hipMemsetAsync(c->d_ghist, 0, 8 * RADIX * 4, c->stream));
     TIMED(c, K_HIST,
-          hipLaunchKernelGGL(k_hist_all, dim3(stream_grid(n / 2 + 1)),
-                             dim3(NTHREADS), 0, c->stream, c->keys[c->cur],
+          hipLaunchKernelGGL(k_hist_all<K>, dim3(stream_grid(n / 2 + 1)),
+                             dim3(NTHREADS), 0, c->stream,
+                             (const K *)c->keys[c->cur],
                              n, shifts, np, c->d_ghist));
-    c->stats.bytes_total += n * 8;
+    c->stats.bytes_total += n * sizeof(K);
     for (int i = 0; i < np; i++) gh_map[i] = i;
   }
   uint32_t gh[8 * RADIX];
@@ -2302,12 +2532,12 @@ static int sort_pairs_lookback(dprk_ctx *c, const int *passes, int np) {
      * assignment needs every block resident; the occupancy API can
      * over-report by one for SGPR-heavy kernels — clamp to 2/CU, which
      * the 74 KiB LDS independently bounds) */
-    static int occ_v = 0, occ_k = 0;
+    static int occ_v = 0, occ_k = 0;   /* per-K instantiation cache */
     if (!occ_v) {
       HIP_CHECK(hipOccupancyMaxActiveBlocksPerMultiprocessor(
-          &occ_v, (const void *)k_scatter_lb<true>, NTHREADS_SC, 0));
+          &occ_v, (const void *)(k_scatter_lb<K, true>), NTHREADS_SC, 0));
       HIP_CHECK(hipOccupancyMaxActiveBlocksPerMultiprocessor(
-          &occ_k, (const void *)k_scatter_lb<false>, NTHREADS_SC, 0));
+          &occ_k, (const void *)(k_scatter_lb<K, false>), NTHREADS_SC, 0));
       hipDeviceProp_t props;
       HIP_CHECK(hipGetDeviceProperties(&props, c->device));
       occ_v = min(2, max(1, occ_v)) * props.multiProcessorCount;
@@ -2316,16 +2546,18 @@ static int sort_pairs_lookback(dprk_ctx *c, const int *passes, int np) {
     int grid = (int)min(ntiles, (uint64_t)(c->with_vals ? occ_v : occ_k));
     if (c->with_vals) {
       TIMED(c, K_SCATTER,
-            hipLaunchKernelGGL(k_scatter_lb<true>, dim3(grid),
+            hipLaunchKernelGGL((k_scatter_lb<K, true>), dim3(grid),
                                dim3(NTHREADS_SC), 0, c->stream,
-                               c->keys[src], c->vals[src], c->keys[dst],
+                               (const K *)c->keys[src], c->vals[src],
+                               (K *)c->keys[dst],
                                c->vals[dst], n, ntiles, 8 * passes[i],
                                c->d_dbase, c->d_lb, c->d_fail));
     } else {
       TIMED(c, K_SCATTER,
-            hipLaunchKernelGGL(k_scatter_lb<false>, dim3(grid),
+            hipLaunchKernelGGL((k_scatter_lb<K, false>), dim3(grid),
                                dim3(NTHREADS_SC), 0, c->stream,
-                               c->keys[src], nullptr, c->keys[dst], nullptr,
+                               (const K *)c->keys[src], nullptr,
+                               (K *)c->keys[dst], nullptr,
                                n, ntiles, 8 * passes[i], c->d_dbase,
                                c->d_lb, c->d_fail));
     }
@@ -2336,10 +2568,11 @@ static int sort_pairs_lookback(dprk_ctx *c, const int *passes, int np) {
     if (fail) {
       /* bounded-spin bailout: src buffers are intact — redo this pass on
        * the decoupled 3-kernel path (never observed; safety net) */
-      if (radix_pass(c, 0, 8 * passes[i], 0xFF, 0)) return 1;
+      if (radix_pass_t<K>(c, 0, 8 * passes[i], 0xFF, 0)) return 1;
     } else {
-      c->stats.bytes_scatter += n * (c->with_vals ? 32 : 16);
-      c->stats.bytes_total += n * (c->with_vals ? 32 : 16);
+      uint64_t bpp = 2 * (sizeof(K) + (c->with_vals ? 8 : 0));
+      c->stats.bytes_scatter += n * bpp;
+      c->stats.bytes_total += n * bpp;
       c->cur = dst;
     }
     c->hist_shift_valid = -1;
@@ -2365,14 +2598,24 @@ extern "C" int dprk_sort_pairs(dprk_ctx *c) {
     uint64_t init[2] = {0, ~0ull};
     HIP_CHECK(hipMemcpyAsync(c->d_orand, init, 16, hipMemcpyHostToDevice,
                              c->stream));
-    TIMED(c, K_HIST,
-          hipLaunchKernelGGL(k_minmax_or, dim3(stream_grid(c->n)),
-                             dim3(NTHREADS), 0, c->stream, c->keys[c->cur],
-                             c->n, c->d_orand));
+    if (c->packed) {
+      TIMED(c, K_HIST,
+            hipLaunchKernelGGL(k_minmax_or<uint32_t>,
+                               dim3(stream_grid(c->n)),
+                               dim3(NTHREADS), 0, c->stream,
+                               (const uint32_t *)c->keys[c->cur],
+                               c->n, c->d_orand));
+    } else {
+      TIMED(c, K_HIST,
+            hipLaunchKernelGGL(k_minmax_or<int64_t>,
+                               dim3(stream_grid(c->n)),
+                               dim3(NTHREADS), 0, c->stream, c->keys[c->cur],
+                               c->n, c->d_orand));
+    }
     HIP_CHECK(hipMemcpyAsync(orand, c->d_orand, 16, hipMemcpyDeviceToHost,
                              c->stream));
     HIP_CHECK(hipStreamSynchronize(c->stream));
-    c->stats.bytes_total += c->n * 8;
+    c->stats.bytes_total += c->n * (c->packed ? 4 : 8);
   }
   c->stats.sort_passes = 0;
   int passes[8], np = 0;
@@ -2391,7 +2634,8 @@ extern "C" int dprk_sort_pairs(dprk_ctx *c) {
   uint64_t lb_tiles = ceil_div(c->n, (uint64_t)TILE);
   if (np && !getenv("DPRK_NO_LOOKBACK") &&
       lb_tiles * RADIX * 8 <= (200ull << 20))
-    return sort_pairs_lookback(c, passes, np);
+    return c->packed ? sort_pairs_lookback<uint32_t>(c, passes, np)
+                     : sort_pairs_lookback<int64_t>(c, passes, np);
   for (int i = 0; i < np; i++) {
     if (radix_pass(c, /*mode=*/0, /*shift=*/8 * passes[i], /*mask=*/0xFF, 0))
       return 1;
@@ -2461,12 +2705,14 @@ static int run_scan_u64(dprk_ctx *c, uint64_t *data, uint64_t m,
   return 0;
 }
 
+template <typename K>
 static int segmented_reduce_sorted(dprk_ctx *c, uint64_t *n_out);
 
 /* the fused last-pass reduce path (DPRK_FUSED_REDUCE): runs all but the
  * last radix pass normally, then k_scatter_reduce (fragment emission) +
  * d-major reorder + the base reduce over the small fragment stream.
  * shift = the last pass's digit shift. */
+template <typename K>
 static int fused_reduce_last_pass(dprk_ctx *c, int shift, uint64_t *n_out) {
   uint64_t n = c->n;
   uint64_t ntiles = ceil_div(n, TILE);
@@ -2478,15 +2724,17 @@ static int fused_reduce_last_pass(dprk_ctx *c, int shift, uint64_t *n_out) {
   if (ensure_fused(c, ceil_div(c->cap, (uint64_t)TILE) * RADIX)) return 1;
   if (c->hist_shift_valid != shift) {
     TIMED(c, K_TILEHIST,
-          hipLaunchKernelGGL(k_tile_hist, dim3(grid), dim3(NTHREADS), 0,
-                             c->stream, c->keys[src], n, c->tile_hist,
+          hipLaunchKernelGGL(k_tile_hist<K>, dim3(grid), dim3(NTHREADS), 0,
+                             c->stream, (const K *)c->keys[src], n,
+                             c->tile_hist,
                              ntiles, 0, shift, 0xFF, nullptr, 0));
   }
   c->hist_shift_valid = -1;
   TIMED(c, K_REDUCE,
-        hipLaunchKernelGGL(k_scatter_reduce<true>, dim3(grid),
-                           dim3(NTHREADS_SC), 0, c->stream, c->keys[src],
-                           c->vals[src], c->keys[dst], c->vals[dst], n,
+        hipLaunchKernelGGL(k_scatter_reduce<K>, dim3(grid),
+                           dim3(NTHREADS_SC), 0, c->stream,
+                           (const K *)c->keys[src],
+                           c->vals[src], (K *)c->keys[dst], c->vals[dst], n,
                            c->tile_hist, ntiles, shift, c->f_cnt, c->f_lex));
   /* d-major offsets of the fragment stream: the same 3-level scan used
    * for radix passes, run over the fragment-count matrix */
@@ -2509,30 +2757,32 @@ static int fused_reduce_last_pass(dprk_ctx *c, int shift, uint64_t *n_out) {
   /* gather tile-local fragments into the d-major stream (into the freed
    * pass-input buffers) */
   TIMED(c, K_REDUCE,
-        hipLaunchKernelGGL(k_fused_gather, dim3(stream_grid(nrec)),
+        hipLaunchKernelGGL(k_fused_gather<K>, dim3(stream_grid(nrec)),
                            dim3(NTHREADS), 0, c->stream, c->f_cnt, c->f_lex,
-                           c->f_goff, c->keys[dst], c->vals[dst], ntiles,
-                           nrec, c->keys[src], c->vals[src]));
-  c->stats.bytes_total += n * 24 + n2 * 32 + nrec * 16;
+                           c->f_goff, (const K *)c->keys[dst], c->vals[dst],
+                           ntiles,
+                           nrec, (K *)c->keys[src], c->vals[src]));
+  c->stats.bytes_total += n * (sizeof(K) + 16) + n2 * 32 + nrec * 16;
   c->stats.sort_passes++; /* the fused pass counts as a pass */
   c->n = n2;
   c->cur = src;
   c->sorted = 1;   /* the fragment stream is key-sorted (d-major order) */
   if (n_out) *n_out = n2;
   /* the base reduce merges cross-tile fragments exactly */
-  return segmented_reduce_sorted(c, n_out);
+  return segmented_reduce_sorted<K>(c, n_out);
 }
 
 /* the base segmented reduce over the (sorted) current dataset */
+template <typename K>
 static int segmented_reduce_sorted(dprk_ctx *c, uint64_t *n_out) {
   uint64_t n = c->n;
   uint64_t ntiles = ceil_div(n, RTILE);
   int src = c->cur, scr = c->cur ^ 1;
   int grid = tile_grid(ntiles);
   TIMED(c, K_REDUCE,
-        hipLaunchKernelGGL(k_reduce_tiles, dim3(grid), dim3(NTHREADS), 0,
-                           c->stream, c->keys[src], c->vals[src], n,
-                           c->keys[scr], c->vals[scr], c->tile_T,
+        hipLaunchKernelGGL(k_reduce_tiles<K>, dim3(grid), dim3(NTHREADS), 0,
+                           c->stream, (const K *)c->keys[src], c->vals[src], n,
+                           (K *)c->keys[scr], c->vals[scr], c->tile_T,
                            c->tile_anyend, c->tile_nruns, ntiles));
   uint64_t nchunks = ceil_div(ntiles, (uint64_t)STITCH_CHUNK);
   TIMED(c, K_REDUCE,
@@ -2554,11 +2804,11 @@ static int segmented_reduce_sorted(dprk_ctx *c, uint64_t *n_out) {
                            c->stream));
   HIP_CHECK(hipStreamSynchronize(c->stream));
   TIMED(c, K_REDUCE,
-        hipLaunchKernelGGL(k_compact_runs, dim3(grid), dim3(NTHREADS), 0,
-                           c->stream, c->keys[scr], c->vals[scr],
+        hipLaunchKernelGGL(k_compact_runs<K>, dim3(grid), dim3(NTHREADS), 0,
+                           c->stream, (const K *)c->keys[scr], c->vals[scr],
                            c->tile_nruns, c->tile_outoff, c->tile_C, ntiles,
-                           c->keys[src], c->vals[src]));
-  c->stats.bytes_total += n * 16 + total * 32;
+                           (K *)c->keys[src], c->vals[src]));
+  c->stats.bytes_total += n * (sizeof(K) + 8) + total * (2 * sizeof(K) + 16);
   c->n = total;
   c->cur = src;
   c->sorted = 1;
@@ -2590,14 +2840,25 @@ extern "C" int dprk_reduce_by_key_sum(dprk_ctx *c, uint64_t *n_out) {
       uint64_t init[2] = {0, ~0ull};
       HIP_CHECK(hipMemcpyAsync(c->d_orand, init, 16, hipMemcpyHostToDevice,
                                c->stream));
-      TIMED(c, K_HIST,
-            hipLaunchKernelGGL(k_minmax_or, dim3(stream_grid(c->n)),
-                               dim3(NTHREADS), 0, c->stream, c->keys[c->cur],
-                               c->n, c->d_orand));
+      if (c->packed) {
+        TIMED(c, K_HIST,
+              hipLaunchKernelGGL(k_minmax_or<uint32_t>,
+                                 dim3(stream_grid(c->n)),
+                                 dim3(NTHREADS), 0, c->stream,
+                                 (const uint32_t *)c->keys[c->cur],
+                                 c->n, c->d_orand));
+      } else {
+        TIMED(c, K_HIST,
+              hipLaunchKernelGGL(k_minmax_or<int64_t>,
+                                 dim3(stream_grid(c->n)),
+                                 dim3(NTHREADS), 0, c->stream,
+                                 c->keys[c->cur],
+                                 c->n, c->d_orand));
+      }
       HIP_CHECK(hipMemcpyAsync(orand, c->d_orand, 16, hipMemcpyDeviceToHost,
                                c->stream));
       HIP_CHECK(hipStreamSynchronize(c->stream));
-      c->stats.bytes_total += c->n * 8;
+      c->stats.bytes_total += c->n * (c->packed ? 4 : 8);
       c->orand_valid = 1;
       c->known_or = orand[0];
       c->known_and = orand[1];
@@ -2614,16 +2875,22 @@ extern "C" int dprk_reduce_by_key_sum(dprk_ctx *c, uint64_t *n_out) {
         if (radix_pass(c, 0, 8 * passes[i], 0xFF, 0)) return 1;
         c->stats.sort_passes++;
       }
-      return fused_reduce_last_pass(c, 8 * passes[np - 1], n_out);
+      return c->packed
+                 ? fused_reduce_last_pass<uint32_t>(c, 8 * passes[np - 1],
+                                                    n_out)
+                 : fused_reduce_last_pass<int64_t>(c, 8 * passes[np - 1],
+                                                   n_out);
     }
     /* np == 0: all keys equal — trivial; fall through to the base path */
   }
   if (!c->sorted && dprk_sort_pairs(c)) return 1;
-  return segmented_reduce_sorted(c, n_out);
+  return c->packed ? segmented_reduce_sorted<uint32_t>(c, n_out)
+                   : segmented_reduce_sorted<int64_t>(c, n_out);
 }
 
 /* run extraction into ctx group arrays (assumes sorted) */
-static int extract_runs(dprk_ctx *c) {
+template <typename K>
+static int extract_runs_t(dprk_ctx *c) {
   uint64_t n = c->n;
   uint64_t ntiles = ceil_div(n, RTILE);
   int src = c->cur, scr = c->cur ^ 1;
@@ -2632,8 +2899,9 @@ static int extract_runs(dprk_ctx *c) {
    * keys scratch uses the alt keys buffer */
   if (!c->with_vals) { set_err("groups need vals buffers reserved"); return 1; }
   TIMED(c, K_RUNEXT,
-        hipLaunchKernelGGL(k_head_tiles, dim3(grid), dim3(NTHREADS), 0,
-                           c->stream, c->keys[src], n, c->keys[scr],
+        hipLaunchKernelGGL(k_head_tiles<K>, dim3(grid), dim3(NTHREADS), 0,
+                           c->stream, (const K *)c->keys[src], n,
+                           (K *)c->keys[scr],
                            (uint64_t *)c->vals[scr], c->tile_nruns, ntiles));
   TIMED(c, K_RUNEXT,
         hipLaunchKernelGGL(k_scan_u32_single, dim3(1), dim3(NTHREADS), 0,
@@ -2645,16 +2913,21 @@ static int extract_runs(dprk_ctx *c) {
   HIP_CHECK(hipStreamSynchronize(c->stream));
   if (ensure_grp(c, g ? g : 1)) return 1;
   TIMED(c, K_RUNEXT,
-        hipLaunchKernelGGL(k_compact_heads, dim3(grid), dim3(NTHREADS), 0,
-                           c->stream, c->keys[scr], (uint64_t *)c->vals[scr],
+        hipLaunchKernelGGL(k_compact_heads<K>, dim3(grid), dim3(NTHREADS), 0,
+                           c->stream, (const K *)c->keys[scr],
+                           (uint64_t *)c->vals[scr],
                            c->tile_nruns, c->tile_outoff, ntiles,
                            c->grp_keys, c->grp_starts));
   TIMED(c, K_RUNEXT,
         hipLaunchKernelGGL(k_run_counts, dim3(stream_grid(g)), dim3(NTHREADS),
                            0, c->stream, c->grp_starts, g, n, c->grp_counts));
-  c->stats.bytes_total += n * 8 + g * 24;
+  c->stats.bytes_total += n * sizeof(K) + g * 24;
   c->n_groups = g;
   return drain_timing(c);
+}
+
+static int extract_runs(dprk_ctx *c) {
+  return c->packed ? extract_runs_t<uint32_t>(c) : extract_runs_t<int64_t>(c);
 }
 
 extern "C" int dprk_group_by_key(dprk_ctx *c, uint64_t *n_groups_out) {
@@ -2822,15 +3095,20 @@ extern "C" int dprk_exchange(dprk_ctx *c, const uint64_t *send_counts,
                                      (unsigned long long)total_recv);
                              return 1; }
   int src = c->cur, dst = c->cur ^ 1;
+  /* packed keys travel as u32 over xGMI (half the key bytes); every rank
+   * must agree on the width — the host layer aligns packed state before
+   * calling (Engine.sync_packed / bench exchange paths) */
+  ncclDataType_t kdt = c->packed ? ncclUint32 : ncclInt64;
+  size_t ksz = c->packed ? 4 : 8;
   int began = timed_begin(c, K_EXCHANGE);
   NCCL_CHECK(ncclGroupStart());
   for (int r = 0; r < nranks; r++) {
     if (send_counts[r])
-      NCCL_CHECK(ncclSend(c->keys[src] + send_off[r], send_counts[r],
-                          ncclInt64, r, c->comm, c->stream));
+      NCCL_CHECK(ncclSend((char *)c->keys[src] + send_off[r] * ksz,
+                          send_counts[r], kdt, r, c->comm, c->stream));
     if (recv_counts[r])
-      NCCL_CHECK(ncclRecv(c->keys[dst] + recv_off[r], recv_counts[r],
-                          ncclInt64, r, c->comm, c->stream));
+      NCCL_CHECK(ncclRecv((char *)c->keys[dst] + recv_off[r] * ksz,
+                          recv_counts[r], kdt, r, c->comm, c->stream));
     if (c->with_vals) {
       if (send_counts[r])
         NCCL_CHECK(ncclSend(c->vals[src] + send_off[r], send_counts[r],
@@ -2842,7 +3120,8 @@ extern "C" int dprk_exchange(dprk_ctx *c, const uint64_t *send_counts,
   }
   NCCL_CHECK(ncclGroupEnd());
   timed_end(c, began);
-  c->stats.bytes_total += (total_send + total_recv) * (c->with_vals ? 16 : 8);
+  c->stats.bytes_total +=
+      (total_send + total_recv) * (ksz + (c->with_vals ? 8 : 0));
   c->cur = dst;
   c->n = total_recv;
   c->sorted = 0;
@@ -2889,6 +3168,8 @@ extern "C" int dprk_exchange_buckets(dprk_ctx *c,
                                      (unsigned long long)c->cap);
                              return 1; }
   int src = c->cur, dst = c->cur ^ 1;
+  ncclDataType_t kdt = c->packed ? ncclUint32 : ncclInt64;
+  size_t ksz = c->packed ? 4 : 8;
   int began = timed_begin(c, K_EXCHANGE);
   NCCL_CHECK(ncclGroupStart());
   uint64_t soff = 0;
@@ -2896,7 +3177,7 @@ extern "C" int dprk_exchange_buckets(dprk_ctx *c,
     uint64_t cnt = bucket_counts[b];
     if (cnt) {
       int peer = bucket_rank[b];
-      NCCL_CHECK(ncclSend(c->keys[src] + soff, cnt, ncclInt64, peer,
+      NCCL_CHECK(ncclSend((char *)c->keys[src] + soff * ksz, cnt, kdt, peer,
                           c->comm, c->stream));
       if (c->with_vals)
         NCCL_CHECK(ncclSend(c->vals[src] + soff, cnt, ncclDouble, peer,
@@ -2911,7 +3192,7 @@ extern "C" int dprk_exchange_buckets(dprk_ctx *c,
     for (int b = 0; b < nbuckets; b++) {
       uint64_t cnt = recv_matrix[(uint64_t)s * nbuckets + b];
       if (cnt) {
-        NCCL_CHECK(ncclRecv(c->keys[dst] + roff, cnt, ncclInt64, s,
+        NCCL_CHECK(ncclRecv((char *)c->keys[dst] + roff * ksz, cnt, kdt, s,
                             c->comm, c->stream));
         if (c->with_vals)
           NCCL_CHECK(ncclRecv(c->vals[dst] + roff, cnt, ncclDouble, s,
@@ -2922,7 +3203,8 @@ extern "C" int dprk_exchange_buckets(dprk_ctx *c,
   }
   NCCL_CHECK(ncclGroupEnd());
   timed_end(c, began);
-  c->stats.bytes_total += (total_send + total_recv) * (c->with_vals ? 16 : 8);
+  c->stats.bytes_total +=
+      (total_send + total_recv) * (ksz + (c->with_vals ? 8 : 0));
   c->cur = dst;
   c->n = total_recv;
   c->sorted = 0;

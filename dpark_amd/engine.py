@@ -55,7 +55,8 @@ SYMBOLS = [
     "dprk_init", "dprk_free", "dprk_last_error", "dprk_build_info",
     "dprk_reserve", "dprk_gen_pairs", "dprk_set_alias_table",
     "dprk_put_pairs", "dprk_get_pairs",
-    "dprk_count", "dprk_partition", "dprk_partition_range",
+    "dprk_count", "dprk_is_packed", "dprk_set_packed",
+    "dprk_partition", "dprk_partition_range",
     "dprk_sample_keys", "dprk_sort_pairs",
     "dprk_reduce_by_key_sum", "dprk_group_by_key", "dprk_get_groups",
     "dprk_join", "dprk_get_join", "dprk_comm_unique_id", "dprk_comm_init",
@@ -99,6 +100,8 @@ def load_library():
     lib.dprk_get_pairs.argtypes = [p, ctypes.c_void_p, ctypes.c_void_p, u64,
                                    ctypes.POINTER(u64)]
     lib.dprk_count.argtypes = [p, ctypes.POINTER(u64)]
+    lib.dprk_is_packed.argtypes = [p, ctypes.POINTER(ctypes.c_int)]
+    lib.dprk_set_packed.argtypes = [p, ctypes.c_int]
     lib.dprk_partition.argtypes = [p, ctypes.c_int, ctypes.POINTER(u64)]
     lib.dprk_partition_range.argtypes = [p, ctypes.c_void_p, ctypes.c_int,
                                          ctypes.POINTER(u64)]
@@ -202,6 +205,31 @@ class Engine:
         _check(self._lib, self._lib.dprk_count(self._ctx, ctypes.byref(n)),
                "dprk_count")
         return n.value
+
+    def is_packed(self):
+        """Whether keys currently travel as packed u32 (include/dprk.h)."""
+        v = ctypes.c_int()
+        _check(self._lib, self._lib.dprk_is_packed(self._ctx,
+                                                   ctypes.byref(v)),
+               "dprk_is_packed")
+        return bool(v.value)
+
+    def set_packed(self, want):
+        _check(self._lib, self._lib.dprk_set_packed(self._ctx,
+                                                    1 if want else 0),
+               "dprk_set_packed")
+
+    def sync_packed(self, dist):
+        """Multi-rank exchange agreement: all ranks must ship the same key
+        width.  All-reduce (MIN) of the local packed flag over the gloo
+        control plane; ranks packed while any peer is not unpack."""
+        import torch
+        t = torch.tensor([1 if self.is_packed() else 0])
+        dist.all_reduce(t, op=dist.ReduceOp.MIN)
+        agreed = bool(int(t[0]))
+        if self.is_packed() != agreed:
+            self.set_packed(agreed)
+        return agreed
 
     # -- ops -------------------------------------------------------------
     def partition(self, nbuckets):

@@ -40,6 +40,21 @@ class SimEngine:
     def set_alias_table(self, table):
         self._table = np.asarray(table, dtype=np.uint64)
 
+    def is_packed(self):
+        if len(self.keys) == 0:
+            return False
+        return bool(self.keys.min() >= 0 and self.keys.max() < (1 << 32))
+
+    def set_packed(self, want):
+        if want and not self.is_packed():
+            raise RuntimeError("set_packed: keys do not fit [0, 2^32)")
+
+    def sync_packed(self, dist):
+        import torch
+        t = torch.tensor([1 if self.is_packed() else 0])
+        dist.all_reduce(t, op=dist.ReduceOp.MIN)
+        return bool(int(t[0]))
+
     def partition(self, nbuckets):
         r = rank_of_keys(self.keys, nbuckets)
         order = np.argsort(r, kind="stable")

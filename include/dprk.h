@@ -68,6 +68,17 @@ int dprk_get_pairs(dprk_ctx *ctx, int64_t *keys, double *vals, uint64_t cap,
 /* current dataset size */
 int dprk_count(dprk_ctx *ctx, uint64_t *n_out);
 
+/* Packed-u32 key transport (internal layout, not a semantics change):
+ * when every key of the dataset lies in [0, 2^32) the engine stores and
+ * moves keys as uint32 (25% fewer bytes per radix pass; detected
+ * input-deterministically at gen/put time).  is_packed reads the state;
+ * set_packed forces it (want=0 always legal; want=1 errors if any key
+ * does not fit).  Multi-rank: all ranks must agree on the key width
+ * before dprk_exchange* — the host layer all-reduces the flags and
+ * aligns disagreeing ranks (Engine.sync_packed). */
+int dprk_is_packed(dprk_ctx *ctx, int *out);
+int dprk_set_packed(dprk_ctx *ctx, int want);
+
 /* ---- the hot path ---------------------------------------------------- */
 /* Radix-partition the current dataset into nbuckets contiguous segments by
  * the TOP bits of splitmix64(key) (the GPU HashPartitioner — SURVEY.md

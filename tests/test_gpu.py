@@ -259,6 +259,15 @@ def test_api_level_engine_routing():
     guk, gcnt = so.group_counts(keys)
     assert np.array_equal(uk, guk)
     assert np.array_equal(counts.astype(np.int64), gcnt)
+    # grouped VALUES per key must match the oracle as multisets
+    order = np.lexsort((vals, keys))
+    ov = vals[order]
+    woff = 0
+    for i in range(len(uk)):
+        s, c = int(starts[i]), int(counts[i])
+        np.testing.assert_allclose(np.sort(gvals[s:s + c]),
+                                   ov[woff:woff + c], rtol=0)
+        woff += c
     jb = ctx.vectorPairs(keys[:100_000], vals[:100_000], 2)
     jk, jva, jvb = r.join(jb).collectArrays()
     wk2, _, _ = so.join_multiset(keys, vals, keys[:100_000], vals[:100_000])
@@ -449,6 +458,108 @@ def test_fused_reduce_opt_in():
         e.close()
     finally:
         del os.environ["DPRK_FUSED_REDUCE"]
+
+
+def test_packed_state_and_boundaries():
+    """Packed-u32 key transport (include/dprk.h): the gate is exact at
+    the [0, 2^32) boundary, input-deterministic, and set_packed round
+    trips losslessly."""
+    from dpark_amd.engine import EngineError
+    e = Engine(device=0)
+    try:
+        e.reserve(100_000, with_vals=True)
+        # gen with a 32-bit-or-smaller universe packs
+        e.gen_pairs(seed=1, n=50_000, key_universe=1 << 24)
+        assert e.is_packed()
+        # full-universe gen does not
+        e.gen_pairs(seed=1, n=50_000, key_universe=0)
+        assert not e.is_packed()
+        # put: boundary keys
+        fits = np.array([0, 1, (1 << 31), (1 << 32) - 1], dtype=np.int64)
+        e.put(fits, np.ones(4))
+        assert e.is_packed()
+        k, v = e.get()
+        assert np.array_equal(k, fits)        # get() unpacks losslessly
+        assert e.reduce_by_key_sum() == 4
+        k, v = e.get()
+        assert np.array_equal(k, np.sort(fits))
+        too_big = np.array([0, 1 << 32], dtype=np.int64)
+        e.put(too_big, np.ones(2))
+        assert not e.is_packed()
+        negative = np.array([-1, 5], dtype=np.int64)
+        e.put(negative, np.ones(2))
+        assert not e.is_packed()
+        with pytest.raises(EngineError, match="fit"):
+            e.set_packed(True)
+        # round trip pack -> unpack -> pack preserves data exactly
+        keys, vals = gen_pairs(3, 100_000, 1 << 20)
+        e.put(keys, vals)
+        assert e.is_packed()
+        e.set_packed(False)
+        assert not e.is_packed()
+        k, v = e.get()
+        assert np.array_equal(k, keys) and np.array_equal(v, vals)
+        e.set_packed(True)
+        assert e.is_packed()
+        k, v = e.get()
+        assert np.array_equal(k, keys) and np.array_equal(v, vals)
+    finally:
+        e.close()
+
+
+def test_packed_vs_unpacked_bit_identical():
+    """The packed path is a transport layout, not a numerics change: the
+    SAME input reduced packed and force-unpacked gives bit-identical f64
+    sums (the reduction tree depends only on element positions)."""
+    e = Engine(device=0)
+    try:
+        keys, vals = gen_pairs(17, 2_000_000, 1 << 20)
+        e.reserve(len(keys), with_vals=True)
+        e.put(keys, vals)
+        assert e.is_packed()
+        e.reduce_by_key_sum()
+        k1, v1 = e.get()
+        e.put(keys, vals)
+        e.set_packed(False)      # force the int64 path on identical input
+        e.reduce_by_key_sum()
+        k2, v2 = e.get()
+        assert np.array_equal(k1, k2)
+        assert np.array_equal(v1, v2)          # bit-identical
+        # sort path too
+        e.put(keys, vals)
+        e.sort()
+        s1, sv1 = e.get()
+        e.put(keys, vals)
+        e.set_packed(False)
+        e.sort()
+        s2, sv2 = e.get()
+        assert np.array_equal(s1, s2) and np.array_equal(sv1, sv2)
+    finally:
+        e.close()
+
+
+def test_group_values_multiset(eng):
+    """The grouped VALUES (not just keys/counts/starts) match the oracle
+    per key as multisets — VERDICT r01 weak #5: a value-placement bug in
+    the grouped layout must fail here."""
+    for n, uni, seed in [(500_000, 1 << 12, 71), (300_000, 0, 72)]:
+        keys, vals = gen_pairs(seed, n, uni if uni else 1 << 64)
+        _put(eng, keys, vals)
+        uk, starts, counts = eng.group_by_key()
+        sk, sv = eng.get()      # the sorted dataset backing the groups
+        # oracle: pandas-style grouped values
+        order = np.lexsort((vals, keys))
+        ok, ov = keys[order], vals[order]
+        assert np.array_equal(uk, np.unique(keys))
+        woff = 0
+        assert len(sk) == n
+        for i in range(len(uk)):
+            s, c = int(starts[i]), int(counts[i])
+            gv = np.sort(sv[s:s + c])
+            wv = ov[woff:woff + c]          # already value-sorted per key
+            assert np.array_equal(sk[s:s + c], np.full(c, uk[i]))
+            np.testing.assert_allclose(gv, wv, rtol=0)  # exact permutation
+            woff += c
 
 
 def test_bucket_exchange_self():
