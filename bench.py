@@ -68,6 +68,91 @@ def run_config1():
                       "ok": True}))
 
 
+def run_wordcount(args):
+    """String-key engine leg: config-1 wordcount
+    through the device-hash path.  One step = H2D + k_hash_bytes + sort +
+    segmented reduce + exemplar grouping (the H2D and hash are part of
+    the step — stated in config; the CPU comparator is --config1)."""
+    from collections import Counter
+    from dpark_amd.engine import Engine
+    from dpark_amd.hashspec import pack_words
+
+    rng = np.random.default_rng(1)
+    vocab = [f"w{i:05d}" for i in range(10_000)]
+    n_lines = args.n_pairs or 1_000_000
+    log(f"wordcount: generating {n_lines} lines ...")
+    blocks = [rng.choice(vocab, size=(min(10_000, n_lines), 10))
+              for _ in range(max(1, n_lines // 10_000))]
+    words = [w for b in blocks for w in b.ravel().tolist()]
+    n = len(words)
+    buf, offs = pack_words(words)
+    vals = np.ones(n, dtype=np.float64)
+    rowids = np.arange(n, dtype=np.float64)
+    log(f"wordcount: {n} words, {len(buf)} bytes packed")
+
+    ea = Engine(device=0)
+    eb = Engine(device=0)
+    ea.reserve(n, with_vals=True)
+    eb.reserve(n, with_vals=True)
+
+    def step():
+        ea.put_bytes_keys(buf, offs, vals)
+        ea.reduce_by_key_sum()
+        uh, sums = ea.get()
+        eb.put_bytes_keys(buf, offs, rowids)
+        uk, starts, counts = eb.group_by_key()
+        _, sorted_rows = eb.get()
+        exemplar = sorted_rows[starts.astype(np.int64)].astype(np.int64)
+        return uh, sums, exemplar
+
+    for _ in range(args.warmup):
+        uh, sums, exemplar = step()
+    # parity vs the CPU dict path (dpark semantics), untimed
+    want = Counter(words)
+    got = {words[i]: s for i, s in zip(exemplar.tolist(), sums.tolist())}
+    assert len(got) == len(want) == 10_000, (len(got), len(want))
+    assert all(got[w] == c for w, c in want.items()), "wordcount mismatch"
+    log("wordcount: parity vs CPU Counter green")
+
+    ea.stats_reset()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    elapsed = time.perf_counter() - t0
+    s = ea.stats()
+    value = n * args.steps / elapsed / 1e6
+    roofline = None
+    if s["n_scatter"] > 0 and s["ms_scatter"] > 0:
+        achieved = s["bytes_scatter"] / s["ms_scatter"] * 1e-6
+        roofline = {"bound": "hbm", "achieved": round(achieved, 1),
+                    "peak": HBM_PEAK_SPEC, "unit": "GB/s",
+                    "frac": round(achieved / HBM_PEAK_SPEC, 4),
+                    "frac_vs_measured_copy":
+                        round(achieved / HBM_PEAK_MEASURED, 4),
+                    "traffic": None, "kernel": "k_scatter",
+                    "ms_per_launch": round(s["ms_scatter"] / s["n_scatter"],
+                                           4),
+                    "launches": int(s["n_scatter"])}
+    print(json.dumps({
+        "metric": "M words/sec through string-key reduceByKey (wordcount)",
+        "value": round(value, 2), "unit": "M words/s", "n_gpus": 1,
+        "steps": args.steps, "warmup": args.warmup,
+        "ms_per_step": round(elapsed * 1000 / args.steps, 3),
+        "higher_is_better": True, "scaling": "weak", "vs_baseline": None,
+        "dtype": "bytes-key+f64", "data": "synthetic",
+        "config": {"workload": "wordcount-engine-string-path "
+                               "(BASELINE.json configs[0] semantics on GPU; "
+                               "step INCLUDES H2D + device hash + exemplar "
+                               "grouping)",
+                   "op": "wordcount", "n_words": n,
+                   "distinct_words": 10_000,
+                   "parallelism": "single-gpu",
+                   "engine": "libdprk.so (HIP gfx950)"},
+        "roofline": roofline, "cpu_baseline": None}))
+    ea.close()
+    eb.close()
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
@@ -75,17 +160,24 @@ def main():
     ap.add_argument("--warmup", type=int, default=3)
     ap.add_argument("--n-pairs", type=int, default=None,
                     help="pairs per GPU (default: workload-specific)")
-    ap.add_argument("--op", choices=["reduce", "sort", "join", "zipf"],
+    ap.add_argument("--op", choices=["reduce", "sort", "join", "zipf",
+                                     "wordcount"],
                     default="reduce",
                     help="reduce=config 2 (headline), sort=config 3 "
                          "(1B keys-only LSD sort), join=config 4 shape, "
-                         "zipf=config 5 (1B Zipf(1.1) skewed reduceByKey)")
+                         "zipf=config 5 (1B Zipf(1.1) skewed reduceByKey); "
+                         "wordcount=config 1 semantics on the GPU "
+                         "string-key path (N=1 only)")
     ap.add_argument("--config1", action="store_true",
                     help="run the CPU plumbing check instead")
     ap.add_argument("--skip-cpu-baseline", action="store_true")
     args = ap.parse_args()
     if args.config1:
         run_config1()
+        return
+    if args.op == "wordcount":
+        assert args.gpus == 1, "wordcount leg is single-GPU"
+        run_wordcount(args)
         return
 
     if args.gpus > 1 and "WORLD_SIZE" not in os.environ:

@@ -67,6 +67,13 @@ class DparkContext:
         from .vector import VectorPairsRDD
         return VectorPairsRDD(self, keys, vals, numSlices)
 
+    def stringPairs(self, words, vals, numSlices=None):
+        """Engine-eligible pair RDD with bytes/str keys: reduceByKey(+)
+        routes through the device-hash string path (dpark_amd/strings.py,
+        SURVEY.md §8(f) rank 1)."""
+        from .strings import StringPairsRDD
+        return StringPairsRDD(self, words, vals, numSlices)
+
     def union(self, rdds):
         return UnionRDD(self, list(rdds))
 

@@ -367,6 +367,33 @@ __global__ __launch_bounds__(NTHREADS) void k_sample(
     out[i] = (int64_t)KT<K>::raw(keys[min(i * stride_elems, n - 1)]);
 }
 
+/* bytes-key hashing (the string-key engine path, SURVEY.md §8(f) rank 1:
+ * TextFileRDD -> wordcount [U]).  Hash spec (normative; numpy mirror in
+ * dpark_amd/hashspec.py :: hash_bytes_words, asserted identical by
+ * tests): the word is split into 8-byte little-endian chunks (last one
+ * zero-padded); h = splitmix64(C1 ^ len); for each chunk c:
+ * h = splitmix64(h ^ c).  Length in the seed keeps "a" and "a\0"
+ * distinct.  One thread per word; words are consecutive in the buffer so
+ * the byte loads stay L2-coalesced across lanes. */
+__global__ __launch_bounds__(NTHREADS) void k_hash_bytes(
+    const uint8_t *__restrict__ buf, const uint64_t *__restrict__ offs,
+    uint64_t n_words, int64_t *__restrict__ keys) {
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < n_words; i += stride) {
+    uint64_t b0 = offs[i], b1 = offs[i + 1];
+    uint64_t h = d_splitmix64(0x9E3779B97F4A7C15ull ^ (b1 - b0));
+    for (uint64_t p = b0; p < b1; p += 8) {
+      uint64_t c = 0;
+      uint64_t lim = min(b1 - p, (uint64_t)8);
+      for (uint64_t j = 0; j < lim; j++)
+        c |= (uint64_t)buf[p + j] << (8 * j);
+      h = d_splitmix64(h ^ c);
+    }
+    keys[i] = (int64_t)h;
+  }
+}
+
 /* pack (int64 -> u32) / unpack (u32 -> int64); caller guarantees fit */
 __global__ __launch_bounds__(NTHREADS) void k_pack_keys(
     const int64_t *__restrict__ in, uint32_t *__restrict__ out, uint64_t n) {
@@ -2259,6 +2286,50 @@ extern "C" int dprk_get_pairs(dprk_ctx *c, int64_t *keys, double *vals,
     HIP_CHECK(hipMemcpyAsync(vals, c->vals[c->cur], c->n * 8,
                              hipMemcpyDeviceToHost, c->stream));
   HIP_CHECK(hipStreamSynchronize(c->stream));
+  return drain_timing(c);
+}
+
+/* string-key path: hash a packed bytes buffer
+ * of n_words words (offs: n_words+1 byte offsets) into the ctx's keys
+ * via k_hash_bytes; vals uploaded alongside (NULL => all 1.0 is NOT
+ * assumed — caller supplies vals or reserves keys-only).  The dataset
+ * then flows through the normal sort/reduce/group pipeline keyed by the
+ * 64-bit hash; the HOST resolves hash -> word via exemplar rows (see
+ * dpark_amd/strings.py).  Collisions: P ~ K^2/2^65 for K distinct words
+ * (~1e-9 at K=10M); DPRK_VERIFY_STRING_HASH=1 host-verifies exactly. */
+extern "C" int dprk_put_bytes_keys(dprk_ctx *c, const uint8_t *buf,
+                                   uint64_t nbytes, const uint64_t *offs,
+                                   uint64_t n_words, const double *vals) {
+  if (!c) { set_err("null ctx"); return 1; }
+  if (n_words > c->cap) { set_err("put_bytes: n > reserved cap"); return 1; }
+  if (vals && !c->with_vals) { set_err("put_bytes: ctx reserved keys-only");
+                               return 1; }
+  HIP_CHECK(hipSetDevice(c->device));
+  uint8_t *d_buf = nullptr;
+  uint64_t *d_offs = nullptr;
+  HIP_CHECK(hipMalloc(&d_buf, nbytes ? nbytes : 1));
+  HIP_CHECK(hipMalloc(&d_offs, (n_words + 1) * 8));
+  HIP_CHECK(hipMemcpyAsync(d_buf, buf, nbytes, hipMemcpyHostToDevice,
+                           c->stream));
+  HIP_CHECK(hipMemcpyAsync(d_offs, offs, (n_words + 1) * 8,
+                           hipMemcpyHostToDevice, c->stream));
+  TIMED(c, K_HIST,
+        hipLaunchKernelGGL(k_hash_bytes, dim3(stream_grid(n_words)),
+                           dim3(NTHREADS), 0, c->stream, d_buf, d_offs,
+                           n_words, c->keys[c->cur]));
+  if (vals && c->with_vals)
+    HIP_CHECK(hipMemcpyAsync(c->vals[c->cur], vals, n_words * 8,
+                             hipMemcpyHostToDevice, c->stream));
+  HIP_CHECK(hipStreamSynchronize(c->stream));
+  HIP_CHECK(hipFree(d_buf));
+  HIP_CHECK(hipFree(d_offs));
+  c->stats.bytes_total += nbytes + n_words * 16;
+  c->n = n_words;
+  c->sorted = 0;
+  c->packed = 0;      /* hashes span the full 64-bit space */
+  c->orand_valid = 0;
+  c->hist_shift_valid = -1;
+  c->ghist_valid = 0;
   return drain_timing(c);
 }
 

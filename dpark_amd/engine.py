@@ -54,7 +54,7 @@ class DprkStats(ctypes.Structure):
 SYMBOLS = [
     "dprk_init", "dprk_free", "dprk_last_error", "dprk_build_info",
     "dprk_reserve", "dprk_gen_pairs", "dprk_set_alias_table",
-    "dprk_put_pairs", "dprk_get_pairs",
+    "dprk_put_pairs", "dprk_get_pairs", "dprk_put_bytes_keys",
     "dprk_count", "dprk_is_packed", "dprk_set_packed",
     "dprk_partition", "dprk_partition_range",
     "dprk_sample_keys", "dprk_sort_pairs",
@@ -99,6 +99,8 @@ def load_library():
     lib.dprk_put_pairs.argtypes = [p, ctypes.c_void_p, ctypes.c_void_p, u64]
     lib.dprk_get_pairs.argtypes = [p, ctypes.c_void_p, ctypes.c_void_p, u64,
                                    ctypes.POINTER(u64)]
+    lib.dprk_put_bytes_keys.argtypes = [p, ctypes.c_void_p, u64,
+                                        ctypes.c_void_p, u64, ctypes.c_void_p]
     lib.dprk_count.argtypes = [p, ctypes.POINTER(u64)]
     lib.dprk_is_packed.argtypes = [p, ctypes.POINTER(ctypes.c_int)]
     lib.dprk_set_packed.argtypes = [p, ctypes.c_int]
@@ -186,6 +188,24 @@ class Engine:
                                         keys.ctypes.data_as(ctypes.c_void_p),
                                         vp, len(keys)),
                "dprk_put_pairs")
+
+    def put_bytes_keys(self, buf, offs, vals=None):
+        """String-key path: device-hash n words (packed bytes `buf`,
+        offsets `offs` of n+1 entries) into the keys; vals alongside.
+        Spec + numpy mirror: dpark_amd/hashspec.py hash_bytes_words."""
+        buf = np.ascontiguousarray(buf, dtype=np.uint8)
+        offs = np.ascontiguousarray(offs, dtype=np.uint64)
+        n = len(offs) - 1
+        vp = None
+        if vals is not None:
+            vals = np.ascontiguousarray(vals, dtype=np.float64)
+            assert len(vals) == n
+            vp = vals.ctypes.data_as(ctypes.c_void_p)
+        _check(self._lib,
+               self._lib.dprk_put_bytes_keys(
+                   self._ctx, buf.ctypes.data_as(ctypes.c_void_p), len(buf),
+                   offs.ctypes.data_as(ctypes.c_void_p), n, vp),
+               "dprk_put_bytes_keys")
 
     def get(self, with_vals=True):
         n = self.count()

@@ -65,6 +65,15 @@ int dprk_put_pairs(dprk_ctx *ctx, const int64_t *keys, const double *vals,
                    uint64_t n); /* vals may be NULL for keys-only */
 int dprk_get_pairs(dprk_ctx *ctx, int64_t *keys, double *vals, uint64_t cap,
                    uint64_t *n_out); /* D2H of the current dataset */
+/* String-key path (SURVEY.md §8(f) rank 1 — TextFileRDD -> wordcount
+ * [U]): device-hash n_words bytes keys (packed buffer + n_words+1 byte
+ * offsets; spec mirrored in dpark_amd/hashspec.py) into the dataset's
+ * int64 keys, vals uploaded alongside.  The hashes then flow through
+ * sort/reduce/group; hash->word resolution is host-side via exemplar
+ * rows (dpark_amd/strings.py documents the collision bound). */
+int dprk_put_bytes_keys(dprk_ctx *ctx, const uint8_t *buf, uint64_t nbytes,
+                        const uint64_t *offs, uint64_t n_words,
+                        const double *vals);
 /* current dataset size */
 int dprk_count(dprk_ctx *ctx, uint64_t *n_out);
 

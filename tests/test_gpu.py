@@ -562,6 +562,72 @@ def test_group_values_multiset(eng):
             woff += c
 
 
+def test_hash_bytes_device_matches_mirror(eng):
+    """k_hash_bytes output is byte-identical to the normative numpy
+    mirror (dpark_amd/hashspec.py) — the string-key path's parity pin."""
+    from dpark_amd.hashspec import hash_bytes_words, pack_words
+    rng = np.random.default_rng(9)
+    words = ([f"w{i:05d}" for i in range(10_000)]
+             + ["", "a", "ab" * 20, "\x00\x01\x02"]
+             + ["".join(chr(97 + c) for c in rng.integers(0, 26, size=l))
+                for l in rng.integers(1, 40, size=5000)])
+    buf, offs = pack_words(words)
+    n = len(words)
+    eng.reserve(n, with_vals=True)
+    eng.put_bytes_keys(buf, offs, np.zeros(n))
+    got = eng.get()[0].view(np.uint64)
+    want = hash_bytes_words(buf, offs)
+    assert np.array_equal(got, want)
+
+
+def test_string_reduce_end_to_end():
+    """engine_string_reduce == collections.Counter on wordcount-shaped
+    data, with the exact collision verification enabled."""
+    from collections import Counter
+    from dpark_amd.strings import engine_string_reduce
+    os.environ["DPRK_VERIFY_STRING_HASH"] = "1"
+    try:
+        rng = np.random.default_rng(12)
+        vocab = [f"word{i:04d}" for i in range(3000)]
+        words = rng.choice(vocab, size=400_000).tolist()
+        vals = np.ones(len(words))
+        out_words, sums = engine_string_reduce(words, vals)
+        want = Counter(words)
+        assert len(out_words) == len(want)
+        got = dict(zip(out_words, sums.tolist()))
+        assert all(got[w] == c for w, c in want.items())
+        # general (non-1.0) values: sums within tolerance of the oracle
+        vals2 = rng.random(len(words))
+        out2, sums2 = engine_string_reduce(words, vals2)
+        acc = {}
+        for w, v in zip(words, vals2.tolist()):
+            acc[w] = acc.get(w, 0.0) + v
+        got2 = dict(zip(out2, sums2.tolist()))
+        for w, v in acc.items():
+            np.testing.assert_allclose(got2[w], v, rtol=RTOL)
+    finally:
+        del os.environ["DPRK_VERIFY_STRING_HASH"]
+
+
+def test_string_api_routing():
+    """ctx.stringPairs(...).reduceByKey(+) routes through the engine on a
+    GPU box and matches the CPU dict semantics."""
+    import operator
+    from collections import Counter
+    from dpark_amd import DparkContext
+    from dpark_amd.strings import StringPairsRDD
+    rng = np.random.default_rng(13)
+    words = rng.choice([f"k{i}" for i in range(500)], size=100_000).tolist()
+    ctx = DparkContext("local[4]")
+    out = ctx.stringPairs(words, np.ones(len(words)), 4).reduceByKey(
+        operator.add)
+    assert isinstance(out, StringPairsRDD)
+    got = out.collectAsMap()
+    want = Counter(words)
+    assert len(got) == len(want)
+    assert all(got[w] == c for w, c in want.items())
+
+
 def test_bucket_exchange_self():
     """Bucket-granular RCCL exchange on the 1-GPU box: 256 buckets, all
     assigned to rank 0 via the real grouped per-bucket ncclSend/Recv
