@@ -776,7 +776,16 @@ __global__ __launch_bounds__(NTHREADS) void k_tile_off(
 #define LB_VAL  ((1ull << 62) - 1)
 #define LB_SPIN_LIMIT (1u << 16)
 
-template <typename K, bool WITH_VALS>
+/* RANK_ATOMIC (experiment, DPRK_RANK_ATOMIC=1): phase-3 stable ranking
+ * via ONE ds_add_rtn per element on the phase-2 cursors instead of the
+ * 8-ballot chain (~20 wave-ops/element).  Equal-digit order within a
+ * wave becomes the hardware's fixed atomic order rather than lane
+ * order: still a deterministic permutation, so run-to-run
+ * bit-repeatability, sorted-key order and per-key value MULTISETS (the
+ * sort contract — duplicate order unspecified) all
+ * hold; only the unspecified duplicate order differs from the ballot
+ * path. */
+template <typename K, bool WITH_VALS, bool RANK_ATOMIC = false>
 __global__ __launch_bounds__(NTHREADS_SC, 4) void k_scatter_lb(
     const K *__restrict__ keys_in, const double *__restrict__ vals_in,
     K *__restrict__ keys_out, double *__restrict__ vals_out,
@@ -872,16 +881,21 @@ __global__ __launch_bounds__(NTHREADS_SC, 4) void k_scatter_lb(
       uint32_t idx = wbeg + i * 64 + lane;
       if (idx < tile_n) {
         uint32_t dig = rd[i];
-        uint64_t same = __ballot(1);
-        for (int bit = 0; bit < RADIX_BITS; bit++) {
-          uint64_t bb = __ballot((dig >> bit) & 1);
-          same &= ((dig >> bit) & 1) ? bb : ~bb;
+        uint32_t pos;
+        if (RANK_ATOMIC) {
+          pos = atomicAdd(&h[wave][dig], 1u);
+        } else {
+          uint64_t same = __ballot(1);
+          for (int bit = 0; bit < RADIX_BITS; bit++) {
+            uint64_t bb = __ballot((dig >> bit) & 1);
+            same &= ((dig >> bit) & 1) ? bb : ~bb;
+          }
+          uint32_t rank_in_wave = (uint32_t)__popcll(same & lowmask);
+          uint32_t base = h[wave][dig];
+          if (rank_in_wave == 0)
+            h[wave][dig] = base + (uint32_t)__popcll(same);
+          pos = base + rank_in_wave;
         }
-        uint32_t rank_in_wave = (uint32_t)__popcll(same & lowmask);
-        uint32_t base = h[wave][dig];
-        if (rank_in_wave == 0)
-          h[wave][dig] = base + (uint32_t)__popcll(same);
-        uint32_t pos = base + rank_in_wave;
         lk[pos] = rk[i];
         if (WITH_VALS) lv[pos] = rv[i];
       }
@@ -916,7 +930,10 @@ __global__ __launch_bounds__(NTHREADS_SC, 4) void k_scatter_lb(
         __hip_atomic_store(&lb[t * RADIX + threadIdx.x],
                            LB_INC | (excl_g + tile_cnt), __ATOMIC_RELAXED,
                            __HIP_MEMORY_SCOPE_AGENT);
-        goff[threadIdx.x] = digit_base[threadIdx.x] + (uint32_t)excl_g;
+        /* combined write base: phase 4 computes dst = goff[dig] + idx
+         * with ONE LDS read (goff already folds -excl[dig]) */
+        goff[threadIdx.x] = digit_base[threadIdx.x] + (uint32_t)excl_g -
+                            excl[threadIdx.x];
       }
     }
     /* uniform post-lookback failure check (reuse wtot[0] as a flag) */
@@ -930,7 +947,7 @@ __global__ __launch_bounds__(NTHREADS_SC, 4) void k_scatter_lb(
     for (uint32_t idx = threadIdx.x; idx < tile_n; idx += NTHREADS_SC) {
       K key = lk[idx];
       uint32_t dig = d_digit(key, 0, shift, mask);
-      uint32_t dst = goff[dig] + (idx - excl[dig]);
+      uint32_t dst = goff[dig] + idx;   /* goff folds -excl[dig] */
       keys_out[dst] = key;
       if (WITH_VALS) vals_out[dst] = lv[idx];
     }
@@ -938,7 +955,7 @@ __global__ __launch_bounds__(NTHREADS_SC, 4) void k_scatter_lb(
   }
 }
 
-template <typename K, bool WITH_VALS>
+template <typename K, bool WITH_VALS, bool RANK_ATOMIC = false>
 __global__ __launch_bounds__(NTHREADS_SC, 2) void k_scatter(
     const K *__restrict__ keys_in, const double *__restrict__ vals_in,
     K *__restrict__ keys_out, double *__restrict__ vals_out,
@@ -989,10 +1006,12 @@ __global__ __launch_bounds__(NTHREADS_SC, 2) void k_scatter(
     for (int j = 0; j < RADIX / 64; j++) h[wave][lane * (RADIX / 64) + j] = 0;
 #endif
     __syncthreads();
-    if (threadIdx.x < RADIX && wave > 0) {
+    if (threadIdx.x < RADIX) {
       uint32_t add = 0;
       for (int w = 0; w < wave; w++) add += wtot[w];
-      excl[threadIdx.x] += add;
+      uint32_t e = excl[threadIdx.x] + add;
+      excl[threadIdx.x] = e;
+      goff[threadIdx.x] -= e;   /* combined: phase 4 dst = goff[dig]+idx */
     }
 
     /* phase 1: load into registers + wave-private histogram */
@@ -1046,6 +1065,13 @@ __global__ __launch_bounds__(NTHREADS_SC, 2) void k_scatter(
       uint32_t idx = wbeg + i * 64 + lane;
       if (idx < tile_n) {
         uint32_t dig = rd[i];
+        uint32_t pos;
+#ifndef DPRK_PACKED_H
+        if (RANK_ATOMIC) {
+          pos = atomicAdd(&h[wave][dig], 1u);
+        } else
+#endif
+        {
         uint64_t same = __ballot(1);
         for (int b = 0; b < RADIX_BITS; b++) {
           uint64_t bb = __ballot((dig >> b) & 1);
@@ -1063,7 +1089,8 @@ __global__ __launch_bounds__(NTHREADS_SC, 2) void k_scatter(
         if (rank_in_wave == 0)                    /* leader advances     */
           h[wave][dig] = base + (uint32_t)__popcll(same);
 #endif
-        uint32_t pos = base + rank_in_wave;
+        pos = base + rank_in_wave;
+        }
         lk[pos] = rk[i];
         if (WITH_VALS) lv[pos] = rv[i];
       }
@@ -1074,7 +1101,7 @@ __global__ __launch_bounds__(NTHREADS_SC, 2) void k_scatter(
     for (uint32_t idx = threadIdx.x; idx < tile_n; idx += NTHREADS_SC) {
       K key = lk[idx];
       uint32_t dig = d_digit_s(key, mode, shift, mask, splitters, nsplit);
-      uint32_t dst = goff[dig] + (idx - excl[dig]);
+      uint32_t dst = goff[dig] + idx;   /* goff folds -excl[dig] */
       keys_out[dst] = key;
       if (WITH_VALS) vals_out[dst] = lv[idx];
     }
@@ -2188,8 +2215,10 @@ extern "C" int dprk_gen_pairs(dprk_ctx *c, uint64_t seed, uint64_t n,
   c->ghist_valid = 0;
   /* packed-u32 fast path: generated keys are in [0, universe) — pack
    * whenever the universe fits 32 bits (uniform policy, a pure function
-   * of the gen arguments => bit-repeatable path selection) */
-  c->packed = (key_universe != 0 && key_universe <= (1ull << 32));
+   * of the gen arguments => bit-repeatable path selection).
+   * DPRK_NO_PACK=1 disables (A/B harness). */
+  c->packed = (key_universe != 0 && key_universe <= (1ull << 32) &&
+               !getenv("DPRK_NO_PACK"));
   return c->packed
              ? gen_pairs_t<uint32_t>(c, seed, n, dist, key_universe, keymask)
              : gen_pairs_t<int64_t>(c, seed, n, dist, key_universe, keymask);
@@ -2216,7 +2245,7 @@ static int maybe_pack_after_put(dprk_ctx *c, uint64_t n) {
   c->known_and = oa[1];
   /* all keys in [0, 2^32)  <=>  biased OR's high 32 bits == 0x80000000
    * (bit 63 set: non-negative; bits 32..62 clear: < 2^32) */
-  if ((oa[0] >> 32) == 0x80000000ull) {
+  if ((oa[0] >> 32) == 0x80000000ull && !getenv("DPRK_NO_PACK")) {
     TIMED(c, K_OTHER,
           hipLaunchKernelGGL(k_pack_keys, dim3(stream_grid(n)),
                              dim3(NTHREADS), 0, c->stream, c->keys[c->cur],
@@ -2432,23 +2461,48 @@ static int radix_pass_t(dprk_ctx *c, int mode, int shift, uint32_t mask,
         hipLaunchKernelGGL(k_tile_off, dim3((uint32_t)nseg), dim3(NTHREADS),
                            0, c->stream, c->tile_hist, ntiles, seg_tiles,
                            c->seg_sums, c->tile_off));
+  static int rank_atomic = -1;
+  if (rank_atomic < 0) rank_atomic = getenv("DPRK_RANK_ATOMIC") ? 1 : 0;
   if (c->with_vals) {
-    TIMED(c, K_SCATTER,
-          hipLaunchKernelGGL((k_scatter<K, true>), dim3(thgrid),
-                             dim3(NTHREADS_SC),
-                             0, c->stream, (const K *)c->keys[src],
-                             c->vals[src],
-                             (K *)c->keys[dst], c->vals[dst], n, c->tile_hist,
-                             c->tile_off, ntiles, mode, shift, mask,
-                             d_split, nsplit));
+    if (rank_atomic) {
+      TIMED(c, K_SCATTER,
+            hipLaunchKernelGGL((k_scatter<K, true, true>), dim3(thgrid),
+                               dim3(NTHREADS_SC),
+                               0, c->stream, (const K *)c->keys[src],
+                               c->vals[src],
+                               (K *)c->keys[dst], c->vals[dst], n,
+                               c->tile_hist,
+                               c->tile_off, ntiles, mode, shift, mask,
+                               d_split, nsplit));
+    } else {
+      TIMED(c, K_SCATTER,
+            hipLaunchKernelGGL((k_scatter<K, true>), dim3(thgrid),
+                               dim3(NTHREADS_SC),
+                               0, c->stream, (const K *)c->keys[src],
+                               c->vals[src],
+                               (K *)c->keys[dst], c->vals[dst], n,
+                               c->tile_hist,
+                               c->tile_off, ntiles, mode, shift, mask,
+                               d_split, nsplit));
+    }
   } else {
-    TIMED(c, K_SCATTER,
-          hipLaunchKernelGGL((k_scatter<K, false>), dim3(thgrid),
-                             dim3(NTHREADS_SC),
-                             0, c->stream, (const K *)c->keys[src], nullptr,
-                             (K *)c->keys[dst], nullptr, n, c->tile_hist,
-                             c->tile_off, ntiles, mode, shift, mask,
-                             d_split, nsplit));
+    if (rank_atomic) {
+      TIMED(c, K_SCATTER,
+            hipLaunchKernelGGL((k_scatter<K, false, true>), dim3(thgrid),
+                               dim3(NTHREADS_SC),
+                               0, c->stream, (const K *)c->keys[src], nullptr,
+                               (K *)c->keys[dst], nullptr, n, c->tile_hist,
+                               c->tile_off, ntiles, mode, shift, mask,
+                               d_split, nsplit));
+    } else {
+      TIMED(c, K_SCATTER,
+            hipLaunchKernelGGL((k_scatter<K, false>), dim3(thgrid),
+                               dim3(NTHREADS_SC),
+                               0, c->stream, (const K *)c->keys[src], nullptr,
+                               (K *)c->keys[dst], nullptr, n, c->tile_hist,
+                               c->tile_off, ntiles, mode, shift, mask,
+                               d_split, nsplit));
+    }
   }
   uint64_t bpp = 2 * (sizeof(K) + (c->with_vals ? 8 : 0));
   c->stats.bytes_scatter += n * bpp;
@@ -2615,22 +2669,44 @@ static int sort_pairs_lookback(dprk_ctx *c, const int *passes, int np) {
       occ_k = min(2, max(1, occ_k)) * props.multiProcessorCount;
     }
     int grid = (int)min(ntiles, (uint64_t)(c->with_vals ? occ_v : occ_k));
+    static int rank_atomic = -1;
+    if (rank_atomic < 0) rank_atomic = getenv("DPRK_RANK_ATOMIC") ? 1 : 0;
     if (c->with_vals) {
-      TIMED(c, K_SCATTER,
-            hipLaunchKernelGGL((k_scatter_lb<K, true>), dim3(grid),
-                               dim3(NTHREADS_SC), 0, c->stream,
-                               (const K *)c->keys[src], c->vals[src],
-                               (K *)c->keys[dst],
-                               c->vals[dst], n, ntiles, 8 * passes[i],
-                               c->d_dbase, c->d_lb, c->d_fail));
+      if (rank_atomic) {
+        TIMED(c, K_SCATTER,
+              hipLaunchKernelGGL((k_scatter_lb<K, true, true>), dim3(grid),
+                                 dim3(NTHREADS_SC), 0, c->stream,
+                                 (const K *)c->keys[src], c->vals[src],
+                                 (K *)c->keys[dst],
+                                 c->vals[dst], n, ntiles, 8 * passes[i],
+                                 c->d_dbase, c->d_lb, c->d_fail));
+      } else {
+        TIMED(c, K_SCATTER,
+              hipLaunchKernelGGL((k_scatter_lb<K, true>), dim3(grid),
+                                 dim3(NTHREADS_SC), 0, c->stream,
+                                 (const K *)c->keys[src], c->vals[src],
+                                 (K *)c->keys[dst],
+                                 c->vals[dst], n, ntiles, 8 * passes[i],
+                                 c->d_dbase, c->d_lb, c->d_fail));
+      }
     } else {
-      TIMED(c, K_SCATTER,
-            hipLaunchKernelGGL((k_scatter_lb<K, false>), dim3(grid),
-                               dim3(NTHREADS_SC), 0, c->stream,
-                               (const K *)c->keys[src], nullptr,
-                               (K *)c->keys[dst], nullptr,
-                               n, ntiles, 8 * passes[i], c->d_dbase,
-                               c->d_lb, c->d_fail));
+      if (rank_atomic) {
+        TIMED(c, K_SCATTER,
+              hipLaunchKernelGGL((k_scatter_lb<K, false, true>), dim3(grid),
+                                 dim3(NTHREADS_SC), 0, c->stream,
+                                 (const K *)c->keys[src], nullptr,
+                                 (K *)c->keys[dst], nullptr,
+                                 n, ntiles, 8 * passes[i], c->d_dbase,
+                                 c->d_lb, c->d_fail));
+      } else {
+        TIMED(c, K_SCATTER,
+              hipLaunchKernelGGL((k_scatter_lb<K, false>), dim3(grid),
+                                 dim3(NTHREADS_SC), 0, c->stream,
+                                 (const K *)c->keys[src], nullptr,
+                                 (K *)c->keys[dst], nullptr,
+                                 n, ntiles, 8 * passes[i], c->d_dbase,
+                                 c->d_lb, c->d_fail));
+      }
     }
     uint32_t fail = 0;
     HIP_CHECK(hipMemcpyAsync(&fail, c->d_fail, 4, hipMemcpyDeviceToHost,

@@ -16,7 +16,9 @@ import os
 import numpy as np
 
 _HERE = os.path.dirname(os.path.abspath(__file__))
-_SO = os.path.join(_HERE, "libdprk.so")
+# DPRK_SO overrides the library path (A/B harness for compile-knob
+# variants, e.g. TILE-size builds); default is the in-tree build.
+_SO = os.environ.get("DPRK_SO") or os.path.join(_HERE, "libdprk.so")
 
 _lib = None
 _load_error = None
