@@ -257,23 +257,49 @@ def main():
         # hash buckets, all-gather the count matrix, deterministically
         # greedy-assign buckets to ranks (hot buckets stay whole but get
         # a rank largely to themselves), then one bucket-granular RCCL
-        # all-to-allv.  reduceByKey correctness is untouched: every
-        # bucket (hence every key) lands wholly on one rank.
-        if nranks > 1:
-            from dpark_amd.distshuffle import greedy_assign, recv_matrix_for
-            e.sync_packed(dist)  # all ranks must ship the same key width
-            nbuckets = 256
-            send = e.partition(nbuckets)
-            counts = torch.tensor(send.astype("int64"))
-            mat = [torch.zeros(nbuckets, dtype=torch.int64)
-                   for _ in range(nranks)]
-            dist.all_gather(mat, counts)              # count matrix (tiny)
-            m = np.stack([r.numpy() for r in mat]).astype(np.uint64)
-            bucket_rank, loads = greedy_assign(m.sum(axis=0), nranks)
-            imbalance["max"] = max(imbalance.get("max", 0.0),
-                                   max(loads) / (sum(loads) / nranks))
-            rm = recv_matrix_for(m, bucket_rank, rank)
-            e.exchange_buckets(send, bucket_rank, rm, nranks)
+        # all-to-allv.  Skew beyond whole-bucket balancing (SURVEY.md §7
+        # M4): the plan degrades to (source, bucket) CELL granularity —
+        # split buckets leave PARTIAL sums on several ranks, and the
+        # caller runs a second (small) exchange + combine after the
+        # local reduce (returns True when that is needed).
+        if nranks <= 1:
+            return False
+        from dpark_amd.distshuffle import (plan_exchange, recv_matrix_for,
+                                           recv_matrix_for_cells)
+        e.sync_packed(dist)  # all ranks must ship the same key width
+        nbuckets = 256
+        send = e.partition(nbuckets)
+        counts = torch.tensor(send.astype("int64"))
+        mat = [torch.zeros(nbuckets, dtype=torch.int64)
+               for _ in range(nranks)]
+        dist.all_gather(mat, counts)              # count matrix (tiny)
+        m = np.stack([r.numpy() for r in mat]).astype(np.uint64)
+        kind, assign, loads = plan_exchange(m, nranks, cap)
+        imbalance["max"] = max(imbalance.get("max", 0.0),
+                               max(loads) / (sum(loads) / nranks))
+        if kind == "whole":
+            rm = recv_matrix_for(m, assign, rank)
+            e.exchange_buckets(send, assign, rm, nranks)
+            return False
+        rm = recv_matrix_for_cells(m, assign, rank)
+        e.exchange_buckets(send, assign[rank], rm, nranks)
+        return True
+
+    def second_combine(e):
+        # after the local reduce of a cell-split exchange: whole-bucket
+        # exchange of the (small) reduced partials + one more combine
+        from dpark_amd.distshuffle import greedy_assign, recv_matrix_for
+        nbuckets = 256
+        send = e.partition(nbuckets)
+        counts = torch.tensor(send.astype("int64"))
+        mat = [torch.zeros(nbuckets, dtype=torch.int64)
+               for _ in range(nranks)]
+        dist.all_gather(mat, counts)
+        m = np.stack([r.numpy() for r in mat]).astype(np.uint64)
+        bucket_rank, _ = greedy_assign(m.sum(axis=0), nranks)
+        rm = recv_matrix_for(m, bucket_rank, rank)
+        e.exchange_buckets(send, bucket_rank, rm, nranks)
+        return e.reduce_by_key_sum()
 
     imbalance = {}
 
@@ -282,8 +308,9 @@ def main():
         if args.op == "reduce":
             eng.gen_pairs(seed=SEED + rank, n=n_local,
                           key_universe=KEY_UNIVERSE)
-            exchange_if_dist(eng)
-            return eng.reduce_by_key_sum()
+            split = exchange_if_dist(eng)
+            n_u = eng.reduce_by_key_sum()
+            return second_combine(eng) if split else n_u
         if args.op == "sort":
             # config 3: 1B int64 keys uniform over full 64 bits, keys only
             eng.gen_pairs(seed=7 + rank, n=n_local, key_universe=0)
@@ -312,14 +339,39 @@ def main():
             # config 5: Zipf(alpha=1.1) skewed keys over 2^24
             eng.gen_pairs(seed=11 + rank, n=n_local, key_universe=1 << 24,
                           dist=1)
-            exchange_if_dist(eng)
-            return eng.reduce_by_key_sum()
+            split = exchange_if_dist(eng)
+            n_u = eng.reduce_by_key_sum()
+            return second_combine(eng) if split else n_u
         # config 4 shape: two sides, keys from 2^26 universe, co-partition
-        # (shared hash digits) -> sort both -> merge join
+        # (shared hash digits) -> sort both -> merge join.  BOTH sides
+        # must use ONE bucket->rank map (key co-location); the plan is
+        # built from the summed count matrices, whole buckets only (a
+        # cell split would break co-partitioning).
         eng.gen_pairs(seed=13 + rank, n=n_local, key_universe=1 << 26)
         eng_b.gen_pairs(seed=17 + rank, n=n_local, key_universe=1 << 26)
-        exchange_if_dist(eng)
-        exchange_if_dist(eng_b)
+        if nranks > 1:
+            from dpark_amd.distshuffle import greedy_assign, recv_matrix_for
+            eng.sync_packed(dist)
+            eng_b.sync_packed(dist)
+            nbuckets = 256
+            mats = []
+            sends = []
+            for e in (eng, eng_b):
+                send = e.partition(nbuckets)
+                sends.append(send)
+                counts = torch.tensor(send.astype("int64"))
+                mat = [torch.zeros(nbuckets, dtype=torch.int64)
+                       for _ in range(nranks)]
+                dist.all_gather(mat, counts)
+                mats.append(np.stack([r.numpy() for r in mat])
+                            .astype(np.uint64))
+            bucket_rank, loads = greedy_assign(mats[0].sum(axis=0)
+                                               + mats[1].sum(axis=0), nranks)
+            imbalance["max"] = max(imbalance.get("max", 0.0),
+                                   max(loads) / (sum(loads) / nranks))
+            for e, m, send in zip((eng, eng_b), mats, sends):
+                rm = recv_matrix_for(m, bucket_rank, rank)
+                e.exchange_buckets(send, bucket_rank, rm, nranks)
         return eng.join_count(eng_b)
 
     # warmup

@@ -40,12 +40,17 @@
 /* ------------------------------------------------------------------ */
 #define RADIX 256
 #define RADIX_BITS 8
+/* TILE 8192 x 1024 threads is the measured default (r02 A/B, same box:
+ * scatter 1.18 -> 0.87 ms/launch on config 2 — half the tiles, half the
+ * barrier convoys, 32-element digit runs).  LDS: staging 96 KiB (u32) /
+ * 144 KiB (i64) + 16 KiB hists fits 160 KiB at 1 WG/CU = 16 waves/CU,
+ * same wave count as the old 4096x512 at 2 WG/CU. */
 #ifndef TILE
-#define TILE 4096          /* elements per radix-pass workgroup (tunable) */
+#define TILE 8192          /* elements per radix-pass workgroup (tunable) */
 #endif
 #define NTHREADS 256       /* 4 waves of 64                              */
 #ifndef NTHREADS_SC
-#define NTHREADS_SC 512    /* scatter workgroup: 8 waves (tunable)       */
+#define NTHREADS_SC 1024   /* scatter workgroup: 16 waves (tunable)      */
 #endif
 #define RTILE 2048         /* elements per reduce/head workgroup (4 WG/CU
                               occupancy: the reduce is latency-bound)    */
@@ -1922,6 +1927,17 @@ static int tile_grid(uint64_t ntiles) {
   return (int)min(ntiles, (uint64_t)cap);
 }
 
+/* lookback L3-residency gate in bytes (granule array must stay L3-hot;
+ * DPRK_LB_GATE_MB tunes, default 200 — see dprk_sort_pairs) */
+static uint64_t lb_gate_bytes() {
+  static long gate_mb = -1;
+  if (gate_mb < 0) {
+    const char *e = getenv("DPRK_LB_GATE_MB");
+    gate_mb = e ? atol(e) : 200;
+  }
+  return (uint64_t)gate_mb << 20;
+}
+
 extern "C" int dprk_init(dprk_ctx **out, int device_id) {
   *out = nullptr;
   HIP_CHECK(hipSetDevice(device_id));
@@ -2144,7 +2160,7 @@ static int gen_pairs_t(dprk_ctx *c, uint64_t seed, uint64_t n, int dist,
     /* the global hists feed only the lookback path; when n exceeds its
      * L3-residency gate (see dprk_sort_pairs) skip the extra LDS-atomic
      * work here */
-    if (ceil_div(n, (uint64_t)TILE) * RADIX * 8 > (200ull << 20))
+    if (ceil_div(n, (uint64_t)TILE) * RADIX * 8 > lb_gate_bytes())
       nact = 0;
     if (!c->d_ghist) HIP_CHECK(hipMalloc(&c->d_ghist, 8 * RADIX * 4));
     HIP_CHECK(hipMemsetAsync(c->d_ghist, 0, 8 * RADIX * 4, c->stream));
@@ -2462,7 +2478,7 @@ static int radix_pass_t(dprk_ctx *c, int mode, int shift, uint32_t mask,
                            0, c->stream, c->tile_hist, ntiles, seg_tiles,
                            c->seg_sums, c->tile_off));
   static int rank_atomic = -1;
-  if (rank_atomic < 0) rank_atomic = getenv("DPRK_RANK_ATOMIC") ? 1 : 0;
+  if (rank_atomic < 0) rank_atomic = getenv("DPRK_RANK_BALLOT") ? 0 : 1;  /* atomic rank default (r02 A/B) */
   if (c->with_vals) {
     if (rank_atomic) {
       TIMED(c, K_SCATTER,
@@ -2670,7 +2686,7 @@ static int sort_pairs_lookback(dprk_ctx *c, const int *passes, int np) {
     }
     int grid = (int)min(ntiles, (uint64_t)(c->with_vals ? occ_v : occ_k));
     static int rank_atomic = -1;
-    if (rank_atomic < 0) rank_atomic = getenv("DPRK_RANK_ATOMIC") ? 1 : 0;
+    if (rank_atomic < 0) rank_atomic = getenv("DPRK_RANK_BALLOT") ? 0 : 1;  /* atomic rank default (r02 A/B) */
     if (c->with_vals) {
       if (rank_atomic) {
         TIMED(c, K_SCATTER,
@@ -2780,7 +2796,7 @@ extern "C" int dprk_sort_pairs(dprk_ctx *c) {
    * a pure function of n => path choice stays input-deterministic. */
   uint64_t lb_tiles = ceil_div(c->n, (uint64_t)TILE);
   if (np && !getenv("DPRK_NO_LOOKBACK") &&
-      lb_tiles * RADIX * 8 <= (200ull << 20))
+      lb_tiles * RADIX * 8 <= lb_gate_bytes())
     return c->packed ? sort_pairs_lookback<uint32_t>(c, passes, np)
                      : sort_pairs_lookback<int64_t>(c, passes, np);
   for (int i = 0; i < np; i++) {
