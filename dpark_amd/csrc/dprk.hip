@@ -1928,14 +1928,19 @@ static int tile_grid(uint64_t ntiles) {
 }
 
 /* lookback L3-residency gate in bytes (granule array must stay L3-hot;
- * DPRK_LB_GATE_MB tunes, default 200 — see dprk_sort_pairs) */
-static uint64_t lb_gate_bytes() {
+ * DPRK_LB_GATE_MB tunes).  Defaults measured r02 on the 1B shapes:
+ * with-values pairs win up to 300 MB (28.4 -> 29.5 G pairs/s at 1B),
+ * keys-only sort loses past 200 (16.0 -> 15.6 G keys/s) — the gate is a
+ * pure function of (n, with_vals), so path choice stays
+ * input-deterministic. */
+static uint64_t lb_gate_bytes(int with_vals) {
   static long gate_mb = -1;
   if (gate_mb < 0) {
     const char *e = getenv("DPRK_LB_GATE_MB");
-    gate_mb = e ? atol(e) : 200;
+    gate_mb = e ? atol(e) : 0;
   }
-  return (uint64_t)gate_mb << 20;
+  if (gate_mb > 0) return (uint64_t)gate_mb << 20;
+  return (uint64_t)(with_vals ? 300 : 200) << 20;
 }
 
 extern "C" int dprk_init(dprk_ctx **out, int device_id) {
@@ -2160,7 +2165,7 @@ static int gen_pairs_t(dprk_ctx *c, uint64_t seed, uint64_t n, int dist,
     /* the global hists feed only the lookback path; when n exceeds its
      * L3-residency gate (see dprk_sort_pairs) skip the extra LDS-atomic
      * work here */
-    if (ceil_div(n, (uint64_t)TILE) * RADIX * 8 > lb_gate_bytes())
+    if (ceil_div(n, (uint64_t)TILE) * RADIX * 8 > lb_gate_bytes(c->with_vals))
       nact = 0;
     if (!c->d_ghist) HIP_CHECK(hipMalloc(&c->d_ghist, 8 * RADIX * 4));
     HIP_CHECK(hipMemsetAsync(c->d_ghist, 0, 8 * RADIX * 4, c->stream));
@@ -2796,7 +2801,7 @@ extern "C" int dprk_sort_pairs(dprk_ctx *c) {
    * a pure function of n => path choice stays input-deterministic. */
   uint64_t lb_tiles = ceil_div(c->n, (uint64_t)TILE);
   if (np && !getenv("DPRK_NO_LOOKBACK") &&
-      lb_tiles * RADIX * 8 <= lb_gate_bytes())
+      lb_tiles * RADIX * 8 <= lb_gate_bytes(c->with_vals))
     return c->packed ? sort_pairs_lookback<uint32_t>(c, passes, np)
                      : sort_pairs_lookback<int64_t>(c, passes, np);
   for (int i = 0; i < np; i++) {

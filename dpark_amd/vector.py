@@ -120,6 +120,39 @@ class VectorPairsRDD(RDD):
             return JoinedVectorRDD(self.ctx, jk, jva, jvb)
         return super().join(other, numSplits)
 
+    def cogroup(self, other, numSplits=None):
+        """Engine-level cogroup: k -> ([vs], [ws])
+        for the UNION of both key sets (CoGroupedRDD's output shape [U,
+        SURVEY.md §3.3]).  Heavy work (sort + run extraction + value
+        grouping) on GPU per side; the union-merge of the two SORTED
+        unique-key lists (tiny vs the data) is host-side numpy."""
+        if not (isinstance(other, VectorPairsRDD) and engine_required()):
+            return super().cogroup(other, numSplits)
+        ea = _engine.Engine(device=0)
+        eb = _engine.Engine(device=0)
+        try:
+            ea.reserve(max(len(self.keys), 1), with_vals=True)
+            ea.put(self.keys, self.vals)
+            uka, sta, cta = ea.group_by_key()
+            _, va = ea.get()          # values grouped in key order
+            eb.reserve(max(len(other.keys), 1), with_vals=True)
+            eb.put(other.keys, other.vals)
+            ukb, stb, ctb = eb.group_by_key()
+            _, vb = eb.get()
+        finally:
+            ea.close()
+            eb.close()
+        uk = np.union1d(uka, ukb)     # both inputs sorted: O(g)
+        ia = np.searchsorted(uka, uk)
+        in_a = (ia < len(uka))
+        in_a[in_a] &= uka[ia[in_a]] == uk[in_a]
+        ib = np.searchsorted(ukb, uk)
+        in_b = (ib < len(ukb))
+        in_b[in_b] &= ukb[ib[in_b]] == uk[in_b]
+        return CoGroupedVectorRDD(self.ctx, uk,
+                                  in_a, ia, sta, cta, va,
+                                  in_b, ib, stb, ctb, vb)
+
     # -- array-level accessors (bulk results without tuple boxing) -------
     def collectArrays(self):
         return self.keys, self.vals
@@ -147,6 +180,47 @@ class GroupedVectorRDD(RDD):
     def collectArrays(self):
         return (self.unique_keys, self.starts, self.counts,
                 self.grouped_vals)
+
+
+class CoGroupedVectorRDD(RDD):
+    """cogroup result: per union key, value lists from each side (empty
+    list where the key is absent — CoGroupMerger semantics [U])."""
+
+    def __init__(self, ctx, uk, in_a, ia, sta, cta, va,
+                 in_b, ib, stb, ctb, vb):
+        super().__init__(ctx)
+        self.unique_keys = uk
+        self._a = (in_a, ia, sta, cta, va)
+        self._b = (in_b, ib, stb, ctb, vb)
+        self._splits = [Split(0)]
+
+    def splits(self):
+        return self._splits
+
+    def _side(self, side, i):
+        present, idx, starts, counts, vals = side
+        if not present[i]:
+            return []
+        j = int(idx[i])
+        s, c = int(starts[j]), int(counts[j])
+        return vals[s:s + c].tolist()
+
+    def compute(self, split):
+        for i, k in enumerate(self.unique_keys.tolist()):
+            yield k, (self._side(self._a, i), self._side(self._b, i))
+
+    def collectArrays(self):
+        """(uk, counts_a, grouped_vals_a_or_empty, counts_b, ...) in the
+        flattened per-union-key layout."""
+        (in_a, ia, sta, cta, va) = self._a
+        (in_b, ib, stb, ctb, vb) = self._b
+        ca = np.zeros(len(self.unique_keys), np.int64)
+        cb = np.zeros(len(self.unique_keys), np.int64)
+        if len(cta):
+            ca[in_a] = cta.astype(np.int64)[ia[in_a]]
+        if len(ctb):
+            cb[in_b] = ctb.astype(np.int64)[ib[in_b]]
+        return self.unique_keys, ca, cb
 
 
 class JoinedVectorRDD(RDD):

@@ -628,6 +628,90 @@ def test_string_api_routing():
     assert all(got[w] == c for w, c in want.items())
 
 
+def test_two_stage_skew_flow_self():
+    """N=1 mechanics of the skew cell-split path:
+    cell-granular bucket_rank row + local reduce + SECOND bucket
+    exchange + combine — parity vs the oracle through the real RCCL
+    self path.  (The capacity-overflow case needs world>=2 and is
+    covered by tests/test_dist_cpu.py::test_world2_skew_split...)"""
+    from dpark_amd.distshuffle import (greedy_assign_cells,
+                                       recv_matrix_for_cells,
+                                       greedy_assign, recv_matrix_for)
+    os.environ["DPRK_FORCE_SELF_EXCHANGE"] = "1"
+    try:
+        e = Engine(device=0)
+        e.comm_init(1, 0, Engine.comm_unique_id())
+        # heavy skew: one key dominates
+        rng = np.random.default_rng(31)
+        n = 1_000_000
+        keys = np.where(rng.random(n) < 0.6, 7,
+                        rng.integers(0, 1 << 16, n)).astype(np.int64)
+        vals = rng.random(n)
+        e.reserve(n, with_vals=True)
+        e.put(keys, vals)
+        send = e.partition(256)
+        m = send[None, :]
+        cells, loads = greedy_assign_cells(m, 1, cap=n)
+        rm = recv_matrix_for_cells(m, cells, 0)
+        e.exchange_buckets(send, cells[0], rm, 1)
+        e.reduce_by_key_sum()
+        # second stage: whole-bucket exchange of reduced partials
+        send2 = e.partition(256)
+        m2 = send2[None, :]
+        br2, _ = greedy_assign(m2.sum(axis=0), 1)
+        rm2 = recv_matrix_for(m2, br2, 0)
+        e.exchange_buckets(send2, br2, rm2, 1)
+        got_n = e.reduce_by_key_sum()
+        k, v = e.get()
+        wk, wv = so.reduce_by_key_sum(keys, vals)
+        assert got_n == len(wk) and np.array_equal(k, wk)
+        np.testing.assert_allclose(v, wv, rtol=RTOL)
+        e.close()
+    finally:
+        del os.environ["DPRK_FORCE_SELF_EXCHANGE"]
+
+
+def test_cogroup_engine_path():
+    """Engine-level cogroup: k -> ([vs], [ws]) over
+    the UNION of key sets, value lists as multisets, empty list where a
+    key is absent — compared per key against a dict-built oracle."""
+    from dpark_amd import DparkContext
+    from dpark_amd.vector import CoGroupedVectorRDD
+    rng = np.random.default_rng(23)
+    # overlapping + disjoint keys: a in [0, 3000), b in [1500, 4500)
+    ka = rng.integers(0, 3000, size=120_000).astype(np.int64)
+    kb = rng.integers(1500, 4500, size=80_000).astype(np.int64)
+    va = rng.random(len(ka))
+    vb = rng.random(len(kb))
+    ctx = DparkContext("local[4]")
+    ra = ctx.vectorPairs(ka, va, 4)
+    rb = ctx.vectorPairs(kb, vb, 4)
+    cg = ra.cogroup(rb)
+    assert isinstance(cg, CoGroupedVectorRDD)
+    want_a, want_b = {}, {}
+    for k, v in zip(ka.tolist(), va.tolist()):
+        want_a.setdefault(k, []).append(v)
+    for k, v in zip(kb.tolist(), vb.tolist()):
+        want_b.setdefault(k, []).append(v)
+    all_keys = sorted(set(want_a) | set(want_b))
+    got = list(cg.compute(cg.splits()[0]))
+    assert [k for k, _ in got] == all_keys
+    for k, (vs, ws) in got:
+        assert sorted(vs) == sorted(want_a.get(k, [])), k
+        assert sorted(ws) == sorted(want_b.get(k, [])), k
+    uk, ca, cb = cg.collectArrays()
+    assert np.array_equal(uk, np.array(all_keys))
+    assert np.array_equal(ca, np.array([len(want_a.get(k, []))
+                                        for k in all_keys]))
+    assert np.array_equal(cb, np.array([len(want_b.get(k, []))
+                                        for k in all_keys]))
+    # one side empty
+    cg2 = ra.cogroup(ctx.vectorPairs(np.empty(0, np.int64),
+                                     np.empty(0), 2))
+    uk2, ca2, cb2 = cg2.collectArrays()
+    assert np.array_equal(uk2, np.unique(ka)) and cb2.sum() == 0
+
+
 def test_bucket_exchange_self():
     """Bucket-granular RCCL exchange on the 1-GPU box: 256 buckets, all
     assigned to rank 0 via the real grouped per-bucket ncclSend/Recv
