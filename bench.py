@@ -153,6 +153,82 @@ def run_wordcount(args):
     eb.close()
 
 
+def run_bigreduce(args):
+    """Out-of-core reduceByKey: n >= 2^32 pairs
+    through the chunked bucket-arena path (default 8B pairs = 128 GB of
+    int64+f64 input, keys uniform over 2^26).  One step = the whole job:
+    chunked device generation + ingest (hash partition into arenas) +
+    per-bucket sort/reduce.  Conservation (device input sum vs output
+    sum) asserted every step."""
+    from dpark_amd.engine import Engine
+    n_total = args.n_pairs or 8_000_000_000
+    CHUNK = 1 << 28
+    UNI = 1 << 26
+    eng = Engine(device=0)
+    eng.reserve(CHUNK, with_vals=True)
+    log(f"bigreduce: {n_total} pairs ({n_total * 16 / 2**30:.0f} GiB of "
+        f"int64+f64 input) in chunks of {CHUNK}")
+
+    def job():
+        eng.big_begin(n_total, nbuckets=256)
+        start = 0
+        in_sum = 0.0
+        while start < n_total:
+            take = min(CHUNK, n_total - start)
+            eng.gen_pairs(seed=SEED, n=take, key_universe=UNI, start=start)
+            in_sum += eng.val_sum()   # conservation leg (8 B/pair read)
+            eng.big_add_chunk()
+            start += take
+        nu = eng.big_reduce()
+        out_sum = eng.big_val_sum()
+        assert abs(out_sum - in_sum) <= 1e-9 * abs(in_sum), \
+            (out_sum, in_sum)
+        eng.big_free()
+        return nu
+
+    nu = None
+    for _ in range(args.warmup):
+        nu = job()
+    log(f"bigreduce: warmup done, {nu} uniques")
+    eng.stats_reset()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        nu = job()
+    elapsed = time.perf_counter() - t0
+    s = eng.stats()
+    value = n_total * args.steps / elapsed / 1e6
+    roofline = None
+    if s["n_scatter"] > 0 and s["ms_scatter"] > 0:
+        achieved = s["bytes_scatter"] / s["ms_scatter"] * 1e-6
+        roofline = {"bound": "hbm", "achieved": round(achieved, 1),
+                    "peak": HBM_PEAK_SPEC, "unit": "GB/s",
+                    "frac": round(achieved / HBM_PEAK_SPEC, 4),
+                    "frac_vs_measured_copy":
+                        round(achieved / HBM_PEAK_MEASURED, 4),
+                    "traffic": None, "kernel": "k_scatter",
+                    "ms_per_launch": round(s["ms_scatter"] / s["n_scatter"],
+                                           4),
+                    "launches": int(s["n_scatter"])}
+    print(json.dumps({
+        "metric": "M KV-pairs/sec through out-of-core reduceByKey",
+        "value": round(value, 2), "unit": "M KV-pairs/s", "n_gpus": 1,
+        "steps": args.steps, "warmup": args.warmup,
+        "ms_per_step": round(elapsed * 1000 / args.steps, 3),
+        "higher_is_better": True, "scaling": "weak", "vs_baseline": None,
+        "dtype": "int64+f64", "data": "synthetic",
+        "config": {"workload": f"out-of-core-reduceByKey-{n_total}-pairs-"
+                               f"2^26-universe (>=2^32 row, SURVEY.md "
+                               f"§8(f)2; step includes chunked gen + "
+                               f"ingest + conservation check)",
+                   "op": "bigreduce", "n_pairs_per_gpu": n_total,
+                   "chunk": CHUNK, "key_universe": "2^26",
+                   "n_uniques": int(nu), "seed": SEED,
+                   "parallelism": "single-gpu",
+                   "engine": "libdprk.so (HIP gfx950)"},
+        "roofline": roofline, "cpu_baseline": None}))
+    eng.close()
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
@@ -161,13 +237,14 @@ def main():
     ap.add_argument("--n-pairs", type=int, default=None,
                     help="pairs per GPU (default: workload-specific)")
     ap.add_argument("--op", choices=["reduce", "sort", "join", "zipf",
-                                     "wordcount"],
+                                     "wordcount", "bigreduce"],
                     default="reduce",
                     help="reduce=config 2 (headline), sort=config 3 "
                          "(1B keys-only LSD sort), join=config 4 shape, "
                          "zipf=config 5 (1B Zipf(1.1) skewed reduceByKey); "
                          "wordcount=config 1 semantics on the GPU "
-                         "string-key path (N=1 only)")
+                         "string-key path; bigreduce=out-of-core >=2^32 "
+                         "pairs (both N=1 only)")
     ap.add_argument("--config1", action="store_true",
                     help="run the CPU plumbing check instead")
     ap.add_argument("--skip-cpu-baseline", action="store_true")
@@ -178,6 +255,10 @@ def main():
     if args.op == "wordcount":
         assert args.gpus == 1, "wordcount leg is single-GPU"
         run_wordcount(args)
+        return
+    if args.op == "bigreduce":
+        assert args.gpus == 1, "bigreduce leg is single-GPU"
+        run_bigreduce(args)
         return
 
     if args.gpus > 1 and "WORLD_SIZE" not in os.environ:

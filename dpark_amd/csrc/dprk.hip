@@ -190,6 +190,21 @@ struct dprk_ctx {
   uint32_t *f_lex = nullptr;    /* tile-local digit bases */
   uint32_t *f_goff = nullptr;   /* d-major global fragment offsets */
 
+  /* out-of-core "big" aggregate (n >= 2^32):
+   * chunk-partitioned bucket arenas on device, reduced bucket-by-bucket
+   * through the normal in-core pipeline.  dpark's disk-spill external
+   * Merger is the semantic precedent [U]. */
+  int64_t *big_k = nullptr;       /* keys arena (u32 when big_packed)   */
+  double *big_v = nullptr;        /* vals arena                          */
+  uint64_t big_cap_total = 0;
+  uint64_t big_cap_b = 0;         /* per-bucket capacity                 */
+  int big_nbuckets = 0;
+  int big_packed = -1;            /* set by the first chunk              */
+  uint64_t big_fill[256] = {0};
+  uint64_t big_out_n = 0;         /* after big_reduce: output pairs      */
+  uint64_t big_or = 0, big_and = ~0ull;
+  int big_valid = 0;
+
   /* comm */
   ncclComm_t comm = nullptr;
   int nranks = 1, rank = 0;
@@ -582,6 +597,27 @@ __global__ __launch_bounds__(NTHREADS) void k_hist_all(
       for (int w = 0; w < NTHREADS / 64; w++) s += h[w][p][d];
       if (s) atomicAdd(&g_hist[p * RADIX + d], s);
     }
+}
+
+/* deterministic f64 value sum (conservation checks at out-of-core sizes
+ * the CPU oracle cannot reach): fixed per-block partial tree, partials
+ * combined in block order host-side */
+__global__ __launch_bounds__(NTHREADS) void k_val_sum(
+    const double *__restrict__ vals, uint64_t n,
+    double *__restrict__ partials /* [gridDim.x] */) {
+  __shared__ double sh[NTHREADS];
+  double acc = 0.0;
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    acc += vals[i];
+  sh[threadIdx.x] = acc;
+  __syncthreads();
+  for (int off = NTHREADS / 2; off; off >>= 1) {
+    if (threadIdx.x < off) sh[threadIdx.x] += sh[threadIdx.x + off];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) partials[blockIdx.x] = sh[0];
 }
 
 /* OR/AND reduction of biased keys -> constant-digit (skip-pass) detection */
@@ -2023,8 +2059,24 @@ static int ensure_fused(dprk_ctx *c, uint64_t nrec) {
   return 0;
 }
 
+static int free_big(dprk_ctx *c) {
+  if (c->big_k) HIP_CHECK(hipFree(c->big_k));
+  if (c->big_v) HIP_CHECK(hipFree(c->big_v));
+  c->big_k = nullptr;
+  c->big_v = nullptr;
+  c->big_cap_total = 0;
+  c->big_cap_b = 0;
+  c->big_nbuckets = 0;
+  c->big_packed = -1;
+  c->big_out_n = 0;
+  c->big_valid = 0;
+  memset(c->big_fill, 0, sizeof(c->big_fill));
+  return 0;
+}
+
 static int free_all(dprk_ctx *c) {
   if (free_fused(c)) return 1;
+  if (free_big(c)) return 1;
   for (int i = 0; i < 2; i++) {
     if (c->keys[i]) HIP_CHECK(hipFree(c->keys[i]));
     if (c->vals[i]) HIP_CHECK(hipFree(c->vals[i]));
@@ -2135,7 +2187,7 @@ extern "C" int dprk_set_alias_table(dprk_ctx *c, const uint64_t *table,
 }
 
 template <typename K>
-static int gen_pairs_t(dprk_ctx *c, uint64_t seed, uint64_t n, int dist,
+static int gen_pairs_t(dprk_ctx *c, uint64_t seed_hi, uint64_t n, int dist,
                        uint64_t key_universe, uint64_t keymask) {
   if (dist == 1) {
     TIMED(c, K_GEN,
@@ -2143,7 +2195,7 @@ static int gen_pairs_t(dprk_ctx *c, uint64_t seed, uint64_t n, int dist,
                              dim3(NTHREADS),
                              0, c->stream, (K *)c->keys[c->cur],
                              c->with_vals ? c->vals[c->cur] : nullptr, n,
-                             seed << 32, c->alias_table, key_universe));
+                             seed_hi, c->alias_table, key_universe));
   } else {
     /* fused gen+hist: per-tile hist for the first sort pass (base path)
      * AND global per-pass histograms (lookback path) come out of the
@@ -2179,7 +2231,7 @@ static int gen_pairs_t(dprk_ctx *c, uint64_t seed, uint64_t n, int dist,
                              dim3(NTHREADS), 0, c->stream,
                              (K *)c->keys[c->cur],
                              c->with_vals ? c->vals[c->cur] : nullptr, n,
-                             seed << 32, keymask, c->tile_hist, ntiles,
+                             seed_hi, keymask, c->tile_hist, ntiles,
                              shift0, shifts, nact, c->d_ghist, c->d_orand));
     uint64_t oa[2];
     HIP_CHECK(hipMemcpyAsync(oa, c->d_orand, 16, hipMemcpyDeviceToHost,
@@ -2218,8 +2270,12 @@ static int gen_pairs_t(dprk_ctx *c, uint64_t seed, uint64_t n, int dist,
   return drain_timing(c);
 }
 
-extern "C" int dprk_gen_pairs(dprk_ctx *c, uint64_t seed, uint64_t n,
-                              int dist, uint64_t key_universe) {
+/* start: index offset into the synthetic stream (x_i = (seed<<32) +
+ * start + i) — lets out-of-core chunks continue ONE stream; start=0
+ * reproduces the original spec exactly. */
+extern "C" int dprk_gen_pairs_at(dprk_ctx *c, uint64_t seed, uint64_t start,
+                                 uint64_t n, int dist,
+                                 uint64_t key_universe) {
   if (!c) { set_err("null ctx"); return 1; }
   if (dist != 0 && dist != 1) { set_err("gen: unknown dist %d", dist);
                                 return 1; }
@@ -2240,9 +2296,17 @@ extern "C" int dprk_gen_pairs(dprk_ctx *c, uint64_t seed, uint64_t n,
    * DPRK_NO_PACK=1 disables (A/B harness). */
   c->packed = (key_universe != 0 && key_universe <= (1ull << 32) &&
                !getenv("DPRK_NO_PACK"));
+  uint64_t seed_hi = (seed << 32) + start;
   return c->packed
-             ? gen_pairs_t<uint32_t>(c, seed, n, dist, key_universe, keymask)
-             : gen_pairs_t<int64_t>(c, seed, n, dist, key_universe, keymask);
+             ? gen_pairs_t<uint32_t>(c, seed_hi, n, dist, key_universe,
+                                     keymask)
+             : gen_pairs_t<int64_t>(c, seed_hi, n, dist, key_universe,
+                                    keymask);
+}
+
+extern "C" int dprk_gen_pairs(dprk_ctx *c, uint64_t seed, uint64_t n,
+                              int dist, uint64_t key_universe) {
+  return dprk_gen_pairs_at(c, seed, 0, n, dist, key_universe);
 }
 
 /* put-side pack gate: run k_minmax_or on the fresh int64 keys (the sort
@@ -2448,6 +2512,250 @@ extern "C" int dprk_set_packed(dprk_ctx *c, int want) {
   c->known_and &= 0xFFFFFFFFull;
   c->stats.bytes_total += c->n * 12;
   return drain_timing(c);
+}
+
+/* deterministic f64 sum of the current values (conservation checks at
+ * out-of-core sizes): per-block fixed-tree partials + host combine in
+ * block order. */
+extern "C" int dprk_val_sum(dprk_ctx *c, double *out) {
+  if (!c || !out) { set_err("null arg"); return 1; }
+  if (!c->with_vals) { set_err("val_sum: keys-only ctx"); return 1; }
+  HIP_CHECK(hipSetDevice(c->device));
+  *out = 0.0;
+  if (c->n == 0) return 0;
+  int grid = stream_grid(c->n);
+  double *d_part = nullptr;
+  HIP_CHECK(hipMalloc(&d_part, (uint64_t)grid * 8));
+  TIMED(c, K_OTHER,
+        hipLaunchKernelGGL(k_val_sum, dim3(grid), dim3(NTHREADS), 0,
+                           c->stream, c->vals[c->cur], c->n, d_part));
+  double *h = (double *)malloc((uint64_t)grid * 8);
+  HIP_CHECK(hipMemcpyAsync(h, d_part, (uint64_t)grid * 8,
+                           hipMemcpyDeviceToHost, c->stream));
+  HIP_CHECK(hipStreamSynchronize(c->stream));
+  double s = 0.0;
+  for (int i = 0; i < grid; i++) s += h[i];
+  free(h);
+  HIP_CHECK(hipFree(d_part));
+  *out = s;
+  return drain_timing(c);
+}
+
+/* ------------------------------------------------------------------ */
+/* out-of-core "big" reduceByKey (n >= 2^32).  */
+/* Chunks (each < 2^32, ingested via gen_pairs_at / put_pairs) are     */
+/* hash-partitioned (the same splitmix64 top-digit HashPartitioner as  */
+/* the multi-rank path) into nbuckets device ARENAS; every key lands   */
+/* wholly in one bucket, so each bucket then runs the normal in-core   */
+/* sort + segmented reduce, writing its (unique key, sum) output       */
+/* compacted forward in the arena.  Output order: bucket-major (hash   */
+/* digit), key-sorted within a bucket — documented, not globally key-  */
+/* sorted (reduceByKey's contract is a map).         */
+/* dpark's external sorted-run Merger is the semantic precedent [U].   */
+/* ------------------------------------------------------------------ */
+
+extern "C" int dprk_big_begin(dprk_ctx *c, uint64_t total_cap,
+                              int nbuckets) {
+  if (!c) { set_err("null ctx"); return 1; }
+  if (nbuckets < 2 || nbuckets > 256 || (nbuckets & (nbuckets - 1))) {
+    set_err("big_begin: nbuckets must be a power of two in [2,256]");
+    return 1;
+  }
+  if (!c->with_vals) { set_err("big_begin: reserve with_vals first");
+                       return 1; }
+  HIP_CHECK(hipSetDevice(c->device));
+  if (free_big(c)) return 1;
+  /* arenas are allocated lazily at the first chunk (key width unknown
+   * until then); per-bucket slack covers hash-uniform fluctuation —
+   * heavy skew that overflows a bucket errors clearly (hot-key
+   * pre-reduction is the documented mitigation) */
+  c->big_cap_total = total_cap;
+  c->big_cap_b = total_cap / nbuckets + total_cap / (nbuckets * 8) + 4096;
+  c->big_nbuckets = nbuckets;
+  c->big_packed = -1;
+  c->big_or = 0;
+  c->big_and = ~0ull;
+  c->big_valid = 1;
+  return 0;
+}
+
+extern "C" int dprk_big_add_chunk(dprk_ctx *c) {
+  if (!c || !c->big_valid) { set_err("big_add_chunk: call big_begin");
+                             return 1; }
+  if (c->n == 0) return 0;
+  HIP_CHECK(hipSetDevice(c->device));
+  if (c->big_packed < 0) {
+    c->big_packed = c->packed;
+    uint64_t ksz = c->big_packed ? 4 : 8;
+    uint64_t nb = (uint64_t)c->big_nbuckets;
+    if (hipMalloc(&c->big_k, nb * c->big_cap_b * ksz) != hipSuccess) {
+      set_err("big: keys arena alloc failed (%llu B)",
+              (unsigned long long)(nb * c->big_cap_b * ksz));
+      return 1;
+    }
+    if (hipMalloc(&c->big_v, nb * c->big_cap_b * 8) != hipSuccess) {
+      (void)hipFree(c->big_k);
+      c->big_k = nullptr;
+      set_err("big: vals arena alloc failed");
+      return 1;
+    }
+  }
+  if (c->packed != c->big_packed) {
+    /* align the chunk's key width to the aggregate's (first chunk wins;
+     * set_packed(1) fails loudly if the chunk's keys do not fit) */
+    if (dprk_set_packed(c, c->big_packed)) return 1;
+  }
+  if (c->orand_valid) {
+    c->big_or |= c->known_or;
+    c->big_and &= c->known_and;
+  } else {
+    c->big_or = ~0ull;
+    c->big_and = 0;
+  }
+  uint64_t counts[256];
+  if (dprk_partition(c, c->big_nbuckets, counts)) return 1;
+  uint64_t ksz = c->big_packed ? 4 : 8;
+  uint64_t off = 0;
+  for (int b = 0; b < c->big_nbuckets; b++) {
+    if (c->big_fill[b] + counts[b] > c->big_cap_b) {
+      set_err("big: bucket %d overflows its arena (%llu + %llu > %llu) — "
+              "skew beyond hash balancing; reduce chunk size or pre-reduce "
+              "chunks", b, (unsigned long long)c->big_fill[b],
+              (unsigned long long)counts[b],
+              (unsigned long long)c->big_cap_b);
+      return 1;
+    }
+    if (counts[b]) {
+      char *dk = (char *)c->big_k + (b * c->big_cap_b + c->big_fill[b]) * ksz;
+      HIP_CHECK(hipMemcpyAsync(dk, (char *)c->keys[c->cur] + off * ksz,
+                               counts[b] * ksz, hipMemcpyDeviceToDevice,
+                               c->stream));
+      HIP_CHECK(hipMemcpyAsync(c->big_v + b * c->big_cap_b + c->big_fill[b],
+                               c->vals[c->cur] + off, counts[b] * 8,
+                               hipMemcpyDeviceToDevice, c->stream));
+      c->big_fill[b] += counts[b];
+    }
+    off += counts[b];
+  }
+  HIP_CHECK(hipStreamSynchronize(c->stream));
+  c->stats.bytes_total += c->n * (ksz + 8) * 2;
+  c->n = 0;   /* chunk consumed */
+  return drain_timing(c);
+}
+
+extern "C" int dprk_big_reduce(dprk_ctx *c, uint64_t *n_out) {
+  if (!c || !c->big_valid) { set_err("big_reduce: call big_begin");
+                             return 1; }
+  if (c->big_packed < 0) { c->big_out_n = 0;
+                           if (n_out) *n_out = 0; return 0; }
+  HIP_CHECK(hipSetDevice(c->device));
+  uint64_t ksz = c->big_packed ? 4 : 8;
+  uint64_t out_off = 0;
+  for (int b = 0; b < c->big_nbuckets; b++) {
+    uint64_t n_b = c->big_fill[b];
+    if (!n_b) continue;
+    if (n_b > c->cap) {
+      set_err("big_reduce: bucket %d (%llu pairs) exceeds the chunk "
+              "capacity %llu", b, (unsigned long long)n_b,
+              (unsigned long long)c->cap);
+      return 1;
+    }
+    char *sk = (char *)c->big_k + b * c->big_cap_b * ksz;
+    HIP_CHECK(hipMemcpyAsync(c->keys[c->cur], sk, n_b * ksz,
+                             hipMemcpyDeviceToDevice, c->stream));
+    HIP_CHECK(hipMemcpyAsync(c->vals[c->cur], c->big_v + b * c->big_cap_b,
+                             n_b * 8, hipMemcpyDeviceToDevice, c->stream));
+    c->n = n_b;
+    c->sorted = 0;
+    c->packed = c->big_packed;
+    c->orand_valid = 1;             /* conservative union over chunks */
+    c->known_or = c->big_or;
+    c->known_and = c->big_and;
+    c->hist_shift_valid = -1;
+    c->ghist_valid = 0;
+    uint64_t nu = 0;
+    if (dprk_reduce_by_key_sum(c, &nu)) return 1;
+    /* compact the bucket's output forward in the arenas (out_off never
+     * passes the consumed input region: output <= input per bucket) */
+    HIP_CHECK(hipMemcpyAsync((char *)c->big_k + out_off * ksz,
+                             c->keys[c->cur], nu * ksz,
+                             hipMemcpyDeviceToDevice, c->stream));
+    HIP_CHECK(hipMemcpyAsync(c->big_v + out_off, c->vals[c->cur], nu * 8,
+                             hipMemcpyDeviceToDevice, c->stream));
+    HIP_CHECK(hipStreamSynchronize(c->stream));
+    c->stats.bytes_total += (n_b + nu) * (ksz + 8);
+    out_off += nu;
+  }
+  c->big_out_n = out_off;
+  c->n = 0;   /* the ctx dataset is scratch now */
+  if (n_out) *n_out = out_off;
+  return 0;
+}
+
+extern "C" int dprk_big_get(dprk_ctx *c, int64_t *keys, double *vals,
+                            uint64_t cap, uint64_t *n_out) {
+  if (!c || !c->big_valid) { set_err("big_get: call big_begin"); return 1; }
+  *n_out = c->big_out_n;
+  if (c->big_out_n > cap) { set_err("big_get: cap too small"); return 1; }
+  HIP_CHECK(hipSetDevice(c->device));
+  if (c->big_out_n == 0) return 0;
+  if (vals)
+    HIP_CHECK(hipMemcpyAsync(vals, c->big_v, c->big_out_n * 8,
+                             hipMemcpyDeviceToHost, c->stream));
+  if (keys) {
+    if (c->big_packed) {
+      /* unpack through the chunk ctx's spare buffer, piecewise */
+      uint64_t done = 0;
+      while (done < c->big_out_n) {
+        uint64_t take = min(c->big_out_n - done, c->cap);
+        hipLaunchKernelGGL(k_unpack_keys, dim3(stream_grid(take)),
+                           dim3(NTHREADS), 0, c->stream,
+                           (const uint32_t *)c->big_k + done,
+                           c->keys[c->cur ^ 1], take);
+        HIP_CHECK(hipGetLastError());
+        HIP_CHECK(hipMemcpyAsync(keys + done, c->keys[c->cur ^ 1], take * 8,
+                                 hipMemcpyDeviceToHost, c->stream));
+        HIP_CHECK(hipStreamSynchronize(c->stream));
+        done += take;
+      }
+    } else {
+      HIP_CHECK(hipMemcpyAsync(keys, c->big_k, c->big_out_n * 8,
+                               hipMemcpyDeviceToHost, c->stream));
+    }
+  }
+  HIP_CHECK(hipStreamSynchronize(c->stream));
+  return 0;
+}
+
+/* big-output val sum without D2H (conservation check) */
+extern "C" int dprk_big_val_sum(dprk_ctx *c, double *out) {
+  if (!c || !c->big_valid || !out) { set_err("big_val_sum: bad args");
+                                     return 1; }
+  HIP_CHECK(hipSetDevice(c->device));
+  *out = 0.0;
+  if (c->big_out_n == 0) return 0;
+  int grid = stream_grid(c->big_out_n);
+  double *d_part = nullptr;
+  HIP_CHECK(hipMalloc(&d_part, (uint64_t)grid * 8));
+  hipLaunchKernelGGL(k_val_sum, dim3(grid), dim3(NTHREADS), 0, c->stream,
+                     c->big_v, c->big_out_n, d_part);
+  HIP_CHECK(hipGetLastError());
+  double *h = (double *)malloc((uint64_t)grid * 8);
+  HIP_CHECK(hipMemcpyAsync(h, d_part, (uint64_t)grid * 8,
+                           hipMemcpyDeviceToHost, c->stream));
+  HIP_CHECK(hipStreamSynchronize(c->stream));
+  double s = 0.0;
+  for (int i = 0; i < grid; i++) s += h[i];
+  free(h);
+  HIP_CHECK(hipFree(d_part));
+  *out = s;
+  return 0;
+}
+
+extern "C" int dprk_big_free(dprk_ctx *c) {
+  if (!c) { set_err("null ctx"); return 1; }
+  HIP_CHECK(hipSetDevice(c->device));
+  return free_big(c);
 }
 
 /* ---- one radix pass: tile_hist -> 3-level scan -> scatter ---------- */

@@ -57,6 +57,9 @@ SYMBOLS = [
     "dprk_init", "dprk_free", "dprk_last_error", "dprk_build_info",
     "dprk_reserve", "dprk_gen_pairs", "dprk_set_alias_table",
     "dprk_put_pairs", "dprk_get_pairs", "dprk_put_bytes_keys",
+    "dprk_gen_pairs_at", "dprk_val_sum",
+    "dprk_big_begin", "dprk_big_add_chunk", "dprk_big_reduce",
+    "dprk_big_get", "dprk_big_val_sum", "dprk_big_free",
     "dprk_count", "dprk_is_packed", "dprk_set_packed",
     "dprk_partition", "dprk_partition_range",
     "dprk_sample_keys", "dprk_sort_pairs",
@@ -103,6 +106,15 @@ def load_library():
                                    ctypes.POINTER(u64)]
     lib.dprk_put_bytes_keys.argtypes = [p, ctypes.c_void_p, u64,
                                         ctypes.c_void_p, u64, ctypes.c_void_p]
+    lib.dprk_gen_pairs_at.argtypes = [p, u64, u64, u64, ctypes.c_int, u64]
+    lib.dprk_val_sum.argtypes = [p, ctypes.POINTER(ctypes.c_double)]
+    lib.dprk_big_begin.argtypes = [p, u64, ctypes.c_int]
+    lib.dprk_big_add_chunk.argtypes = [p]
+    lib.dprk_big_reduce.argtypes = [p, ctypes.POINTER(u64)]
+    lib.dprk_big_get.argtypes = [p, ctypes.c_void_p, ctypes.c_void_p, u64,
+                                 ctypes.POINTER(u64)]
+    lib.dprk_big_val_sum.argtypes = [p, ctypes.POINTER(ctypes.c_double)]
+    lib.dprk_big_free.argtypes = [p]
     lib.dprk_count.argtypes = [p, ctypes.POINTER(u64)]
     lib.dprk_is_packed.argtypes = [p, ctypes.POINTER(ctypes.c_int)]
     lib.dprk_set_packed.argtypes = [p, ctypes.c_int]
@@ -164,10 +176,61 @@ class Engine:
                                                  1 if with_vals else 0),
                "dprk_reserve")
 
-    def gen_pairs(self, seed, n, key_universe=0, dist=0):
-        _check(self._lib, self._lib.dprk_gen_pairs(self._ctx, seed, n, dist,
-                                                   key_universe),
-               "dprk_gen_pairs")
+    def gen_pairs(self, seed, n, key_universe=0, dist=0, start=0):
+        _check(self._lib,
+               self._lib.dprk_gen_pairs_at(self._ctx, seed, start, n, dist,
+                                           key_universe),
+               "dprk_gen_pairs_at")
+
+    def val_sum(self):
+        s = ctypes.c_double()
+        _check(self._lib, self._lib.dprk_val_sum(self._ctx, ctypes.byref(s)),
+               "dprk_val_sum")
+        return s.value
+
+    # -- out-of-core (n >= 2^32) -----------------------------------------
+    def big_begin(self, total_cap, nbuckets=256):
+        _check(self._lib, self._lib.dprk_big_begin(self._ctx, total_cap,
+                                                   nbuckets),
+               "dprk_big_begin")
+
+    def big_add_chunk(self):
+        _check(self._lib, self._lib.dprk_big_add_chunk(self._ctx),
+               "dprk_big_add_chunk")
+
+    def big_reduce(self):
+        n = ctypes.c_uint64()
+        _check(self._lib, self._lib.dprk_big_reduce(self._ctx,
+                                                    ctypes.byref(n)),
+               "dprk_big_reduce")
+        return n.value
+
+    def big_get(self, n=None, with_vals=True):
+        if n is None:
+            n = self.big_reduce_count if hasattr(self, "big_reduce_count") \
+                else None
+        assert n is not None, "pass n from big_reduce()"
+        keys = np.empty(n, dtype=np.int64)
+        vals = np.empty(n, dtype=np.float64) if with_vals else None
+        nout = ctypes.c_uint64()
+        _check(self._lib,
+               self._lib.dprk_big_get(
+                   self._ctx, keys.ctypes.data_as(ctypes.c_void_p),
+                   vals.ctypes.data_as(ctypes.c_void_p) if with_vals else None,
+                   n, ctypes.byref(nout)),
+               "dprk_big_get")
+        return (keys, vals) if with_vals else keys
+
+    def big_val_sum(self):
+        s = ctypes.c_double()
+        _check(self._lib, self._lib.dprk_big_val_sum(self._ctx,
+                                                     ctypes.byref(s)),
+               "dprk_big_val_sum")
+        return s.value
+
+    def big_free(self):
+        _check(self._lib, self._lib.dprk_big_free(self._ctx),
+               "dprk_big_free")
 
     def set_alias_table(self, table):
         import numpy as _np

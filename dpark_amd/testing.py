@@ -32,9 +32,10 @@ class SimEngine:
         self._cap = max_n
         self._with_vals = with_vals
 
-    def gen_pairs(self, seed, n, key_universe=0, dist=0):
+    def gen_pairs(self, seed, n, key_universe=0, dist=0, start=0):
         assert n <= self._cap, "gen: n > reserved cap"
-        k, v = gen_pairs(seed, n, key_universe if key_universe else 1 << 64)
+        k, v = gen_pairs(seed, n, key_universe if key_universe else 1 << 64,
+                         start=start)
         self.keys, self.vals = k, v
 
     def set_alias_table(self, table):

@@ -88,6 +88,29 @@ int dprk_count(dprk_ctx *ctx, uint64_t *n_out);
 int dprk_is_packed(dprk_ctx *ctx, int *out);
 int dprk_set_packed(dprk_ctx *ctx, int want);
 
+/* gen with an index offset into the synthetic stream (x_i = (seed<<32)
+ * + start + i): out-of-core chunks continue ONE stream; start=0 ==
+ * dprk_gen_pairs. */
+int dprk_gen_pairs_at(dprk_ctx *ctx, uint64_t seed, uint64_t start,
+                      uint64_t n, int dist, uint64_t key_universe);
+/* deterministic f64 sum of the current values (conservation checks) */
+int dprk_val_sum(dprk_ctx *ctx, double *out);
+
+/* ---- out-of-core reduceByKey (n >= 2^32) ----
+ * Chunks (< 2^32 each, via gen_pairs_at/put_pairs) are hash-partitioned
+ * into device bucket ARENAS (every key lands wholly in one bucket);
+ * big_reduce then runs the normal in-core sort + segmented reduce per
+ * bucket, compacting (unique key, f64 sum) outputs forward.  Output
+ * order: bucket-major (hash digit), key-sorted within a bucket.
+ * dpark's external sorted-run Merger is the semantic precedent [U]. */
+int dprk_big_begin(dprk_ctx *ctx, uint64_t total_cap, int nbuckets);
+int dprk_big_add_chunk(dprk_ctx *ctx);   /* consumes the current dataset */
+int dprk_big_reduce(dprk_ctx *ctx, uint64_t *n_out);
+int dprk_big_get(dprk_ctx *ctx, int64_t *keys, double *vals, uint64_t cap,
+                 uint64_t *n_out);
+int dprk_big_val_sum(dprk_ctx *ctx, double *out);
+int dprk_big_free(dprk_ctx *ctx);
+
 /* ---- the hot path ---------------------------------------------------- */
 /* Radix-partition the current dataset into nbuckets contiguous segments by
  * the TOP bits of splitmix64(key) (the GPU HashPartitioner — SURVEY.md

@@ -33,11 +33,15 @@ def splitmix64(x):
         np.seterr(**old)
 
 
-def gen_pairs(seed: int, n: int, key_universe: int):
-    """CPU mirror of dprk_gen_pairs(dist=0 uniform). Returns (keys int64,
-    vals float64)."""
-    i = np.arange(n, dtype=np.uint64)
-    x = (np.uint64(seed) << np.uint64(32)) + i
+def gen_pairs(seed: int, n: int, key_universe: int, start: int = 0):
+    """CPU mirror of dprk_gen_pairs_at(dist=0 uniform): x_i = (seed<<32)
+    + start + i (wraparound).  Returns (keys int64, vals float64)."""
+    old = np.seterr(over="ignore")
+    try:
+        i = np.arange(n, dtype=np.uint64)
+        x = (np.uint64(seed) << np.uint64(32)) + np.uint64(start) + i
+    finally:
+        np.seterr(**old)
     ks = splitmix64(x)
     if key_universe != 0 and key_universe < (1 << 64):
         ks = ks & np.uint64(key_universe - 1)

@@ -628,6 +628,103 @@ def test_string_api_routing():
     assert all(got[w] == c for w, c in want.items())
 
 
+def test_big_chunked_reduce_matches_oracle():
+    """Out-of-core machinery at oracle scale:
+    40M pairs ingested as 5 chunks of 8M through the bucket arenas ==
+    the direct oracle reduce of the whole stream.  The SAME machinery
+    then runs at >=2^32 in scripts/big128.sh (property-checked)."""
+    C = 8_000_000
+    total = 40_000_000
+    uni = 1 << 20
+    e = Engine(device=0)
+    try:
+        e.reserve(C, with_vals=True)
+        e.big_begin(total, nbuckets=64)
+        start = 0
+        in_sum = 0.0
+        while start < total:
+            e.gen_pairs(seed=77, n=C, key_universe=uni, start=start)
+            in_sum += e.val_sum()
+            e.big_add_chunk()
+            start += C
+        nu = e.big_reduce()
+        k, v = e.big_get(n=nu)
+        # conservation (device-computed input sum vs output sum)
+        np.testing.assert_allclose(e.big_val_sum(), in_sum, rtol=1e-9)
+        # exact parity vs the oracle on the SAME stream (gen mirror)
+        all_k, all_v = gen_pairs(77, total, uni)
+        wk, wv = so.reduce_by_key_sum(all_k, all_v)
+        assert nu == len(wk)
+        # big output is bucket-major (hash digit), key-sorted within a
+        # bucket — sort both for comparison (documented order)
+        order = np.argsort(k, kind="stable")
+        assert np.array_equal(k[order], wk)
+        np.testing.assert_allclose(v[order], wv, rtol=RTOL)
+        # within-bucket sortedness of the raw output
+        from dpark_amd.hashspec import splitmix64
+        buck = (splitmix64(k.view(np.uint64)) >> np.uint64(58)).astype(int)
+        for b in np.unique(buck):
+            kb = k[buck == b]
+            assert np.all(kb[1:] > kb[:-1])
+        e.big_free()
+    finally:
+        e.close()
+
+
+def test_big_put_chunks_and_errors():
+    """big path fed by put_pairs (host chunks), plus the clear-error
+    contracts (big_begin required; bucket-overflow message)."""
+    from dpark_amd.engine import EngineError
+    e = Engine(device=0)
+    try:
+        e.reserve(100_000, with_vals=True)
+        with pytest.raises(EngineError, match="big_begin"):
+            e.big_add_chunk()
+        e.big_begin(300_000, nbuckets=16)
+        rng = np.random.default_rng(41)
+        all_k, all_v = [], []
+        for i in range(3):
+            keys = rng.integers(-(1 << 40), 1 << 40, 100_000).astype(np.int64)
+            vals = rng.random(100_000)
+            all_k.append(keys)
+            all_v.append(vals)
+            e.put(keys, vals)
+            e.big_add_chunk()
+        nu = e.big_reduce()
+        k, v = e.big_get(n=nu)
+        wk, wv = so.reduce_by_key_sum(np.concatenate(all_k),
+                                      np.concatenate(all_v))
+        assert nu == len(wk)
+        order = np.argsort(k, kind="stable")
+        assert np.array_equal(k[order], wk)
+        np.testing.assert_allclose(v[order], wv, rtol=RTOL)
+        e.big_free()
+        # skew overflow: one key only -> one bucket gets everything
+        e.big_begin(300_000, nbuckets=16)
+        e.put(np.zeros(100_000, dtype=np.int64), np.ones(100_000))
+        e.big_add_chunk()
+        e.put(np.zeros(100_000, dtype=np.int64), np.ones(100_000))
+        with pytest.raises(EngineError, match="overflow"):
+            e.big_add_chunk()
+        e.big_free()
+    finally:
+        e.close()
+
+
+def test_gen_pairs_at_continues_stream():
+    """gen_pairs_at(start) == the corresponding slice of one long gen
+    (device vs numpy mirror, byte-identical)."""
+    e = Engine(device=0)
+    try:
+        e.reserve(100_000, with_vals=True)
+        e.gen_pairs(seed=5, n=50_000, key_universe=1 << 16, start=262_144)
+        gk, gv = e.get()
+        mk, mv = gen_pairs(5, 50_000, 1 << 16, start=262_144)
+        assert np.array_equal(gk, mk) and np.array_equal(gv, mv)
+    finally:
+        e.close()
+
+
 def test_two_stage_skew_flow_self():
     """N=1 mechanics of the skew cell-split path:
     cell-granular bucket_rank row + local reduce + SECOND bucket
