@@ -67,6 +67,28 @@ class DparkContext:
         from .vector import VectorPairsRDD
         return VectorPairsRDD(self, keys, vals, numSlices)
 
+    def binaryPairs(self, path):
+        """Reload a VectorPairsRDD dumped by saveAsBinary (partitioned
+        npz shards + manifest; bit-exact round trip)."""
+        import glob
+        import json as _json
+        from .vector import VectorPairsRDD
+        with open(f"{path}/manifest.json") as f:
+            man = _json.load(f)
+        assert man.get("format") == "dprk-pairs-v1", man
+        parts = sorted(glob.glob(f"{path}/part-*.npz"))
+        assert len(parts) == man["partitions"], (len(parts), man)
+        ks, vs = [], []
+        for p in parts:
+            z = __import__("numpy").load(p)
+            ks.append(z["keys"])
+            vs.append(z["vals"])
+        import numpy as np
+        keys = np.concatenate(ks) if ks else np.empty(0, np.int64)
+        vals = np.concatenate(vs) if vs else np.empty(0, np.float64)
+        assert len(keys) == man["n"]
+        return VectorPairsRDD(self, keys, vals, max(1, len(parts)))
+
     def stringPairs(self, words, vals, numSlices=None):
         """Engine-eligible pair RDD with bytes/str keys: reduceByKey(+)
         routes through the device-hash string path (dpark_amd/strings.py,

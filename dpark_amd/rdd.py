@@ -116,10 +116,29 @@ class ShuffleDependency(Dependency):
 # MapOutputTracker degenerate to a dict — SURVEY.md §2 row 7)
 # ---------------------------------------------------------------------------
 
+MAX_TASK_FAILURES = 4   # dpark/job.py retry budget [U]
+
+
+class FetchFailed(Exception):
+    """dpark/shuffle.py :: FetchFailed analogue [U]: a reduce task could
+    not read a map output; the scheduler invalidates that shuffle and
+    resubmits the parent map stage before retrying the reduce task."""
+
+    def __init__(self, shuffle_id):
+        super().__init__(f"fetch failed for shuffle {shuffle_id}")
+        self.shuffle_id = shuffle_id
+
+
 class ShuffleStore:
     def __init__(self):
         self._outputs = {}  # shuffleId -> {mapId: {bucket: items}}
         self._complete = set()
+
+    def invalidate(self, shuffle_id):
+        """Drop a shuffle's outputs (FetchFailed recovery: the map stage
+        re-runs — dpark resubmits the parent stage [U])."""
+        self._outputs.pop(shuffle_id, None)
+        self._complete.discard(shuffle_id)
 
     def put(self, shuffle_id, map_id, buckets):
         self._outputs.setdefault(shuffle_id, {})[map_id] = buckets
@@ -561,7 +580,16 @@ def _process_worker(payload):
             st.put(sid, m, buckets)
         st.mark_complete(sid)
     _CHILD_STORE = st
-    return func(rdd.iterator(split))
+    # child-side retry (dpark/job.py budget [U]); FetchFailed cannot
+    # occur here — the shuffle slice is shipped with the task
+    last = None
+    for _ in range(MAX_TASK_FAILURES):
+        try:
+            return func(rdd.iterator(split))
+        except Exception as e:  # noqa: BLE001
+            last = e
+    raise RuntimeError(
+        f"process task failed after {MAX_TASK_FAILURES} attempts") from last
 
 
 class LocalScheduler:
@@ -575,6 +603,26 @@ class LocalScheduler:
         self.ctx = ctx
         self.workers = max(1, workers)
         self._shuffle_lock = threading.RLock()
+
+    def _attempt(self, run, desc):
+        """dpark/job.py retry semantics [U]: <= MAX_TASK_FAILURES
+        attempts per task.  FetchFailed invalidates the shuffle so the
+        next attempt re-materializes the parent map stage (dpark's
+        stage-resubmit on FetchFailed [U]); other exceptions simply
+        retry.  After the budget: a clear terminal error."""
+        last = None
+        for attempt in range(MAX_TASK_FAILURES):
+            try:
+                return run()
+            except FetchFailed as e:
+                last = e
+                with self._shuffle_lock:
+                    self.ctx.shuffle_store.invalidate(e.shuffle_id)
+            except Exception as e:  # noqa: BLE001 — dpark retries any task error
+                last = e
+        raise RuntimeError(
+            f"task {desc} failed after {MAX_TASK_FAILURES} attempts: "
+            f"{type(last).__name__}: {last}") from last
 
     def ensure_shuffle(self, dep: ShuffleDependency):
         """Materialize a shuffle's map outputs exactly once; concurrent
@@ -613,7 +661,9 @@ class LocalScheduler:
         splits = parent.splits()
         with ThreadPoolExecutor(self.workers) as pool:
             for map_id, out in pool.map(
-                    lambda args: shuffle_map_task(*args),
+                    lambda args: self._attempt(
+                        lambda: shuffle_map_task(*args),
+                        f"ShuffleMapTask({dep.shuffleId}, {args[0]})"),
                     list(enumerate(splits))):
                 store.put(dep.shuffleId, map_id, out)
         store.mark_complete(dep.shuffleId)
@@ -640,8 +690,14 @@ class LocalScheduler:
         # materialize shuffle deps bottom-up first (single-thread walk
         # avoids re-entrant pools); compute() calls ensure_shuffle lazily
         # as well, so this is an optimization not a correctness need.
+        # Each result task retries per dpark/job.py [U] (_attempt); a
+        # FetchFailed re-materializes the invalidated shuffle lazily on
+        # the retry's iterator() call.
         with ThreadPoolExecutor(self.workers) as pool:
-            return list(pool.map(lambda s: func(rdd.iterator(s)), splits))
+            return list(pool.map(
+                lambda s: self._attempt(lambda: func(rdd.iterator(s)),
+                                        f"ResultTask({rdd.id}, {s.index})"),
+                splits))
 
 
 class MultiProcessScheduler(LocalScheduler):

@@ -69,6 +69,17 @@ class VectorPairsRDD(RDD):
 
     # -- engine-routed hot path ------------------------------------------
     def _engine_run(self, op, other=None):
+        """Engine op with retry-once-on-EngineError (VERDICT r01 item 8 /
+        dpark task-retry analogue on the engine path [U]): a transient
+        EngineError tears the ctx down and retries on a FRESH ctx once;
+        a second failure surfaces.  EngineUnavailable (no native lib on
+        a GPU box) is NOT retried — that is a hard configuration error."""
+        try:
+            return self._engine_run_once(op, other)
+        except _engine.EngineError:
+            return self._engine_run_once(op, other)
+
+    def _engine_run_once(self, op, other=None):
         eng = _engine.Engine(device=0)
         try:
             eng.reserve(max(len(self.keys), 1), with_vals=True)
@@ -156,6 +167,39 @@ class VectorPairsRDD(RDD):
     # -- array-level accessors (bulk results without tuple boxing) -------
     def collectArrays(self):
         return self.keys, self.vals
+
+    # -- output/ingest formats ------------------
+    def saveAsBinary(self, path):
+        """Partitioned binary dump (checkpoint analogue of dpark's
+        OutputTextFileRDD [U]): one part-NNNNN.npz per partition + a
+        manifest; reload with ctx.binaryPairs(path).  Round-trip is
+        bit-exact (int64 keys, float64 vals)."""
+        import json as _json
+        os.makedirs(path, exist_ok=True)
+        for i in range(len(self._splits)):
+            lo, hi = self._bounds[i], self._bounds[i + 1]
+            np.savez(os.path.join(path, f"part-{i:05d}.npz"),
+                     keys=self.keys[lo:hi], vals=self.vals[lo:hi])
+        with open(os.path.join(path, "manifest.json"), "w") as f:
+            _json.dump({"format": "dprk-pairs-v1", "n": len(self.keys),
+                        "partitions": len(self._splits)}, f)
+        return path
+
+    def saveAsTextFile(self, path, ext="", overwrite=True):
+        """GPU-path text dump: same one-file-per-partition layout and
+        str((k, v)) line format as the generic CPU path (rdd.py),
+        written array-at-a-time."""
+        os.makedirs(path, exist_ok=True)
+        out = []
+        for i in range(len(self._splits)):
+            lo, hi = self._bounds[i], self._bounds[i + 1]
+            fname = os.path.join(path, "%05d%s" % (i, ext))
+            with open(fname, "w") as f:
+                f.writelines(f"{(k, v)}\n" for k, v in
+                             zip(self.keys[lo:hi].tolist(),
+                                 self.vals[lo:hi].tolist()))
+            out.append(fname)
+        return out
 
 
 class GroupedVectorRDD(RDD):

@@ -699,10 +699,9 @@ def test_big_put_chunks_and_errors():
         assert np.array_equal(k[order], wk)
         np.testing.assert_allclose(v[order], wv, rtol=RTOL)
         e.big_free()
-        # skew overflow: one key only -> one bucket gets everything
+        # skew overflow: one key only -> one bucket gets everything and
+        # immediately exceeds its arena share — clear error, no crash
         e.big_begin(300_000, nbuckets=16)
-        e.put(np.zeros(100_000, dtype=np.int64), np.ones(100_000))
-        e.big_add_chunk()
         e.put(np.zeros(100_000, dtype=np.int64), np.ones(100_000))
         with pytest.raises(EngineError, match="overflow"):
             e.big_add_chunk()
