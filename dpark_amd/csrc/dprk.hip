@@ -996,6 +996,208 @@ __global__ __launch_bounds__(NTHREADS_SC, 4) void k_scatter_lb(
   }
 }
 
+/* Software-pipelined lookback scatter (DPRK_PIPE experiment; u32 keys
+ * only — the double hist + staging fits 160 KiB LDS only at 12 B/pair):
+ * tile t+grid's global loads are ISSUED during tile t's staging and
+ * consumed after t's write phase, so the ~900-cycle HBM load latency
+ * hides under the lookback walk + store issue — at TILE 8192 there is
+ * only 1 WG/CU, so no partner workgroup fills those gaps (the r01
+ * rejection of cross-tile pipelining was measured at 2 WG/CU).  Rank is
+ * ds_add_rtn (the r02 default).  Identical output permutation to
+ * k_scatter_lb<u32, WV, true>. */
+template <typename K, bool WITH_VALS>
+__global__ __launch_bounds__(NTHREADS_SC, 4) void k_scatter_lb_pipe(
+    const K *__restrict__ keys_in, const double *__restrict__ vals_in,
+    K *__restrict__ keys_out, double *__restrict__ vals_out,
+    uint64_t n, uint64_t ntiles, int shift,
+    const uint32_t *__restrict__ digit_base,
+    unsigned long long *__restrict__ lb,
+    unsigned int *__restrict__ d_fail) {
+  constexpr int NW = NTHREADS_SC / 64;
+  constexpr int WBLK = TILE / NW;
+  constexpr int EPL = WBLK / 64;
+  __shared__ K lk[TILE];
+  __shared__ double lv[WITH_VALS ? TILE : 1];
+  __shared__ uint32_t excl[RADIX];
+  __shared__ uint32_t goff[RADIX];
+  __shared__ uint32_t h[2][NW][RADIX];
+  __shared__ uint32_t wtot[NW > 4 ? NW : 4];
+  __shared__ uint32_t sfail;
+
+  const int wave = threadIdx.x / 64, lane = threadIdx.x % 64;
+  const uint32_t mask = 0xFF;
+  const uint32_t wbeg = wave * WBLK;
+
+  uint64_t t0 = blockIdx.x;
+  if (t0 >= ntiles) return;
+  for (int j = 0; j < RADIX / 64; j++) {
+    h[0][wave][lane * (RADIX / 64) + j] = 0;
+    h[1][wave][lane * (RADIX / 64) + j] = 0;
+  }
+  if (threadIdx.x == 0) sfail = 0;
+
+  K rk0[EPL], rk1[EPL];
+  double rv0[EPL], rv1[EPL];
+  uint32_t rd0[EPL], rd1[EPL];
+
+  /* prolog: load + hist tile t0 into buffer 0 (barrier below publishes
+   * the zeroed hists before the atomics) */
+  __syncthreads();
+  {
+    const uint64_t start = t0 * (uint64_t)TILE;
+    const uint64_t end = min(start + TILE, n);
+    const uint32_t tile_n = (uint32_t)(end - start);
+#pragma unroll
+    for (int i = 0; i < EPL; i++) {
+      uint32_t idx = wbeg + i * 64 + lane;
+      if (idx < tile_n) {
+        rk0[i] = keys_in[start + idx];
+        if (WITH_VALS) rv0[i] = vals_in[start + idx];
+        rd0[i] = d_digit(rk0[i], 0, shift, mask);
+        atomicAdd(&h[0][wave][rd0[i]], 1u);
+      }
+    }
+  }
+
+  /* one pipeline stage: process tile t from (rkc, hc); prefetch + hist
+   * tile t+grid into (rkn, hn).  Returns false on lookback failure. */
+  auto stage = [&](K (&rkc)[EPL], double (&rvc)[EPL], uint32_t (&rdc)[EPL],
+                   K (&rkn)[EPL], double (&rvn)[EPL], uint32_t (&rdn)[EPL],
+                   uint32_t (*hc)[RADIX], uint32_t (*hn)[RADIX],
+                   uint64_t t) -> bool {
+    if (threadIdx.x == 0)       /* cross-block failure bailout */
+      sfail = __hip_atomic_load(d_fail, __ATOMIC_RELAXED,
+                                __HIP_MEMORY_SCOPE_AGENT);
+    __syncthreads();            /* hist(t) complete; sfail visible */
+    if (sfail) return false;
+    const uint64_t start = t * (uint64_t)TILE;
+    const uint64_t end = min(start + TILE, n);
+    const uint32_t tile_n = (uint32_t)(end - start);
+    const uint64_t tn = t + gridDim.x;
+
+    /* phase 2: publish AGG; excl scan; wave cursors in place */
+    uint32_t tile_cnt = 0;
+    if (threadIdx.x < RADIX) {
+      for (int w = 0; w < NW; w++) tile_cnt += hc[w][threadIdx.x];
+      __hip_atomic_store(&lb[t * RADIX + threadIdx.x], LB_AGG | tile_cnt,
+                         __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+      uint32_t inc = d_wave_incl_scan(tile_cnt, lane);
+      if (lane == 63) wtot[wave] = inc;
+      excl[threadIdx.x] = inc - tile_cnt;
+    }
+    __syncthreads();
+    if (threadIdx.x < RADIX) {
+      if (wave > 0) {
+        uint32_t add = 0;
+        for (int w = 0; w < wave; w++) add += wtot[w];
+        excl[threadIdx.x] += add;
+      }
+      uint32_t b = excl[threadIdx.x];
+      for (int w = 0; w < NW; w++) {
+        uint32_t cnt = hc[w][threadIdx.x];
+        hc[w][threadIdx.x] = b;
+        b += cnt;
+      }
+    }
+    __syncthreads();
+
+    /* phase 3: rank (ds_add_rtn) + stage; then ISSUE next tile's loads */
+#pragma unroll
+    for (int i = 0; i < EPL; i++) {
+      uint32_t idx = wbeg + i * 64 + lane;
+      if (idx < tile_n) {
+        uint32_t pos = atomicAdd(&hc[wave][rdc[i]], 1u);
+        lk[pos] = rkc[i];
+        if (WITH_VALS) lv[pos] = rvc[i];
+      }
+    }
+    uint32_t tile_nn = 0;
+    if (tn < ntiles) {
+      const uint64_t startn = tn * (uint64_t)TILE;
+      tile_nn = (uint32_t)(min(startn + TILE, n) - startn);
+#pragma unroll
+      for (int i = 0; i < EPL; i++) {
+        uint32_t idx = wbeg + i * 64 + lane;
+        if (idx < tile_nn) {
+          rkn[i] = keys_in[startn + idx];
+          if (WITH_VALS) rvn[i] = vals_in[startn + idx];
+        }
+      }
+    }
+
+    /* lookback (thread = digit) overlaps the in-flight prefetch */
+    if (threadIdx.x < RADIX) {
+      uint64_t excl_g = 0;
+      uint32_t spins = 0;
+      bool ok = true;
+      int64_t s = (int64_t)t - 1;
+      while (s >= 0) {
+        uint64_t e = __hip_atomic_load(&lb[(uint64_t)s * RADIX + threadIdx.x],
+                                       __ATOMIC_RELAXED,
+                                       __HIP_MEMORY_SCOPE_AGENT);
+        if (!(e >> 62)) {
+          __builtin_amdgcn_s_sleep(2);
+          if (++spins > LB_SPIN_LIMIT) {
+            __hip_atomic_store(d_fail, 1u, __ATOMIC_RELAXED,
+                               __HIP_MEMORY_SCOPE_AGENT);
+            ok = false;
+            break;
+          }
+          continue;
+        }
+        excl_g += e & LB_VAL;
+        if ((e >> 62) >= 2) break;
+        s--;
+      }
+      if (ok) {
+        __hip_atomic_store(&lb[t * RADIX + threadIdx.x],
+                           LB_INC | (excl_g + tile_cnt), __ATOMIC_RELAXED,
+                           __HIP_MEMORY_SCOPE_AGENT);
+        goff[threadIdx.x] = digit_base[threadIdx.x] + (uint32_t)excl_g -
+                            excl[threadIdx.x];
+      } else {
+        sfail = 1;
+      }
+    }
+    __syncthreads();
+    if (sfail) return false;
+
+    /* phase 4: write out; stores drain while the tail runs */
+    for (uint32_t idx = threadIdx.x; idx < tile_n; idx += NTHREADS_SC) {
+      K key = lk[idx];
+      uint32_t dig = d_digit(key, 0, shift, mask);
+      keys_out[goff[dig] + idx] = key;
+      if (WITH_VALS) vals_out[goff[dig] + idx] = lv[idx];
+    }
+
+    /* tail: zero the consumed hist (becomes t+2*grid's buffer) and
+     * histogram the prefetched tile (consumes the early loads) */
+    for (int j = 0; j < RADIX / 64; j++)
+      hc[wave][lane * (RADIX / 64) + j] = 0;
+    if (tn < ntiles) {
+#pragma unroll
+      for (int i = 0; i < EPL; i++) {
+        uint32_t idx = wbeg + i * 64 + lane;
+        if (idx < tile_nn) {
+          rdn[i] = d_digit(rkn[i], 0, shift, mask);
+          atomicAdd(&hn[wave][rdn[i]], 1u);
+        }
+      }
+    }
+    return true;
+  };
+
+  uint64_t t = t0;
+  for (;;) {
+    if (!stage(rk0, rv0, rd0, rk1, rv1, rd1, h[0], h[1], t)) return;
+    t += gridDim.x;
+    if (t >= ntiles) return;
+    if (!stage(rk1, rv1, rd1, rk0, rv0, rd0, h[1], h[0], t)) return;
+    t += gridDim.x;
+    if (t >= ntiles) return;
+  }
+}
+
 template <typename K, bool WITH_VALS, bool RANK_ATOMIC = false>
 __global__ __launch_bounds__(NTHREADS_SC, 2) void k_scatter(
     const K *__restrict__ keys_in, const double *__restrict__ vals_in,
@@ -3000,6 +3202,60 @@ static int sort_pairs_lookback(dprk_ctx *c, const int *passes, int np) {
     int grid = (int)min(ntiles, (uint64_t)(c->with_vals ? occ_v : occ_k));
     static int rank_atomic = -1;
     if (rank_atomic < 0) rank_atomic = getenv("DPRK_RANK_BALLOT") ? 0 : 1;  /* atomic rank default (r02 A/B) */
+    static int pipe = -1;
+    if (pipe < 0) pipe = getenv("DPRK_PIPE") ? 1 : 0;
+    if constexpr (sizeof(K) == 4) {
+      if (pipe) {
+      /* software-pipelined variant (u32 only; experiment knob) */
+      static int occ_pv = 0, occ_pk = 0;
+      if (!occ_pv) {
+        HIP_CHECK(hipOccupancyMaxActiveBlocksPerMultiprocessor(
+            &occ_pv, (const void *)(k_scatter_lb_pipe<K, true>),
+            NTHREADS_SC, 0));
+        HIP_CHECK(hipOccupancyMaxActiveBlocksPerMultiprocessor(
+            &occ_pk, (const void *)(k_scatter_lb_pipe<K, false>),
+            NTHREADS_SC, 0));
+        hipDeviceProp_t props;
+        HIP_CHECK(hipGetDeviceProperties(&props, c->device));
+        occ_pv = min(2, max(1, occ_pv)) * props.multiProcessorCount;
+        occ_pk = min(2, max(1, occ_pk)) * props.multiProcessorCount;
+      }
+      int pgrid = (int)min(ntiles, (uint64_t)(c->with_vals ? occ_pv
+                                                           : occ_pk));
+      if (c->with_vals) {
+        TIMED(c, K_SCATTER,
+              hipLaunchKernelGGL((k_scatter_lb_pipe<K, true>), dim3(pgrid),
+                                 dim3(NTHREADS_SC), 0, c->stream,
+                                 (const K *)c->keys[src], c->vals[src],
+                                 (K *)c->keys[dst], c->vals[dst], n, ntiles,
+                                 8 * passes[i], c->d_dbase, c->d_lb,
+                                 c->d_fail));
+      } else {
+        TIMED(c, K_SCATTER,
+              hipLaunchKernelGGL((k_scatter_lb_pipe<K, false>), dim3(pgrid),
+                                 dim3(NTHREADS_SC), 0, c->stream,
+                                 (const K *)c->keys[src], nullptr,
+                                 (K *)c->keys[dst], nullptr, n, ntiles,
+                                 8 * passes[i], c->d_dbase, c->d_lb,
+                                 c->d_fail));
+      }
+      uint32_t failp = 0;
+      HIP_CHECK(hipMemcpyAsync(&failp, c->d_fail, 4, hipMemcpyDeviceToHost,
+                               c->stream));
+      HIP_CHECK(hipStreamSynchronize(c->stream));
+      if (failp) {
+        if (radix_pass_t<K>(c, 0, 8 * passes[i], 0xFF, 0)) return 1;
+      } else {
+        uint64_t bpp = 2 * (sizeof(K) + (c->with_vals ? 8 : 0));
+        c->stats.bytes_scatter += n * bpp;
+        c->stats.bytes_total += n * bpp;
+        c->cur = dst;
+      }
+      c->hist_shift_valid = -1;
+      c->stats.sort_passes++;
+      continue;
+      }
+    }
     if (c->with_vals) {
       if (rank_atomic) {
         TIMED(c, K_SCATTER,
