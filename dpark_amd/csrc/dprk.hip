@@ -3215,7 +3215,9 @@ static int sort_pairs_lookback(dprk_ctx *c, const int *passes, int np) {
     static int rank_atomic = -1;
     if (rank_atomic < 0) rank_atomic = getenv("DPRK_RANK_BALLOT") ? 0 : 1;  /* atomic rank default (r02 A/B) */
     static int pipe = -1;
-    if (pipe < 0) pipe = getenv("DPRK_PIPE") ? 1 : 0;
+    if (pipe < 0) pipe = getenv("DPRK_NO_PIPE") ? 0 : 1;  /* default on:
+        r02 same-box A/B: 0.868 -> 0.715 ms/launch at 100M (+16% value),
+        +4.7% at 1B, +1.3% zipf; parity suite green */
     if constexpr (sizeof(K) == 4) {
       if (pipe) {
       /* software-pipelined variant (u32 only; experiment knob) */
