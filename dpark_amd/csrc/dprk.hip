@@ -1101,7 +1101,50 @@ __global__ __launch_bounds__(NTHREADS_SC, 4) void k_scatter_lb_pipe(
     }
     __syncthreads();
 
-    /* phase 3: rank (ds_add_rtn) + stage; then ISSUE next tile's loads */
+    /* phase 3: rank (ds_add_rtn) + stage; then ISSUE next tile's loads.
+     * DPRK_EARLY_WALK (compile variant): the RADIX lookback threads walk
+     * FIRST — their spins overlap the other 12 waves' staging, and the
+     * post-staging barrier doubles as the fail check (one barrier
+     * saved); the cost is waves 0-3's staging starting after the walk. */
+#ifdef DPRK_EARLY_WALK
+#define DPRK_WALK_NOW 1
+#else
+#define DPRK_WALK_NOW 0
+#endif
+    uint32_t tile_nn = 0;
+    if (DPRK_WALK_NOW && threadIdx.x < RADIX) {
+      uint64_t excl_g = 0;
+      uint32_t spins = 0;
+      bool ok = true;
+      int64_t s = (int64_t)t - 1;
+      while (s >= 0) {
+        uint64_t e = __hip_atomic_load(&lb[(uint64_t)s * RADIX + threadIdx.x],
+                                       __ATOMIC_RELAXED,
+                                       __HIP_MEMORY_SCOPE_AGENT);
+        if (!(e >> 62)) {
+          __builtin_amdgcn_s_sleep(2);
+          if (++spins > LB_SPIN_LIMIT) {
+            __hip_atomic_store(d_fail, 1u, __ATOMIC_RELAXED,
+                               __HIP_MEMORY_SCOPE_AGENT);
+            ok = false;
+            break;
+          }
+          continue;
+        }
+        excl_g += e & LB_VAL;
+        if ((e >> 62) >= 2) break;
+        s--;
+      }
+      if (ok) {
+        __hip_atomic_store(&lb[t * RADIX + threadIdx.x],
+                           LB_INC | (excl_g + tile_cnt), __ATOMIC_RELAXED,
+                           __HIP_MEMORY_SCOPE_AGENT);
+        goff[threadIdx.x] = digit_base[threadIdx.x] + (uint32_t)excl_g -
+                            excl[threadIdx.x];
+      } else {
+        sfail = 1;
+      }
+    }
 #pragma unroll
     for (int i = 0; i < EPL; i++) {
       uint32_t idx = wbeg + i * 64 + lane;
@@ -1111,7 +1154,6 @@ __global__ __launch_bounds__(NTHREADS_SC, 4) void k_scatter_lb_pipe(
         if (WITH_VALS) lv[pos] = rvc[i];
       }
     }
-    uint32_t tile_nn = 0;
     if (tn < ntiles) {
       const uint64_t startn = tn * (uint64_t)TILE;
       tile_nn = (uint32_t)(min(startn + TILE, n) - startn);
@@ -1126,7 +1168,7 @@ __global__ __launch_bounds__(NTHREADS_SC, 4) void k_scatter_lb_pipe(
     }
 
     /* lookback (thread = digit) overlaps the in-flight prefetch */
-    if (threadIdx.x < RADIX) {
+    if (!DPRK_WALK_NOW && threadIdx.x < RADIX) {
       uint64_t excl_g = 0;
       uint32_t spins = 0;
       bool ok = true;
