@@ -1240,6 +1240,213 @@ __global__ __launch_bounds__(NTHREADS_SC, 4) void k_scatter_lb_pipe(
   }
 }
 
+/* Two-deep key prefetch variant (DPRK_PIPE2): keys of tile t+2 issue at
+ * tile t (a FULL tile of latency cover before their tail-hist use —
+ * the one-deep pipe only covers them with the short write phase), vals
+ * stay one-deep (consumed at staging, already a full tile of cover).
+ * 3 key buffers (+8 VGPRs, alloc 120 <= 128: occupancy preserved at 4
+ * waves/SIMD). */
+template <typename K, bool WITH_VALS>
+__global__ __launch_bounds__(NTHREADS_SC, 4) void k_scatter_lb_pipe2(
+    const K *__restrict__ keys_in, const double *__restrict__ vals_in,
+    K *__restrict__ keys_out, double *__restrict__ vals_out,
+    uint64_t n, uint64_t ntiles, int shift,
+    const uint32_t *__restrict__ digit_base,
+    unsigned long long *__restrict__ lb,
+    unsigned int *__restrict__ d_fail) {
+  constexpr int NW = NTHREADS_SC / 64;
+  constexpr int WBLK = TILE / NW;
+  constexpr int EPL = WBLK / 64;
+  __shared__ K lk[TILE];
+  __shared__ double lv[WITH_VALS ? TILE : 1];
+  __shared__ uint32_t excl[RADIX];
+  __shared__ uint32_t goff[RADIX];
+  __shared__ uint32_t h[2][NW][RADIX];
+  __shared__ uint32_t wtot[NW > 4 ? NW : 4];
+  __shared__ uint32_t sfail;
+
+  const int wave = threadIdx.x / 64, lane = threadIdx.x % 64;
+  const uint32_t mask = 0xFF;
+  const uint32_t wbeg = wave * WBLK;
+
+  uint64_t t0 = blockIdx.x;
+  if (t0 >= ntiles) return;
+  for (int j = 0; j < RADIX / 64; j++) {
+    h[0][wave][lane * (RADIX / 64) + j] = 0;
+    h[1][wave][lane * (RADIX / 64) + j] = 0;
+  }
+  if (threadIdx.x == 0) sfail = 0;
+
+  K ka[EPL], kb[EPL], kc3[EPL];
+  double va[EPL], vb[EPL];
+  uint32_t da[EPL], db[EPL];
+
+  auto loadk = [&](K (&dst)[EPL], uint64_t t) {
+    if (t >= ntiles) return;
+    const uint64_t start = t * (uint64_t)TILE;
+    const uint32_t tile_n = (uint32_t)(min(start + TILE, n) - start);
+#pragma unroll
+    for (int i = 0; i < EPL; i++) {
+      uint32_t idx = wbeg + i * 64 + lane;
+      if (idx < tile_n) dst[i] = keys_in[start + idx];
+    }
+  };
+  auto loadv = [&](double (&dst)[EPL], uint64_t t) {
+    if (!WITH_VALS || t >= ntiles) return;
+    const uint64_t start = t * (uint64_t)TILE;
+    const uint32_t tile_n = (uint32_t)(min(start + TILE, n) - start);
+#pragma unroll
+    for (int i = 0; i < EPL; i++) {
+      uint32_t idx = wbeg + i * 64 + lane;
+      if (idx < tile_n) dst[i] = vals_in[start + idx];
+    }
+  };
+  auto histd = [&](K (&keys)[EPL], uint32_t (&dig)[EPL],
+                   uint32_t (*hh)[RADIX], uint64_t t) {
+    if (t >= ntiles) return;
+    const uint64_t start = t * (uint64_t)TILE;
+    const uint32_t tile_n = (uint32_t)(min(start + TILE, n) - start);
+#pragma unroll
+    for (int i = 0; i < EPL; i++) {
+      uint32_t idx = wbeg + i * 64 + lane;
+      if (idx < tile_n) {
+        dig[i] = d_digit(keys[i], 0, shift, mask);
+        atomicAdd(&hh[wave][dig[i]], 1u);
+      }
+    }
+  };
+
+  /* prolog: keys t0+t1 + vals t0 in flight; hist(t0) */
+  __syncthreads();          /* zeroed hists visible */
+  loadk(ka, t0);
+  loadk(kb, t0 + gridDim.x);
+  loadv(va, t0);
+  histd(ka, da, h[0], t0);
+
+  /* one stage; (kcur, vcur, dcur, hcur) hold tile t; kload receives the
+   * keys of t+2, vload the vals of t+1; ktail feeds t+1's tail hist. */
+  auto stage = [&](K (&kcur)[EPL], double (&vcur)[EPL],
+                   uint32_t (&dcur)[EPL], K (&ktail)[EPL],
+                   uint32_t (&dnext)[EPL], K (&kload)[EPL],
+                   double (&vload)[EPL], uint32_t (*hcur)[RADIX],
+                   uint32_t (*hnext)[RADIX], uint64_t t) -> bool {
+    if (threadIdx.x == 0)
+      sfail = __hip_atomic_load(d_fail, __ATOMIC_RELAXED,
+                                __HIP_MEMORY_SCOPE_AGENT);
+    __syncthreads();
+    if (sfail) return false;
+    const uint64_t start = t * (uint64_t)TILE;
+    const uint32_t tile_n = (uint32_t)(min(start + TILE, n) - start);
+
+    uint32_t tile_cnt = 0;
+    if (threadIdx.x < RADIX) {
+      for (int w = 0; w < NW; w++) tile_cnt += hcur[w][threadIdx.x];
+      __hip_atomic_store(&lb[t * RADIX + threadIdx.x], LB_AGG | tile_cnt,
+                         __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+      uint32_t inc = d_wave_incl_scan(tile_cnt, lane);
+      if (lane == 63) wtot[wave] = inc;
+      excl[threadIdx.x] = inc - tile_cnt;
+    }
+    __syncthreads();
+    if (threadIdx.x < RADIX) {
+      if (wave > 0) {
+        uint32_t add = 0;
+        for (int w = 0; w < wave; w++) add += wtot[w];
+        excl[threadIdx.x] += add;
+      }
+      uint32_t b = excl[threadIdx.x];
+      for (int w = 0; w < NW; w++) {
+        uint32_t cnt = hcur[w][threadIdx.x];
+        hcur[w][threadIdx.x] = b;
+        b += cnt;
+      }
+    }
+    __syncthreads();
+
+    /* rank + stage; issue t+2 keys and t+1 vals */
+#pragma unroll
+    for (int i = 0; i < EPL; i++) {
+      uint32_t idx = wbeg + i * 64 + lane;
+      if (idx < tile_n) {
+        uint32_t pos = atomicAdd(&hcur[wave][dcur[i]], 1u);
+        lk[pos] = kcur[i];
+        if (WITH_VALS) lv[pos] = vcur[i];
+      }
+    }
+    loadk(kload, t + 2 * gridDim.x);
+    loadv(vload, t + gridDim.x);
+
+    /* lookback walk */
+    if (threadIdx.x < RADIX) {
+      uint64_t excl_g = 0;
+      uint32_t spins = 0;
+      bool ok = true;
+      int64_t s = (int64_t)t - 1;
+      while (s >= 0) {
+        uint64_t e = __hip_atomic_load(&lb[(uint64_t)s * RADIX + threadIdx.x],
+                                       __ATOMIC_RELAXED,
+                                       __HIP_MEMORY_SCOPE_AGENT);
+        if (!(e >> 62)) {
+          __builtin_amdgcn_s_sleep(2);
+          if (++spins > LB_SPIN_LIMIT) {
+            __hip_atomic_store(d_fail, 1u, __ATOMIC_RELAXED,
+                               __HIP_MEMORY_SCOPE_AGENT);
+            ok = false;
+            break;
+          }
+          continue;
+        }
+        excl_g += e & LB_VAL;
+        if ((e >> 62) >= 2) break;
+        s--;
+      }
+      if (ok) {
+        __hip_atomic_store(&lb[t * RADIX + threadIdx.x],
+                           LB_INC | (excl_g + tile_cnt), __ATOMIC_RELAXED,
+                           __HIP_MEMORY_SCOPE_AGENT);
+        goff[threadIdx.x] = digit_base[threadIdx.x] + (uint32_t)excl_g -
+                            excl[threadIdx.x];
+      } else {
+        sfail = 1;
+      }
+    }
+    __syncthreads();
+    if (sfail) return false;
+
+    for (uint32_t idx = threadIdx.x; idx < tile_n; idx += NTHREADS_SC) {
+      K key = lk[idx];
+      uint32_t dig = d_digit(key, 0, shift, mask);
+      keys_out[goff[dig] + idx] = key;
+      if (WITH_VALS) vals_out[goff[dig] + idx] = lv[idx];
+    }
+
+    /* tail: zero the consumed hist; hist + digits of t+1 (keys were
+     * issued a FULL tile ago) */
+    for (int j = 0; j < RADIX / 64; j++)
+      hcur[wave][lane * (RADIX / 64) + j] = 0;
+    histd(ktail, dnext, hnext, t + gridDim.x);
+    return true;
+  };
+
+  /* 6-stage rotation: keys period 3 (a,b,c), vals/digits/hists period 2 */
+  uint64_t t = t0;
+#define PSTEP(KC, VC, DC, KT, DN, KL, VL, HC, HN)                         \
+  do {                                                                     \
+    if (!stage(KC, VC, DC, KT, DN, KL, VL, HC, HN, t)) return;             \
+    t += gridDim.x;                                                        \
+    if (t >= ntiles) return;                                               \
+  } while (0)
+  for (;;) {
+    PSTEP(ka, va, da, kb, db, kc3, vb, h[0], h[1]);
+    PSTEP(kb, vb, db, kc3, da, ka, va, h[1], h[0]);
+    PSTEP(kc3, va, da, ka, db, kb, vb, h[0], h[1]);
+    PSTEP(ka, vb, db, kb, da, kc3, va, h[1], h[0]);
+    PSTEP(kb, va, da, kc3, db, ka, vb, h[0], h[1]);
+    PSTEP(kc3, vb, db, ka, da, kb, va, h[1], h[0]);
+  }
+#undef PSTEP
+}
+
 template <typename K, bool WITH_VALS, bool RANK_ATOMIC = false>
 __global__ __launch_bounds__(NTHREADS_SC, 2) void k_scatter(
     const K *__restrict__ keys_in, const double *__restrict__ vals_in,
@@ -3260,16 +3467,21 @@ static int sort_pairs_lookback(dprk_ctx *c, const int *passes, int np) {
     if (pipe < 0) pipe = getenv("DPRK_NO_PIPE") ? 0 : 1;  /* default on:
         r02 same-box A/B: 0.868 -> 0.715 ms/launch at 100M (+16% value),
         +4.7% at 1B, +1.3% zipf; parity suite green */
+    static int pipe2 = -1;
+    if (pipe2 < 0) pipe2 = getenv("DPRK_PIPE2") ? 1 : 0;
     if constexpr (sizeof(K) == 4) {
       if (pipe) {
-      /* software-pipelined variant (u32 only; experiment knob) */
+      /* software-pipelined variant (u32 only); DPRK_PIPE2 selects the
+       * two-deep key prefetch form */
       static int occ_pv = 0, occ_pk = 0;
       if (!occ_pv) {
         HIP_CHECK(hipOccupancyMaxActiveBlocksPerMultiprocessor(
-            &occ_pv, (const void *)(k_scatter_lb_pipe<K, true>),
+            &occ_pv, pipe2 ? (const void *)(k_scatter_lb_pipe2<K, true>)
+                           : (const void *)(k_scatter_lb_pipe<K, true>),
             NTHREADS_SC, 0));
         HIP_CHECK(hipOccupancyMaxActiveBlocksPerMultiprocessor(
-            &occ_pk, (const void *)(k_scatter_lb_pipe<K, false>),
+            &occ_pk, pipe2 ? (const void *)(k_scatter_lb_pipe2<K, false>)
+                           : (const void *)(k_scatter_lb_pipe<K, false>),
             NTHREADS_SC, 0));
         hipDeviceProp_t props;
         HIP_CHECK(hipGetDeviceProperties(&props, c->device));
@@ -3279,21 +3491,46 @@ static int sort_pairs_lookback(dprk_ctx *c, const int *passes, int np) {
       int pgrid = (int)min(ntiles, (uint64_t)(c->with_vals ? occ_pv
                                                            : occ_pk));
       if (c->with_vals) {
-        TIMED(c, K_SCATTER,
-              hipLaunchKernelGGL((k_scatter_lb_pipe<K, true>), dim3(pgrid),
-                                 dim3(NTHREADS_SC), 0, c->stream,
-                                 (const K *)c->keys[src], c->vals[src],
-                                 (K *)c->keys[dst], c->vals[dst], n, ntiles,
-                                 8 * passes[i], c->d_dbase, c->d_lb,
-                                 c->d_fail));
+        if (pipe2) {
+          TIMED(c, K_SCATTER,
+                hipLaunchKernelGGL((k_scatter_lb_pipe2<K, true>),
+                                   dim3(pgrid),
+                                   dim3(NTHREADS_SC), 0, c->stream,
+                                   (const K *)c->keys[src], c->vals[src],
+                                   (K *)c->keys[dst], c->vals[dst], n,
+                                   ntiles,
+                                   8 * passes[i], c->d_dbase, c->d_lb,
+                                   c->d_fail));
+        } else {
+          TIMED(c, K_SCATTER,
+                hipLaunchKernelGGL((k_scatter_lb_pipe<K, true>), dim3(pgrid),
+                                   dim3(NTHREADS_SC), 0, c->stream,
+                                   (const K *)c->keys[src], c->vals[src],
+                                   (K *)c->keys[dst], c->vals[dst], n,
+                                   ntiles,
+                                   8 * passes[i], c->d_dbase, c->d_lb,
+                                   c->d_fail));
+        }
       } else {
-        TIMED(c, K_SCATTER,
-              hipLaunchKernelGGL((k_scatter_lb_pipe<K, false>), dim3(pgrid),
-                                 dim3(NTHREADS_SC), 0, c->stream,
-                                 (const K *)c->keys[src], nullptr,
-                                 (K *)c->keys[dst], nullptr, n, ntiles,
-                                 8 * passes[i], c->d_dbase, c->d_lb,
-                                 c->d_fail));
+        if (pipe2) {
+          TIMED(c, K_SCATTER,
+                hipLaunchKernelGGL((k_scatter_lb_pipe2<K, false>),
+                                   dim3(pgrid),
+                                   dim3(NTHREADS_SC), 0, c->stream,
+                                   (const K *)c->keys[src], nullptr,
+                                   (K *)c->keys[dst], nullptr, n, ntiles,
+                                   8 * passes[i], c->d_dbase, c->d_lb,
+                                   c->d_fail));
+        } else {
+          TIMED(c, K_SCATTER,
+                hipLaunchKernelGGL((k_scatter_lb_pipe<K, false>),
+                                   dim3(pgrid),
+                                   dim3(NTHREADS_SC), 0, c->stream,
+                                   (const K *)c->keys[src], nullptr,
+                                   (K *)c->keys[dst], nullptr, n, ntiles,
+                                   8 * passes[i], c->d_dbase, c->d_lb,
+                                   c->d_fail));
+        }
       }
       uint32_t failp = 0;
       HIP_CHECK(hipMemcpyAsync(&failp, c->d_fail, 4, hipMemcpyDeviceToHost,
