@@ -549,10 +549,17 @@ __global__ __launch_bounds__(NTHREADS) void k_gen_alias(
     uint64_t e = table[slot];
     uint64_t u2 = d_splitmix64(x ^ 0x14057B7EF767814Full);
     uint64_t k = ((u2 >> 32) < (e >> 32)) ? slot : (e & 0xFFFFFFFFull);
-    keys[i] = KT<K>::from_raw(k);
+    /* NONTEMPORAL stores: the generated stream (12 GB at config 5) is
+     * re-read from HBM regardless, but streamed normally it EVICTS the
+     * 128 MiB alias table from L3 — the table's random 8-B loads then
+     * miss to HBM (the r02-measured ~20 ms/step generator cost).  Same
+     * values, different store hint: the numpy mirror stays
+     * byte-identical. */
+    __builtin_nontemporal_store(KT<K>::from_raw(k), &keys[i]);
     if (vals) {
       uint64_t vb = d_splitmix64(x ^ 0xDEADBEEFCAFEBABEull);
-      vals[i] = (double)(vb >> 11) * (1.0 / 9007199254740992.0); /* 2^-53 */
+      __builtin_nontemporal_store(
+          (double)(vb >> 11) * (1.0 / 9007199254740992.0), &vals[i]);
     }
   }
 }
