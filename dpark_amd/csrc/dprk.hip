@@ -540,26 +540,49 @@ template <typename K>
 __global__ __launch_bounds__(NTHREADS) void k_gen_alias(
     K *__restrict__ keys, double *__restrict__ vals, uint64_t n,
     uint64_t seed_hi, const uint64_t *__restrict__ table, uint64_t universe) {
+  /* 4-way batched iteration: four independent table loads in flight per
+   * thread before any resolve (the random 8-B loads over the 128 MiB
+   * table are the generator's cost — one dependent load per iteration
+   * left most of the latency unhidden).  Stores are nontemporal (the
+   * 12 GB output stream cannot stay cached anyway; measured neutral on
+   * its own).  Values identical to the numpy mirror either way. */
   uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
-  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += stride) {
+  uint64_t i0 = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  for (; i0 + 3 * stride < n; i0 += 4 * stride) {
+    uint64_t x[4], u1[4], slot[4], e[4];
+#pragma unroll
+    for (int j = 0; j < 4; j++) {
+      x[j] = seed_hi + i0 + j * stride;
+      u1[j] = d_splitmix64(x[j] ^ 0x5851F42D4C957F2Dull);
+      slot[j] = __umul64hi(u1[j], universe);
+      e[j] = table[slot[j]];
+    }
+#pragma unroll
+    for (int j = 0; j < 4; j++) {
+      uint64_t u2 = d_splitmix64(x[j] ^ 0x14057B7EF767814Full);
+      uint64_t k = ((u2 >> 32) < (e[j] >> 32)) ? slot[j]
+                                               : (e[j] & 0xFFFFFFFFull);
+      __builtin_nontemporal_store(KT<K>::from_raw(k),
+                                  &keys[i0 + j * stride]);
+      if (vals) {
+        uint64_t vb = d_splitmix64(x[j] ^ 0xDEADBEEFCAFEBABEull);
+        __builtin_nontemporal_store(
+            (double)(vb >> 11) * (1.0 / 9007199254740992.0),
+            &vals[i0 + j * stride]);
+      }
+    }
+  }
+  for (uint64_t i = i0; i < n; i += stride) {
     uint64_t x = seed_hi + i;
     uint64_t u1 = d_splitmix64(x ^ 0x5851F42D4C957F2Dull);
     uint64_t slot = __umul64hi(u1, universe);
     uint64_t e = table[slot];
     uint64_t u2 = d_splitmix64(x ^ 0x14057B7EF767814Full);
     uint64_t k = ((u2 >> 32) < (e >> 32)) ? slot : (e & 0xFFFFFFFFull);
-    /* NONTEMPORAL stores: the generated stream (12 GB at config 5) is
-     * re-read from HBM regardless, but streamed normally it EVICTS the
-     * 128 MiB alias table from L3 — the table's random 8-B loads then
-     * miss to HBM (the r02-measured ~20 ms/step generator cost).  Same
-     * values, different store hint: the numpy mirror stays
-     * byte-identical. */
-    __builtin_nontemporal_store(KT<K>::from_raw(k), &keys[i]);
+    keys[i] = KT<K>::from_raw(k);
     if (vals) {
       uint64_t vb = d_splitmix64(x ^ 0xDEADBEEFCAFEBABEull);
-      __builtin_nontemporal_store(
-          (double)(vb >> 11) * (1.0 / 9007199254740992.0), &vals[i]);
+      vals[i] = (double)(vb >> 11) * (1.0 / 9007199254740992.0);
     }
   }
 }
@@ -3554,6 +3577,43 @@ static int sort_pairs_lookback(dprk_ctx *c, const int *passes, int np) {
       c->hist_shift_valid = -1;
       c->stats.sort_passes++;
       continue;
+      }
+    } else if (!c->with_vals) {
+      /* int64 KEYS-ONLY also fits the pipelined kernel (64 KiB staging
+       * + 32 KiB double hist ~ 98 KiB LDS); with-values int64 does not
+       * (160+ KiB) and takes the non-pipelined path below */
+      if (pipe) {
+        static int occ_p8 = 0;
+        if (!occ_p8) {
+          HIP_CHECK(hipOccupancyMaxActiveBlocksPerMultiprocessor(
+              &occ_p8, (const void *)(k_scatter_lb_pipe<K, false>),
+              NTHREADS_SC, 0));
+          hipDeviceProp_t props;
+          HIP_CHECK(hipGetDeviceProperties(&props, c->device));
+          occ_p8 = min(2, max(1, occ_p8)) * props.multiProcessorCount;
+        }
+        int pgrid = (int)min(ntiles, (uint64_t)occ_p8);
+        TIMED(c, K_SCATTER,
+              hipLaunchKernelGGL((k_scatter_lb_pipe<K, false>), dim3(pgrid),
+                                 dim3(NTHREADS_SC), 0, c->stream,
+                                 (const K *)c->keys[src], nullptr,
+                                 (K *)c->keys[dst], nullptr, n, ntiles,
+                                 8 * passes[i], c->d_dbase, c->d_lb,
+                                 c->d_fail));
+        uint32_t failp = 0;
+        HIP_CHECK(hipMemcpyAsync(&failp, c->d_fail, 4,
+                                 hipMemcpyDeviceToHost, c->stream));
+        HIP_CHECK(hipStreamSynchronize(c->stream));
+        if (failp) {
+          if (radix_pass_t<K>(c, 0, 8 * passes[i], 0xFF, 0)) return 1;
+        } else {
+          c->stats.bytes_scatter += n * 2 * sizeof(K);
+          c->stats.bytes_total += n * 2 * sizeof(K);
+          c->cur = dst;
+        }
+        c->hist_shift_valid = -1;
+        c->stats.sort_passes++;
+        continue;
       }
     }
     if (c->with_vals) {
