@@ -2445,19 +2445,19 @@ static int tile_grid(uint64_t ntiles) {
 }
 
 /* lookback L3-residency gate in bytes (granule array must stay L3-hot;
- * DPRK_LB_GATE_MB tunes).  Defaults measured r02 on the 1B shapes:
- * with-values pairs win up to 300 MB (28.4 -> 29.5 G pairs/s at 1B),
- * keys-only sort loses past 200 (16.0 -> 15.6 G keys/s) — the gate is a
- * pure function of (n, with_vals), so path choice stays
- * input-deterministic. */
+ * DPRK_LB_GATE_MB tunes).  Default 300 MB, measured r02 on the 1B
+ * shapes WITH the pipelined scatters: pairs 28.4 -> 29.5 G pairs/s,
+ * keys-only sort 16.0 -> 19.6 G keys/s (the earlier 200 MB keys-only
+ * preference belonged to the pre-pipeline kernel).  A pure function of
+ * n => input-deterministic path choice. */
 static uint64_t lb_gate_bytes(int with_vals) {
+  (void)with_vals;
   static long gate_mb = -1;
   if (gate_mb < 0) {
     const char *e = getenv("DPRK_LB_GATE_MB");
-    gate_mb = e ? atol(e) : 0;
+    gate_mb = e ? atol(e) : 300;
   }
-  if (gate_mb > 0) return (uint64_t)gate_mb << 20;
-  return (uint64_t)(with_vals ? 300 : 200) << 20;
+  return (uint64_t)gate_mb << 20;
 }
 
 extern "C" int dprk_init(dprk_ctx **out, int device_id) {
