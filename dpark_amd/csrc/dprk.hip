@@ -2381,10 +2381,12 @@ __global__ __launch_bounds__(NTHREADS) void k_join_count(
 }
 
 /* emit: per-key cross products (dpark join semantics — SURVEY.md §3.3).
- * One 16-lane sub-wave group per A-run (avg products are small — config
- * 4: ~14 — so a full wave per run left 78% of lanes idle; measured 37 ms
- * for 9.3e8 outputs).  Consecutive groups emit consecutive runs, whose
- * output ranges are adjacent => coalesced-ish stores. */
+ * One GLANES-lane sub-wave group per A-run (products are small, so a
+ * full wave per run left most lanes idle at low multiplicity — measured
+ * r01; GLANES tunable: wider groups write fuller lines per iteration).
+ * Consecutive groups emit consecutive runs, whose output ranges are
+ * adjacent => coalesced-ish stores. */
+template <int GLANES>
 __global__ __launch_bounds__(NTHREADS) void k_join_emit(
     const int64_t *__restrict__ uka, const uint64_t *__restrict__ starta,
     uint64_t ga,
@@ -2393,9 +2395,10 @@ __global__ __launch_bounds__(NTHREADS) void k_join_emit(
     const uint64_t *__restrict__ bidx, const double *__restrict__ va,
     const double *__restrict__ vb, int64_t *__restrict__ out_k,
     double *__restrict__ out_a, double *__restrict__ out_b) {
-  const uint32_t lane16 = threadIdx.x & 15;
-  const uint64_t ngroups = (uint64_t)gridDim.x * (NTHREADS / 16);
-  const uint64_t gid = (uint64_t)blockIdx.x * (NTHREADS / 16) + threadIdx.x / 16;
+  const uint32_t laneg = threadIdx.x & (GLANES - 1);
+  const uint64_t ngroups = (uint64_t)gridDim.x * (NTHREADS / GLANES);
+  const uint64_t gid =
+      (uint64_t)blockIdx.x * (NTHREADS / GLANES) + threadIdx.x / GLANES;
   for (uint64_t r = gid; r < ga; r += ngroups) {
     uint64_t p = prod[r];
     if (p == 0) continue;
@@ -2405,7 +2408,7 @@ __global__ __launch_bounds__(NTHREADS) void k_join_emit(
     uint64_t sb = startb[b];
     uint32_t cb = (uint32_t)cntb[b];
     uint64_t base = prod_off[r];
-    for (uint64_t j = lane16; j < p; j += 16) {
+    for (uint64_t j = laneg; j < p; j += GLANES) {
       out_k[base + j] = k;
       out_a[base + j] = va[sa + j / cb];
       out_b[base + j] = vb[sb + j % cb];
@@ -4104,13 +4107,42 @@ extern "C" int dprk_join(dprk_ctx *a, dprk_ctx *b, uint64_t *n_out) {
     HIP_CHECK(hipMalloc(&a->join_b, cap * 8));
     a->join_cap = cap;
   }
-  TIMED(a, K_JOIN,
-        hipLaunchKernelGGL(k_join_emit, dim3(2048), dim3(NTHREADS), 0,
-                           a->stream, a->grp_keys, a->grp_starts,
-                           ga, b->grp_starts, b->grp_counts,
-                           a->run_boff, a->run_prod, d_bidx,
-                           a->vals[a->cur], b->vals[b->cur], a->join_k,
-                           a->join_a, a->join_b));
+  {
+    static int glanes = -1;
+    if (glanes < 0) {
+      const char *e = getenv("DPRK_JOIN_GLANES");
+      glanes = e ? atoi(e) : 16;
+      if (glanes != 16 && glanes != 32 && glanes != 64) glanes = 16;
+    }
+    if (glanes == 64) {
+      TIMED(a, K_JOIN,
+            hipLaunchKernelGGL(k_join_emit<64>, dim3(2048), dim3(NTHREADS),
+                               0,
+                               a->stream, a->grp_keys, a->grp_starts,
+                               ga, b->grp_starts, b->grp_counts,
+                               a->run_boff, a->run_prod, d_bidx,
+                               a->vals[a->cur], b->vals[b->cur], a->join_k,
+                               a->join_a, a->join_b));
+    } else if (glanes == 32) {
+      TIMED(a, K_JOIN,
+            hipLaunchKernelGGL(k_join_emit<32>, dim3(2048), dim3(NTHREADS),
+                               0,
+                               a->stream, a->grp_keys, a->grp_starts,
+                               ga, b->grp_starts, b->grp_counts,
+                               a->run_boff, a->run_prod, d_bidx,
+                               a->vals[a->cur], b->vals[b->cur], a->join_k,
+                               a->join_a, a->join_b));
+    } else {
+      TIMED(a, K_JOIN,
+            hipLaunchKernelGGL(k_join_emit<16>, dim3(2048), dim3(NTHREADS),
+                               0,
+                               a->stream, a->grp_keys, a->grp_starts,
+                               ga, b->grp_starts, b->grp_counts,
+                               a->run_boff, a->run_prod, d_bidx,
+                               a->vals[a->cur], b->vals[b->cur], a->join_k,
+                               a->join_a, a->join_b));
+    }
+  }
   a->stats.bytes_total += total * 24 + ga * 32 + gb * 16;
   int rc = drain_timing(a);
   (void)d_bidx;
