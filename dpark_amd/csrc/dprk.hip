@@ -2408,10 +2408,27 @@ __global__ __launch_bounds__(NTHREADS) void k_join_emit(
     uint64_t sb = startb[b];
     uint32_t cb = (uint32_t)cntb[b];
     uint64_t base = prod_off[r];
-    for (uint64_t j = laneg; j < p; j += GLANES) {
-      out_k[base + j] = k;
-      out_a[base + j] = va[sa + j / cb];
-      out_b[base + j] = vb[sb + j % cb];
+    /* index math in u32 (u64 div is a ~50-100 cycle libcall, paid twice
+     * per output element); cb==1 needs no division at all */
+    if (cb == 1) {
+      double vbv = vb[sb];
+      for (uint64_t j = laneg; j < p; j += GLANES) {
+        out_k[base + j] = k;
+        out_a[base + j] = va[sa + j];
+        out_b[base + j] = vbv;
+      }
+    } else if (p <= 0xFFFFFFFFull) {
+      for (uint32_t j = laneg; j < (uint32_t)p; j += GLANES) {
+        out_k[base + j] = k;
+        out_a[base + j] = va[sa + j / cb];
+        out_b[base + j] = vb[sb + j % cb];
+      }
+    } else {
+      for (uint64_t j = laneg; j < p; j += GLANES) {
+        out_k[base + j] = k;
+        out_a[base + j] = va[sa + j / cb];
+        out_b[base + j] = vb[sb + j % cb];
+      }
     }
   }
 }
