@@ -205,11 +205,8 @@ class Engine:
                "dprk_big_reduce")
         return n.value
 
-    def big_get(self, n=None, with_vals=True):
-        if n is None:
-            n = self.big_reduce_count if hasattr(self, "big_reduce_count") \
-                else None
-        assert n is not None, "pass n from big_reduce()"
+    def big_get(self, n, with_vals=True):
+        """n = the count returned by big_reduce()."""
         keys = np.empty(n, dtype=np.int64)
         vals = np.empty(n, dtype=np.float64) if with_vals else None
         nout = ctypes.c_uint64()

@@ -706,6 +706,30 @@ def test_big_put_chunks_and_errors():
         with pytest.raises(EngineError, match="overflow"):
             e.big_add_chunk()
         e.big_free()
+        # the documented mitigation COMPOSES from existing ops: reduce
+        # each chunk in-core BEFORE adding — a hot key then contributes
+        # one pair per chunk and the same data succeeds
+        e.big_begin(300_000, nbuckets=16)
+        total = {}
+        for i in range(3):
+            keys = np.where(rng.random(100_000) < 0.9, 7,
+                            rng.integers(0, 1 << 10, 100_000)
+                            ).astype(np.int64)
+            vals = rng.random(100_000)
+            for kk, vv in zip(keys.tolist(), vals.tolist()):
+                total[kk] = total.get(kk, 0.0) + vv
+            e.put(keys, vals)
+            e.reduce_by_key_sum()        # chunk pre-reduction
+            e.big_add_chunk()
+        nu = e.big_reduce()
+        k, v = e.big_get(n=nu)
+        assert nu == len(total)
+        order = np.argsort(k, kind="stable")
+        wk = np.array(sorted(total), dtype=np.int64)
+        assert np.array_equal(k[order], wk)
+        np.testing.assert_allclose(v[order], [total[int(x)] for x in wk],
+                                   rtol=RTOL)
+        e.big_free()
     finally:
         e.close()
 
