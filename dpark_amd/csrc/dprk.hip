@@ -52,8 +52,10 @@
 #ifndef NTHREADS_SC
 #define NTHREADS_SC 1024   /* scatter workgroup: 16 waves (tunable)      */
 #endif
+#ifndef RTILE
 #define RTILE 2048         /* elements per reduce/head workgroup (4 WG/CU
                               occupancy: the reduce is latency-bound)    */
+#endif
 #define EPT_R (RTILE / NTHREADS) /* 8 consecutive elements per thread    */
 #define ELEMS_PER_THREAD (TILE / NTHREADS) /* 16 */
 #define SCAN_CHUNK 16384   /* generic u64 scan: elements per workgroup   */
