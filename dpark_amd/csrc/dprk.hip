@@ -287,11 +287,6 @@ static int drain_timing(dprk_ctx *c) {
 template <typename K> struct KT;
 template <> struct KT<int64_t> {
   using V2 = longlong2;                       /* 2 keys = 16 B          */
-  using V16 = longlong2;
-  static __device__ __forceinline__ int64_t get16(const longlong2 &v,
-                                                  int j) {
-    return j == 0 ? (int64_t)v.x : (int64_t)v.y;
-  }
   static constexpr int KSZ = 8;
   static __device__ __forceinline__ uint64_t bias(int64_t k) {
     return (uint64_t)k ^ SIGNFLIP;            /* signed -> unsigned order */
@@ -307,11 +302,6 @@ template <> struct KT<int64_t> {
  * an identical specialization keeps vector-element digit calls typed */
 template <> struct KT<long long> {
   using V2 = longlong2;
-  using V16 = longlong2;                      /* 2 keys = 16 B          */
-  static __device__ __forceinline__ long long get16(const longlong2 &v,
-                                                    int j) {
-    return j == 0 ? v.x : v.y;
-  }
   static constexpr int KSZ = 8;
   static __device__ __forceinline__ uint64_t bias(long long k) {
     return (uint64_t)k ^ SIGNFLIP;
@@ -325,10 +315,6 @@ template <> struct KT<long long> {
 };
 template <> struct KT<uint32_t> {
   using V2 = uint2;                           /* 2 keys = 8 B           */
-  using V16 = uint4;                          /* 4 keys = 16 B          */
-  static __device__ __forceinline__ uint32_t get16(const uint4 &v, int j) {
-    return j == 0 ? v.x : j == 1 ? v.y : j == 2 ? v.z : v.w;
-  }
   static constexpr int KSZ = 4;
   static __device__ __forceinline__ uint64_t bias(uint32_t k) {
     return (uint64_t)k;                       /* non-negative: no flip  */
@@ -1086,59 +1072,19 @@ __global__ __launch_bounds__(NTHREADS_SC, 4) void k_scatter_lb_pipe(
   double rv0[EPL], rv1[EPL];
   uint32_t rd0[EPL], rd1[EPL];
 
-  /* BLOCKED element ownership: lane owns EPL CONSECUTIVE elements at
-   * wbeg + lane*EPL — 16-B vector loads stay fully coalesced across the
-   * wave (narrow per-lane streams run at 0.54-0.70x the 16-B rate,
-   * MI355X_MICROARCH.md §HBM).  Legal here because ranking is
-   * ds_add_rtn (hardware atomic order), which never promised lane-order
-   * placement — the sort contracts (key order, per-key multisets,
-   * run-to-run determinism) are index-mapping independent. */
-  constexpr int EV = 16 / sizeof(K);        /* keys per 16-B load   */
-  constexpr int NVK = EPL / EV;
-  auto load_blocked = [&](K (&rkd)[EPL], double (&rvd)[EPL], uint64_t t) {
-    const uint64_t start = t * (uint64_t)TILE;
-    const uint32_t tile_n = (uint32_t)(min(start + TILE, n) - start);
-    const uint32_t base = wbeg + lane * EPL;
-    if (base + EPL <= tile_n) {
-#pragma unroll
-      for (int v = 0; v < NVK; v++) {
-        typename KT<K>::V16 kv = *reinterpret_cast<const typename KT<
-            K>::V16 *>(&keys_in[start + base + v * EV]);
-#pragma unroll
-        for (int j = 0; j < EV; j++) rkd[v * EV + j] = KT<K>::get16(kv, j);
-      }
-      if (WITH_VALS) {
-#pragma unroll
-        for (int v = 0; v < EPL / 2; v++) {
-          double2 vv = *reinterpret_cast<const double2 *>(
-              &vals_in[start + base + v * 2]);
-          rvd[v * 2] = vv.x;
-          rvd[v * 2 + 1] = vv.y;
-        }
-      }
-    } else {
-#pragma unroll
-      for (int i = 0; i < EPL; i++) {
-        uint32_t idx = base + i;
-        if (idx < tile_n) {
-          rkd[i] = keys_in[start + idx];
-          if (WITH_VALS) rvd[i] = vals_in[start + idx];
-        }
-      }
-    }
-  };
-
   /* prolog: load + hist tile t0 into buffer 0 (barrier below publishes
    * the zeroed hists before the atomics) */
   __syncthreads();
   {
     const uint64_t start = t0 * (uint64_t)TILE;
-    const uint32_t tile_n = (uint32_t)(min(start + TILE, n) - start);
-    load_blocked(rk0, rv0, t0);
+    const uint64_t end = min(start + TILE, n);
+    const uint32_t tile_n = (uint32_t)(end - start);
 #pragma unroll
     for (int i = 0; i < EPL; i++) {
-      uint32_t idx = wbeg + lane * EPL + i;
+      uint32_t idx = wbeg + i * 64 + lane;
       if (idx < tile_n) {
+        rk0[i] = keys_in[start + idx];
+        if (WITH_VALS) rv0[i] = vals_in[start + idx];
         rd0[i] = d_digit(rk0[i], 0, shift, mask);
         atomicAdd(&h[0][wave][rd0[i]], 1u);
       }
@@ -1197,6 +1143,7 @@ __global__ __launch_bounds__(NTHREADS_SC, 4) void k_scatter_lb_pipe(
 #else
 #define DPRK_WALK_NOW 0
 #endif
+    uint32_t tile_nn = 0;
     if (DPRK_WALK_NOW && threadIdx.x < RADIX) {
       uint64_t excl_g = 0;
       uint32_t spins = 0;
@@ -1232,14 +1179,25 @@ __global__ __launch_bounds__(NTHREADS_SC, 4) void k_scatter_lb_pipe(
     }
 #pragma unroll
     for (int i = 0; i < EPL; i++) {
-      uint32_t idx = wbeg + lane * EPL + i;
+      uint32_t idx = wbeg + i * 64 + lane;
       if (idx < tile_n) {
         uint32_t pos = atomicAdd(&hc[wave][rdc[i]], 1u);
         lk[pos] = rkc[i];
         if (WITH_VALS) lv[pos] = rvc[i];
       }
     }
-    if (tn < ntiles) load_blocked(rkn, rvn, tn);
+    if (tn < ntiles) {
+      const uint64_t startn = tn * (uint64_t)TILE;
+      tile_nn = (uint32_t)(min(startn + TILE, n) - startn);
+#pragma unroll
+      for (int i = 0; i < EPL; i++) {
+        uint32_t idx = wbeg + i * 64 + lane;
+        if (idx < tile_nn) {
+          rkn[i] = keys_in[startn + idx];
+          if (WITH_VALS) rvn[i] = vals_in[startn + idx];
+        }
+      }
+    }
 
     /* lookback (thread = digit) overlaps the in-flight prefetch */
     if (!DPRK_WALK_NOW && threadIdx.x < RADIX) {
@@ -1291,11 +1249,9 @@ __global__ __launch_bounds__(NTHREADS_SC, 4) void k_scatter_lb_pipe(
     for (int j = 0; j < RADIX / 64; j++)
       hc[wave][lane * (RADIX / 64) + j] = 0;
     if (tn < ntiles) {
-      const uint64_t startn = tn * (uint64_t)TILE;
-      const uint32_t tile_nn = (uint32_t)(min(startn + TILE, n) - startn);
 #pragma unroll
       for (int i = 0; i < EPL; i++) {
-        uint32_t idx = wbeg + lane * EPL + i;
+        uint32_t idx = wbeg + i * 64 + lane;
         if (idx < tile_nn) {
           rdn[i] = d_digit(rkn[i], 0, shift, mask);
           atomicAdd(&hn[wave][rdn[i]], 1u);
