@@ -460,6 +460,22 @@ def test_fused_reduce_opt_in():
         del os.environ["DPRK_FUSED_REDUCE"]
 
 
+def test_lsd_cross_pass_stability_regression(eng):
+    """Regression pin for the r02 lane-blocked-loads bug (reverted): a
+    multi-pass sort whose passes each see massive digit collisions comes
+    out BIT-EXACT in key order — any within-digit rank order that is not
+    position order breaks LSD across passes and fails here (the output
+    is then multiset-correct but jumbled; scripts/dbg_pipe.py)."""
+    rng = np.random.default_rng(77)
+    for n in [5000, 8191, 8192, 8193, 16384, 100_000, 3_000_000]:
+        keys = rng.integers(0, 1 << 20, n).astype(np.int64)  # 3 passes
+        vals = rng.random(n)
+        _put(eng, keys, vals)
+        eng.sort()
+        k, _ = eng.get()
+        assert np.array_equal(k, np.sort(keys)), f"n={n}"
+
+
 def test_packed_state_and_boundaries():
     """Packed-u32 key transport (include/dprk.h): the gate is exact at
     the [0, 2^32) boundary, input-deterministic, and set_packed round
