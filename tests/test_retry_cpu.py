@@ -73,6 +73,23 @@ def test_fetch_failed_resubmits_map_stage():
     assert map_calls["n"] == 24          # map stage ran twice (12 + 12)
 
 
+def _flaky_part(it, state={"n": 0}):  # noqa: B006 — per-payload state is the point
+    state["n"] += 1
+    if state["n"] == 1:
+        raise ValueError("transient in forked worker (injected)")
+    return list(it)
+
+
+def test_process_mode_child_retry():
+    """MultiProcessScheduler: the child-side retry loop recovers a
+    transient task failure inside the forked worker."""
+    ctx = DparkContext("process[2]")
+    parts = ctx.runJob(ctx.parallelize(range(8), 2).map(lambda x: x * 2),
+                       _flaky_part)
+    got = sorted(x for p in parts for x in p)
+    assert got == [0, 2, 4, 6, 8, 10, 12, 14]
+
+
 def test_engine_retry_once_on_engine_error():
     """vector._engine_run retries exactly once on EngineError with a
     fresh ctx; EngineUnavailable is never retried."""
