@@ -2382,6 +2382,45 @@ __global__ __launch_bounds__(NTHREADS) void k_join_count(
   }
 }
 
+/* merge co-walk variant of the match step (DPRK_JOIN_MP experiment):
+ * one thread owns WCHUNK consecutive A-runs — ONE binary search finds
+ * its B start, then both sorted unique lists advance together
+ * sequentially (L2-hot streams), ~0.4+2 memory ops per run vs ~8
+ * probes for the wave-bracketed per-lane searches. */
+#define JOIN_WCHUNK 64
+__global__ __launch_bounds__(NTHREADS) void k_join_count_mp(
+    const int64_t *__restrict__ uka, const uint64_t *__restrict__ cnta,
+    uint64_t ga, const int64_t *__restrict__ ukb,
+    const uint64_t *__restrict__ cntb, uint64_t gb,
+    uint64_t *__restrict__ prod, uint64_t *__restrict__ bidx) {
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  uint64_t nchunks = (ga + JOIN_WCHUNK - 1) / JOIN_WCHUNK;
+  for (uint64_t c = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       c < nchunks; c += stride) {
+    uint64_t r0 = c * JOIN_WCHUNK;
+    uint64_t r1 = min(r0 + JOIN_WCHUNK, ga);
+    /* lower_bound(B, uka[r0]) */
+    int64_t k0 = uka[r0];
+    uint64_t lo = 0, hi = gb;
+    while (lo < hi) {
+      uint64_t m = (lo + hi) >> 1;
+      if (ukb[m] < k0) lo = m + 1; else hi = m;
+    }
+    uint64_t b = lo;
+    for (uint64_t r = r0; r < r1; r++) {
+      int64_t k = uka[r];
+      while (b < gb && ukb[b] < k) b++;
+      if (b < gb && ukb[b] == k) {
+        prod[r] = cnta[r] * cntb[b];
+        bidx[r] = b;
+      } else {
+        prod[r] = 0;
+        bidx[r] = (uint64_t)-1;
+      }
+    }
+  }
+}
+
 /* emit: per-key cross products (dpark join semantics — SURVEY.md §3.3).
  * One GLANES-lane sub-wave group per A-run (products are small, so a
  * full wave per run left most lanes idle at low multiplicity — measured
@@ -4105,11 +4144,24 @@ extern "C" int dprk_join(dprk_ctx *a, dprk_ctx *b, uint64_t *n_out) {
     a->run_prod_cap = cap;
   }
   uint64_t *d_bidx = a->d_bidx;
-  TIMED(a, K_JOIN,
-        hipLaunchKernelGGL(k_join_count, dim3(stream_grid(ga)), dim3(NTHREADS),
-                           0, a->stream, a->grp_keys, a->grp_counts, ga,
-                           b->grp_keys, b->grp_counts, gb, a->run_prod,
-                           d_bidx));
+  static int join_mp = -1;
+  if (join_mp < 0) join_mp = getenv("DPRK_JOIN_MP") ? 1 : 0;
+  if (join_mp) {
+    TIMED(a, K_JOIN,
+          hipLaunchKernelGGL(k_join_count_mp,
+                             dim3(stream_grid(ga / JOIN_WCHUNK + 1)),
+                             dim3(NTHREADS),
+                             0, a->stream, a->grp_keys, a->grp_counts, ga,
+                             b->grp_keys, b->grp_counts, gb, a->run_prod,
+                             d_bidx));
+  } else {
+    TIMED(a, K_JOIN,
+          hipLaunchKernelGGL(k_join_count, dim3(stream_grid(ga)),
+                             dim3(NTHREADS),
+                             0, a->stream, a->grp_keys, a->grp_counts, ga,
+                             b->grp_keys, b->grp_counts, gb, a->run_prod,
+                             d_bidx));
+  }
   /* exclusive scan of products -> output offsets (run_boff := scan copy) */
   HIP_CHECK(hipMemcpyAsync(a->run_boff, a->run_prod, ga * 8,
                            hipMemcpyDeviceToDevice, a->stream));
