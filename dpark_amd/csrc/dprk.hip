@@ -128,6 +128,13 @@ struct dprk_ctx {
   int sorted = 0;
   int packed = 0;   /* keys currently stored as packed u32 (all keys in
                        [0, 2^32); input-deterministic — see KT traits) */
+  /* lazy uniform generation (gen->sort fusion): dprk_gen_pairs(dist=0)
+   * only RECORDS the stream; the first sort pass then GENERATES pairs
+   * straight into partitioned layout (dpark's ShuffleMapTask.run never
+   * materializes input before the bucket loop either [U]).  Any op
+   * needing materialized data calls materialize_lazy() first. */
+  int lazy_gen = 0;
+  uint64_t lazy_seed_hi = 0, lazy_keymask = 0, lazy_universe = 0;
   /* conservative OR/AND of biased keys known from generation (digit-skip
    * without the k_minmax_or pass); valid only straight after gen /
    * permutation-only ops */
@@ -652,6 +659,57 @@ __global__ __launch_bounds__(NTHREADS) void k_val_sum(
   if (threadIdx.x == 0) partials[blockIdx.x] = sh[0];
 }
 
+/* hist-only generation prepass (gen->sort fusion): replay the PRNG
+ * stream to produce every active pass's global digit histogram and the
+ * exact biased OR/AND — ZERO global data traffic; the fused first sort
+ * pass then generates pairs straight into partitioned layout. */
+template <typename K>
+__global__ __launch_bounds__(NTHREADS) void k_gen_hist_only(
+    uint64_t n, uint64_t seed_hi, uint64_t keymask, uint64_t active_shifts,
+    int nact, uint32_t *__restrict__ g_hist /* [nact][RADIX] */,
+    uint64_t *__restrict__ orand /* [2] pre-set to {0, ~0} */) {
+  __shared__ uint32_t h[NTHREADS / 64][8][RADIX];
+  __shared__ uint64_t s_or[NTHREADS / 64], s_and[NTHREADS / 64];
+  const int wave = threadIdx.x / 64;
+  for (int i = threadIdx.x; i < (NTHREADS / 64) * 8 * RADIX; i += NTHREADS)
+    ((uint32_t *)h)[i] = 0;
+  __syncthreads();
+  uint64_t acc_or = 0, acc_and = ~0ull;
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    K k = KT<K>::from_raw(d_splitmix64(seed_hi + i) & keymask);
+    uint64_t u = KT<K>::bias(k);
+    acc_or |= u;
+    acc_and &= u;
+    for (int p = 0; p < nact; p++) {
+      int sh = (int)((active_shifts >> (8 * p)) & 0xFF);
+      atomicAdd(&h[wave][p][(u >> sh) & 0xFF], 1u);
+    }
+  }
+  __syncthreads();
+  for (int p = 0; p < nact; p++)
+    for (int d = threadIdx.x; d < RADIX; d += NTHREADS) {
+      uint32_t s = 0;
+      for (int w = 0; w < NTHREADS / 64; w++) s += h[w][p][d];
+      if (s) atomicAdd(&g_hist[p * RADIX + d], s);
+    }
+  for (int off = 32; off; off >>= 1) {
+    acc_or |= (uint64_t)__shfl_down((long long)acc_or, off, 64);
+    acc_and &= (uint64_t)__shfl_down((long long)acc_and, off, 64);
+  }
+  if (threadIdx.x % 64 == 0) { s_or[wave] = acc_or; s_and[wave] = acc_and; }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    for (int w = 1; w < NTHREADS / 64; w++) {
+      acc_or |= s_or[w];
+      acc_and &= s_and[w];
+    }
+    atomicOr((unsigned long long *)&orand[0], (unsigned long long)acc_or);
+    atomicAnd((unsigned long long *)&orand[1], (unsigned long long)acc_and);
+  }
+}
+
 /* OR/AND reduction of biased keys -> constant-digit (skip-pass) detection */
 template <typename K>
 __global__ __launch_bounds__(NTHREADS) void k_minmax_or(
@@ -1037,14 +1095,21 @@ __global__ __launch_bounds__(NTHREADS_SC, 4) void k_scatter_lb(
  * rejection of cross-tile pipelining was measured at 2 WG/CU).  Rank is
  * ds_add_rtn (the r02 default).  Identical output permutation to
  * k_scatter_lb<u32, WV, true>. */
-template <typename K, bool WITH_VALS>
+/* GEN mode (third template param): the "load" of a tile is the PRNG
+ * replay itself — pairs are generated straight into pass-1-partitioned
+ * layout (dpark's map task feeds its bucket loop from the parent
+ * iterator without materializing input either [U]); seed_hi/keymask
+ * are unused when !GEN.  Same strided element mapping and rank order
+ * as the load form. */
+template <typename K, bool WITH_VALS, bool GEN = false>
 __global__ __launch_bounds__(NTHREADS_SC, 4) void k_scatter_lb_pipe(
     const K *__restrict__ keys_in, const double *__restrict__ vals_in,
     K *__restrict__ keys_out, double *__restrict__ vals_out,
     uint64_t n, uint64_t ntiles, int shift,
     const uint32_t *__restrict__ digit_base,
     unsigned long long *__restrict__ lb,
-    unsigned int *__restrict__ d_fail) {
+    unsigned int *__restrict__ d_fail,
+    uint64_t seed_hi = 0, uint64_t keymask = 0) {
   constexpr int NW = NTHREADS_SC / 64;
   constexpr int WBLK = TILE / NW;
   constexpr int EPL = WBLK / 64;
@@ -1083,8 +1148,16 @@ __global__ __launch_bounds__(NTHREADS_SC, 4) void k_scatter_lb_pipe(
     for (int i = 0; i < EPL; i++) {
       uint32_t idx = wbeg + i * 64 + lane;
       if (idx < tile_n) {
-        rk0[i] = keys_in[start + idx];
-        if (WITH_VALS) rv0[i] = vals_in[start + idx];
+        if (GEN) {
+          uint64_t x = seed_hi + start + idx;
+          rk0[i] = KT<K>::from_raw(d_splitmix64(x) & keymask);
+          if (WITH_VALS)
+            rv0[i] = (double)(d_splitmix64(x ^ 0xDEADBEEFCAFEBABEull) >> 11)
+                     * (1.0 / 9007199254740992.0);
+        } else {
+          rk0[i] = keys_in[start + idx];
+          if (WITH_VALS) rv0[i] = vals_in[start + idx];
+        }
         rd0[i] = d_digit(rk0[i], 0, shift, mask);
         atomicAdd(&h[0][wave][rd0[i]], 1u);
       }
@@ -1193,6 +1266,14 @@ __global__ __launch_bounds__(NTHREADS_SC, 4) void k_scatter_lb_pipe(
       for (int i = 0; i < EPL; i++) {
         uint32_t idx = wbeg + i * 64 + lane;
         if (idx < tile_nn) {
+          if (GEN) {
+            uint64_t x = seed_hi + startn + idx;
+            rkn[i] = KT<K>::from_raw(d_splitmix64(x) & keymask);
+            if (WITH_VALS)
+              rvn[i] = (double)(d_splitmix64(x ^ 0xDEADBEEFCAFEBABEull)
+                                >> 11) * (1.0 / 9007199254740992.0);
+            continue;
+          }
           rkn[i] = keys_in[startn + idx];
           if (WITH_VALS) rvn[i] = vals_in[startn + idx];
         }
@@ -2489,6 +2570,7 @@ static int stream_grid(uint64_t work_items) {
 }
 
 static int free_all(dprk_ctx *c);
+static int materialize_lazy(dprk_ctx *c);
 
 /* workgroups for tile-mapped kernels: min(ntiles, cap); cap tunable via
  * DPRK_GRID_CAP (default 2048 — measured flat 512..4096, see DESIGN) */
@@ -2705,6 +2787,7 @@ extern "C" int dprk_reserve(dprk_ctx *c, uint64_t max_n, int with_vals) {
   c->cur = 0;
   c->sorted = 0;
   c->packed = 0;
+  c->lazy_gen = 0;
   return 0;
 }
 
@@ -2839,11 +2922,44 @@ extern "C" int dprk_gen_pairs_at(dprk_ctx *c, uint64_t seed, uint64_t start,
   c->packed = (key_universe != 0 && key_universe <= (1ull << 32) &&
                !getenv("DPRK_NO_PACK"));
   uint64_t seed_hi = (seed << 32) + start;
+  /* gen->sort fusion (dist 0): only RECORD the stream — if the next op
+   * is a sort/reduce, the first pass GENERATES pairs straight into
+   * partitioned layout (saves the gen write + pass-1 read, 24 B/pair);
+   * any other op materializes first (materialize_lazy).  Results are
+   * deterministic per CALL SEQUENCE; fused vs materialized reductions
+   * may differ in the last ulp (different element permutations, same
+   * <=1e-6 oracle contract — like every other path pair here). */
+  if (dist == 0 && n > 0 && !getenv("DPRK_NO_FUSED_GEN")) {
+    c->lazy_gen = 1;
+    c->lazy_seed_hi = seed_hi;
+    c->lazy_keymask = keymask;
+    c->lazy_universe = key_universe;
+    c->n = n;
+    c->sorted = 0;
+    c->orand_valid = 0;
+    return 0;
+  }
+  c->lazy_gen = 0;
   return c->packed
              ? gen_pairs_t<uint32_t>(c, seed_hi, n, dist, key_universe,
                                      keymask)
              : gen_pairs_t<int64_t>(c, seed_hi, n, dist, key_universe,
                                     keymask);
+}
+
+/* materialize a lazily-recorded uniform generation (the normal fused
+ * k_gen_hist path); no-op otherwise */
+static int materialize_lazy(dprk_ctx *c) {
+  if (!c->lazy_gen) return 0;
+  c->lazy_gen = 0;
+  HIP_CHECK(hipSetDevice(c->device));
+  c->hist_shift_valid = -1;
+  c->ghist_valid = 0;
+  return c->packed
+             ? gen_pairs_t<uint32_t>(c, c->lazy_seed_hi, c->n, 0,
+                                     c->lazy_universe, c->lazy_keymask)
+             : gen_pairs_t<int64_t>(c, c->lazy_seed_hi, c->n, 0,
+                                    c->lazy_universe, c->lazy_keymask);
 }
 
 extern "C" int dprk_gen_pairs(dprk_ctx *c, uint64_t seed, uint64_t n,
@@ -2896,6 +3012,7 @@ extern "C" int dprk_put_pairs(dprk_ctx *c, const int64_t *keys,
                               const double *vals, uint64_t n) {
   if (!c) { set_err("null ctx"); return 1; }
   if (n > c->cap) { set_err("put: n > reserved cap"); return 1; }
+  c->lazy_gen = 0;   /* overwritten */
   if (vals && !c->with_vals) { set_err("put: ctx reserved keys-only"); return 1; }
   HIP_CHECK(hipSetDevice(c->device));
   HIP_CHECK(hipMemcpyAsync(c->keys[c->cur], keys, n * 8,
@@ -2917,6 +3034,7 @@ extern "C" int dprk_put_pairs(dprk_ctx *c, const int64_t *keys,
 extern "C" int dprk_get_pairs(dprk_ctx *c, int64_t *keys, double *vals,
                               uint64_t cap, uint64_t *n_out) {
   if (!c) { set_err("null ctx"); return 1; }
+  if (materialize_lazy(c)) return 1;
   *n_out = c->n;
   if (c->n > cap) { set_err("get: cap %llu < n %llu",
                             (unsigned long long)cap,
@@ -2958,6 +3076,7 @@ extern "C" int dprk_put_bytes_keys(dprk_ctx *c, const uint8_t *buf,
                                    uint64_t n_words, const double *vals) {
   if (!c) { set_err("null ctx"); return 1; }
   if (n_words > c->cap) { set_err("put_bytes: n > reserved cap"); return 1; }
+  c->lazy_gen = 0;   /* overwritten */
   if (vals && !c->with_vals) { set_err("put_bytes: ctx reserved keys-only");
                                return 1; }
   HIP_CHECK(hipSetDevice(c->device));
@@ -3000,6 +3119,7 @@ extern "C" int dprk_is_packed(dprk_ctx *c, int *out) {
 
 extern "C" int dprk_set_packed(dprk_ctx *c, int want) {
   if (!c) { set_err("null ctx"); return 1; }
+  if (materialize_lazy(c)) return 1;
   HIP_CHECK(hipSetDevice(c->device));
   if (want == c->packed) return 0;
   if (c->n == 0) { c->packed = want ? 1 : 0; return 0; }
@@ -3061,6 +3181,7 @@ extern "C" int dprk_set_packed(dprk_ctx *c, int want) {
  * block order. */
 extern "C" int dprk_val_sum(dprk_ctx *c, double *out) {
   if (!c || !out) { set_err("null arg"); return 1; }
+  if (materialize_lazy(c)) return 1;
   if (!c->with_vals) { set_err("val_sum: keys-only ctx"); return 1; }
   HIP_CHECK(hipSetDevice(c->device));
   *out = 0.0;
@@ -3407,6 +3528,7 @@ static int radix_pass(dprk_ctx *c, int mode, int shift, uint32_t mask,
 extern "C" int dprk_sample_keys(dprk_ctx *c, uint64_t n_samples,
                                 int64_t *out_host) {
   if (!c) { set_err("null ctx"); return 1; }
+  if (materialize_lazy(c)) return 1;
   HIP_CHECK(hipSetDevice(c->device));
   if (c->n == 0 || n_samples == 0) return 0;
   if (n_samples > c->n) n_samples = c->n;
@@ -3439,6 +3561,7 @@ extern "C" int dprk_sample_keys(dprk_ctx *c, uint64_t n_samples,
 extern "C" int dprk_partition_range(dprk_ctx *c, const int64_t *splitters,
                                     int nbuckets, uint64_t *bucket_counts) {
   if (!c) { set_err("null ctx"); return 1; }
+  if (materialize_lazy(c)) return 1;
   if (nbuckets < 1 || nbuckets > 256) {
     set_err("partition_range: nbuckets must be in [1,256]");
     return 1;
@@ -3560,6 +3683,91 @@ static int sort_pairs_lookback(dprk_ctx *c, const int *passes, int np) {
         +4.7% at 1B, +1.3% zipf; parity suite green */
     static int pipe2 = -1;
     if (pipe2 < 0) pipe2 = getenv("DPRK_PIPE2") ? 1 : 0;
+    if (c->lazy_gen && i == 0) {
+      /* FUSED gen + first pass: pairs are generated straight into
+         pass-0-partitioned layout (write-only: saves the gen write +
+         pass read, 24 B/pair).  Eligibility (pipe form exists) was
+         checked before the prepass in dprk_sort_pairs. */
+      bool launched = false;
+      static int occ_gv = 0, occ_gk = 0;
+      if constexpr (sizeof(K) == 4) {
+        if (!occ_gv) {
+          HIP_CHECK(hipOccupancyMaxActiveBlocksPerMultiprocessor(
+              &occ_gv, (const void *)(k_scatter_lb_pipe<K, true, true>),
+              NTHREADS_SC, 0));
+          HIP_CHECK(hipOccupancyMaxActiveBlocksPerMultiprocessor(
+              &occ_gk, (const void *)(k_scatter_lb_pipe<K, false, true>),
+              NTHREADS_SC, 0));
+          hipDeviceProp_t props;
+          HIP_CHECK(hipGetDeviceProperties(&props, c->device));
+          occ_gv = min(2, max(1, occ_gv)) * props.multiProcessorCount;
+          occ_gk = min(2, max(1, occ_gk)) * props.multiProcessorCount;
+        }
+        int ggrid = (int)min(ntiles, (uint64_t)(c->with_vals ? occ_gv
+                                                             : occ_gk));
+        if (c->with_vals) {
+          TIMED(c, K_SCATTER,
+                hipLaunchKernelGGL((k_scatter_lb_pipe<K, true, true>),
+                                   dim3(ggrid), dim3(NTHREADS_SC), 0,
+                                   c->stream, nullptr, nullptr,
+                                   (K *)c->keys[dst], c->vals[dst], n,
+                                   ntiles, 8 * passes[i], c->d_dbase,
+                                   c->d_lb, c->d_fail, c->lazy_seed_hi,
+                                   c->lazy_keymask));
+        } else {
+          TIMED(c, K_SCATTER,
+                hipLaunchKernelGGL((k_scatter_lb_pipe<K, false, true>),
+                                   dim3(ggrid), dim3(NTHREADS_SC), 0,
+                                   c->stream, nullptr, nullptr,
+                                   (K *)c->keys[dst], nullptr, n,
+                                   ntiles, 8 * passes[i], c->d_dbase,
+                                   c->d_lb, c->d_fail, c->lazy_seed_hi,
+                                   c->lazy_keymask));
+        }
+        launched = true;
+      } else {
+        if (!c->with_vals) {
+          if (!occ_gk) {
+            HIP_CHECK(hipOccupancyMaxActiveBlocksPerMultiprocessor(
+                &occ_gk, (const void *)(k_scatter_lb_pipe<K, false, true>),
+                NTHREADS_SC, 0));
+            hipDeviceProp_t props;
+            HIP_CHECK(hipGetDeviceProperties(&props, c->device));
+            occ_gk = min(2, max(1, occ_gk)) * props.multiProcessorCount;
+          }
+          int ggrid = (int)min(ntiles, (uint64_t)occ_gk);
+          TIMED(c, K_SCATTER,
+                hipLaunchKernelGGL((k_scatter_lb_pipe<K, false, true>),
+                                   dim3(ggrid), dim3(NTHREADS_SC), 0,
+                                   c->stream, nullptr, nullptr,
+                                   (K *)c->keys[dst], nullptr, n,
+                                   ntiles, 8 * passes[i], c->d_dbase,
+                                   c->d_lb, c->d_fail, c->lazy_seed_hi,
+                                   c->lazy_keymask));
+          launched = true;
+        }
+      }
+      if (launched) {
+        uint32_t failg = 0;
+        HIP_CHECK(hipMemcpyAsync(&failg, c->d_fail, 4,
+                                 hipMemcpyDeviceToHost, c->stream));
+        HIP_CHECK(hipStreamSynchronize(c->stream));
+        if (failg) {
+          if (materialize_lazy(c)) return 1;
+          if (radix_pass_t<K>(c, 0, 8 * passes[i], 0xFF, 0)) return 1;
+        } else {
+          c->lazy_gen = 0;
+          uint64_t bpp = sizeof(K) + (c->with_vals ? 8 : 0); /* write only */
+          c->stats.bytes_scatter += n * bpp;
+          c->stats.bytes_total += n * bpp;
+          c->cur = dst;
+        }
+        c->hist_shift_valid = -1;
+        c->stats.sort_passes++;
+        continue;
+      }
+      if (materialize_lazy(c)) return 1;   /* safety net */
+    }
     if constexpr (sizeof(K) == 4) {
       if (pipe) {
       /* software-pipelined variant (u32 only); DPRK_PIPE2 selects the
@@ -3739,8 +3947,59 @@ static int sort_pairs_lookback(dprk_ctx *c, const int *passes, int np) {
 extern "C" int dprk_sort_pairs(dprk_ctx *c) {
   if (!c) { set_err("null ctx"); return 1; }
   HIP_CHECK(hipSetDevice(c->device));
-  if (c->n == 0) { c->sorted = 1; return 0; }
+  if (c->n == 0) { c->sorted = 1; c->lazy_gen = 0; return 0; }
   if (c->sorted) return 0;
+  if (c->lazy_gen) {
+    /* gen->sort fusion: the fused first pass needs the lookback path
+       and a pipelined kernel form; otherwise materialize and go the
+       normal way */
+    uint64_t lbt = ceil_div(c->n, (uint64_t)TILE);
+    int pipe_ok = !getenv("DPRK_NO_PIPE") &&
+                  (c->packed || !c->with_vals);
+    if (getenv("DPRK_NO_LOOKBACK") || !pipe_ok ||
+        lbt * RADIX * 8 > lb_gate_bytes(c->with_vals)) {
+      if (materialize_lazy(c)) return 1;
+    } else {
+      /* hist-only PRNG prepass: every candidate pass's global digit
+         histogram + the exact biased OR/AND — zero data traffic */
+      uint64_t km = c->lazy_keymask;
+      uint64_t shifts = 0;
+      int nact = 0;
+      for (int p = 0; p < 8; p++) {
+        uint32_t mb = (uint32_t)(km >> (8 * p)) & 0xFF;
+        if (km == ~0ull || mb)
+          shifts |= ((uint64_t)(8 * p)) << (8 * (nact++));
+      }
+      if (!c->d_ghist) HIP_CHECK(hipMalloc(&c->d_ghist, 8 * RADIX * 4));
+      HIP_CHECK(hipMemsetAsync(c->d_ghist, 0, 8 * RADIX * 4, c->stream));
+      uint64_t oa_init[2] = {0, ~0ull};
+      HIP_CHECK(hipMemcpyAsync(c->d_orand, oa_init, 16,
+                               hipMemcpyHostToDevice, c->stream));
+      if (c->packed) {
+        TIMED(c, K_GEN,
+              hipLaunchKernelGGL(k_gen_hist_only<uint32_t>,
+                                 dim3(stream_grid(c->n)), dim3(NTHREADS), 0,
+                                 c->stream, c->n, c->lazy_seed_hi, km,
+                                 shifts, nact, c->d_ghist, c->d_orand));
+      } else {
+        TIMED(c, K_GEN,
+              hipLaunchKernelGGL(k_gen_hist_only<int64_t>,
+                                 dim3(stream_grid(c->n)), dim3(NTHREADS), 0,
+                                 c->stream, c->n, c->lazy_seed_hi, km,
+                                 shifts, nact, c->d_ghist, c->d_orand));
+      }
+      uint64_t oa[2];
+      HIP_CHECK(hipMemcpyAsync(oa, c->d_orand, 16, hipMemcpyDeviceToHost,
+                               c->stream));
+      HIP_CHECK(hipStreamSynchronize(c->stream));
+      c->ghist_shifts = shifts;
+      c->ghist_nact = nact;
+      c->ghist_valid = 1;
+      c->orand_valid = 1;
+      c->known_or = oa[0];
+      c->known_and = oa[1];
+    }
+  }
   /* digit-skip detection: OR/AND of biased keys (or the conservative
    * generation-time bound, saving one full read of the keys) */
   uint64_t orand[2];
@@ -3784,11 +4043,16 @@ extern "C" int dprk_sort_pairs(dprk_ctx *c) {
    * lookback hop is an HBM-random load and the walk cost explodes
    * (measured: 1B keys, 500 MB granules -> +4.3 ms/pass).  The gate is
    * a pure function of n => path choice stays input-deterministic. */
+  if (c->lazy_gen && np == 0) {
+    /* all keys identical: nothing to sort — materialize and finish */
+    if (materialize_lazy(c)) return 1;
+  }
   uint64_t lb_tiles = ceil_div(c->n, (uint64_t)TILE);
   if (np && !getenv("DPRK_NO_LOOKBACK") &&
       lb_tiles * RADIX * 8 <= lb_gate_bytes(c->with_vals))
     return c->packed ? sort_pairs_lookback<uint32_t>(c, passes, np)
                      : sort_pairs_lookback<int64_t>(c, passes, np);
+  if (c->lazy_gen && materialize_lazy(c)) return 1;
   for (int i = 0; i < np; i++) {
     if (radix_pass(c, /*mode=*/0, /*shift=*/8 * passes[i], /*mask=*/0xFF, 0))
       return 1;
@@ -3801,6 +4065,7 @@ extern "C" int dprk_sort_pairs(dprk_ctx *c) {
 extern "C" int dprk_partition(dprk_ctx *c, int nbuckets,
                               uint64_t *bucket_counts) {
   if (!c) { set_err("null ctx"); return 1; }
+  if (materialize_lazy(c)) return 1;
   if (nbuckets < 1 || nbuckets > 256 || (nbuckets & (nbuckets - 1))) {
     set_err("partition: nbuckets must be a power of two in [1,256]");
     return 1;
@@ -4267,6 +4532,7 @@ extern "C" int dprk_comm_init(dprk_ctx *c, int nranks, int rank,
 extern "C" int dprk_exchange(dprk_ctx *c, const uint64_t *send_counts,
                              const uint64_t *recv_counts, int nranks) {
   if (!c) { set_err("null ctx"); return 1; }
+  if (materialize_lazy(c)) return 1;
   /* N=1 is a no-op — unless DPRK_FORCE_SELF_EXCHANGE is set, which runs
    * the real grouped ncclSend/ncclRecv self path so the RCCL exchange is
    * exercised on the single-GPU test box. */
@@ -4344,6 +4610,7 @@ extern "C" int dprk_exchange_buckets(dprk_ctx *c,
                                      const uint64_t *recv_matrix,
                                      int nranks) {
   if (!c) { set_err("null ctx"); return 1; }
+  if (materialize_lazy(c)) return 1;
   if (!c->comm) { set_err("exchange_buckets: comm not initialized");
                   return 1; }
   if (nranks != c->nranks) { set_err("exchange_buckets: nranks mismatch");

@@ -460,6 +460,59 @@ def test_fused_reduce_opt_in():
         del os.environ["DPRK_FUSED_REDUCE"]
 
 
+def test_fused_gen_paths():
+    """gen->sort fusion (the first pass GENERATES pairs into
+    partitioned layout — dpark's map task feeds the bucket loop from an
+    iterator without materializing input either [U]): parity vs the
+    oracle on the mirrored stream, determinism, and the materialize
+    fallbacks for ineligible shapes."""
+    e = Engine(device=0)
+    try:
+        n = 4_000_000
+        e.reserve(n, with_vals=True)
+        # fused: gen -> reduce directly (packed, lookback+pipe eligible)
+        e.gen_pairs(seed=3, n=n, key_universe=1 << 22)
+        n1 = e.reduce_by_key_sum()
+        k1, v1 = e.get()
+        mk, mv = gen_pairs(3, n, 1 << 22)
+        wk, wv = so.reduce_by_key_sum(mk, mv)
+        assert n1 == len(wk)
+        assert np.array_equal(k1, wk)
+        np.testing.assert_allclose(v1, wv, rtol=RTOL)
+        # run-to-run bit-identical on the same call sequence
+        e.gen_pairs(seed=3, n=n, key_universe=1 << 22)
+        e.reduce_by_key_sum()
+        k2, v2 = e.get()
+        assert np.array_equal(k1, k2) and np.array_equal(v1, v2)
+        # full-universe WITH values: no pipelined int64 form exists —
+        # must materialize-fallback and stay oracle-exact
+        e.gen_pairs(seed=5, n=1_000_000, key_universe=0)
+        e.reduce_by_key_sum()
+        k3, v3 = e.get()
+        mk3, mv3 = gen_pairs(5, 1_000_000, 1 << 64)
+        wk3, wv3 = so.reduce_by_key_sum(mk3, mv3)
+        assert np.array_equal(k3, wk3)
+        np.testing.assert_allclose(v3, wv3, rtol=RTOL)
+        # gen -> get (materialize path) stays byte-identical to the mirror
+        e.gen_pairs(seed=3, n=100_000, key_universe=1 << 22)
+        gk, gv = e.get()
+        mmk, mmv = gen_pairs(3, 100_000, 1 << 22)
+        assert np.array_equal(gk, mmk) and np.array_equal(gv, mmv)
+    finally:
+        e.close()
+    # keys-only int64 fused sort (config-3 shape, small)
+    e2 = Engine(device=0)
+    try:
+        e2.reserve(2_000_000, with_vals=False)
+        e2.gen_pairs(seed=9, n=2_000_000, key_universe=0)
+        e2.sort()
+        k = e2.get(with_vals=False)
+        mk2, _ = gen_pairs(9, 2_000_000, 1 << 64)
+        assert np.array_equal(k, np.sort(mk2))
+    finally:
+        e2.close()
+
+
 def test_lsd_cross_pass_stability_regression(eng):
     """Regression pin for the r02 lane-blocked-loads bug (reverted): a
     multi-pass sort whose passes each see massive digit collisions comes
