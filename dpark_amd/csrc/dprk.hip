@@ -4249,7 +4249,9 @@ extern "C" int dprk_reduce_by_key_sum(dprk_ctx *c, uint64_t *n_out) {
    * are individually deterministic and oracle-exact. */
   if (!c->sorted && getenv("DPRK_FUSED_REDUCE")) {
     /* fused path: all but the last pass as normal sort passes, the last
-     * fused with the segmented reduce (needs >=1 active pass) */
+     * fused with the segmented reduce (needs >=1 active pass).  This
+     * opt-in path has no gen-fusion form — materialize lazy gen. */
+    if (materialize_lazy(c)) return 1;
     uint64_t orand[2];
     if (c->orand_valid) {
       orand[0] = c->known_or;
