@@ -455,6 +455,15 @@ def test_fused_reduce_opt_in():
             wk, wv = so.reduce_by_key_sum(keys, vals)
             assert got == len(wk) and np.array_equal(k, wk), (n, uni)
             np.testing.assert_allclose(v, wv, rtol=RTOL)
+        # lazy gen + fused-reduce env: must materialize first (r02 fix)
+        e.reserve(1_000_000, with_vals=True)
+        e.gen_pairs(seed=65, n=1_000_000, key_universe=1 << 16)
+        got = e.reduce_by_key_sum()
+        k, v = e.get()
+        mk, mv = gen_pairs(65, 1_000_000, 1 << 16)
+        wk, wv = so.reduce_by_key_sum(mk, mv)
+        assert got == len(wk) and np.array_equal(k, wk)
+        np.testing.assert_allclose(v, wv, rtol=RTOL)
         e.close()
     finally:
         del os.environ["DPRK_FUSED_REDUCE"]
