@@ -272,7 +272,8 @@ def main():
         # --gpus N` relaunches itself under torch.distributed.run, one
         # rank per GPU, so the driver's literal command measures N GPUs
         # instead of silently clamping to 1.
-        port = 29400 + (os.getpid() % 1000)
+        # port range disjoint from the fixed test ports (295xx/296xx)
+        port = 25000 + (os.getpid() % 2000)
         cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
                f"--nproc-per-node={args.gpus}", "--master-addr=127.0.0.1",
                f"--master-port={port}",
